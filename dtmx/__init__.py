@@ -1,0 +1,21 @@
+"""dtmx: MI355X-native elastic data-parallel training framework with the
+capabilities (and Python surface) of the Dynamic Training MXNet fork
+(awslabs/dynamic-training-with-apache-mxnet-on-aws). See README.md.
+
+Usage mirrors `import mxnet as mx`:
+
+    import dtmx as mx
+    kv = mx.kvstore.create('dist_sync')
+    net = mx.models.get_symbol('resnet', num_layers=50)
+    mod = mx.mod.Module(net, context=mx.gpu(0))
+    mod.bind(...); mod.fit(train_iter, kvstore=kv, ...)
+"""
+from . import callback, context, initializer, io, lr_scheduler, metric, model
+from . import kvstore, models, ndarray, optimizer, parallel, profiler
+from . import module as mod
+from .context import Context, cpu, gpu, num_gpus
+from .module import Module
+
+nd = ndarray
+
+__version__ = "0.1.0"

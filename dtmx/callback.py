@@ -1,0 +1,70 @@
+"""Training callbacks (reference python/mxnet/callback.py)."""
+from __future__ import annotations
+
+import logging
+import time
+
+
+class BatchEndParam:
+    def __init__(self, epoch, nbatch, eval_metric, locals=None):
+        self.epoch = epoch
+        self.nbatch = nbatch
+        self.eval_metric = eval_metric
+        self.locals = locals
+
+
+class Speedometer:
+    """Logs samples/sec every `frequent` batches (reference callback.py:140)."""
+
+    def __init__(self, batch_size: int, frequent: int = 50, auto_reset: bool = True):
+        self.batch_size = batch_size
+        self.frequent = frequent
+        self.auto_reset = auto_reset
+        self.init = False
+        self.tic = 0.0
+        self.last_count = 0
+
+    def __call__(self, param: BatchEndParam):
+        count = param.nbatch
+        if self.last_count > count:
+            self.init = False
+        self.last_count = count
+        if self.init:
+            if count % self.frequent == 0:
+                speed = self.frequent * self.batch_size / (time.time() - self.tic)
+                if param.eval_metric is not None:
+                    name_value = param.eval_metric.get_name_value()
+                    if self.auto_reset:
+                        param.eval_metric.reset()
+                    msg = "Epoch[%d] Batch [%d]\tSpeed: %.2f samples/sec"
+                    msg += "\t%s=%f" * len(name_value)
+                    logging.info(msg, param.epoch, count, speed, *sum(name_value, ()))
+                else:
+                    logging.info(
+                        "Iter[%d] Batch [%d]\tSpeed: %.2f samples/sec",
+                        param.epoch, count, speed,
+                    )
+                self.tic = time.time()
+        else:
+            self.init = True
+            self.tic = time.time()
+
+
+def do_checkpoint(prefix: str, period: int = 1):
+    """Epoch-end checkpoint callback (reference callback.py:38; fit.py:87-91)."""
+    period = int(max(1, period))
+
+    def _callback(iter_no, sym, arg, aux):
+        from . import model
+        if (iter_no + 1) % period == 0:
+            model.save_checkpoint(prefix, iter_no + 1, sym, arg, aux)
+
+    return _callback
+
+
+class LogValidationMetricsCallback:
+    def __call__(self, param):
+        if param.eval_metric is None:
+            return
+        for name, value in param.eval_metric.get_name_value():
+            logging.info("Epoch[%d] Validation-%s=%f", param.epoch, name, value)

@@ -1,0 +1,254 @@
+"""Autograd ops: CDNA4 HIP kernels on GPU, torch reference on CPU.
+
+Layout contract (MI355X-first): 4D activations are NHWC in memory (torch
+"channels_last" on logical NCHW tensors); conv weights logical (K,C,R,S) with
+channels_last memory = (K,R,S,C) — the implicit-GEMM B operand layout.
+Compute dtype bf16 (fp32 accumulate inside kernels); CPU path is fp32 torch.
+
+Reference kernel parity:
+  conv fwd/dgrad/wgrad   <- src/operator/nn/convolution.cu (cuDNN/im2col)
+  batch_norm fwd/bwd     <- src/operator/nn/batch_norm.cu:208-360
+  pooling                <- src/operator/nn/pool.cuh
+  relu / add(+relu)      <- mshadow_op elementwise kernels
+  softmax+CE (fused)     <- src/operator/nn/softmax-inl.h:166-260, softmax_output
+  linear                 <- fully_connected-inl.h (cuBLAS) -> dtmx MFMA GEMM
+  sgd/momentum updates   <- optimizer_op-inl.h:86,305,430 -> fused multi-tensor
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from .hip import require_ext
+
+
+def _use_hip(*tensors) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# --------------------------------------------------------------------- conv
+
+class _ConvNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, stride, padding):
+        ext = require_ext()
+        ctx.save_for_backward(x, w)
+        ctx.stride = stride
+        ctx.padding = padding
+        return ext.conv_fwd(x, w, stride, padding)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv_dgrad(dy, w, ctx.stride, ctx.padding, x.shape[2], x.shape[3])
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv_wgrad(x, dy, w.shape[2], w.shape[3], ctx.stride, ctx.padding)
+        return dx, dw, None, None
+
+
+def conv2d(x, w, stride: int = 1, padding: int = 0):
+    if _use_hip(x):
+        return _ConvNHWC.apply(x, w, stride, padding)
+    return F.conv2d(x, w, None, stride, padding)
+
+
+# --------------------------------------------------------------- batch norm
+
+class _BatchNormNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum,
+                eps, fuse_relu):
+        ext = require_ext()
+        if training:
+            y, save_mean, save_invstd = ext.bn_fwd_train(
+                x, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu
+            )
+            ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+            ctx.fuse_relu = fuse_relu
+        else:
+            y = ext.bn_fwd_infer(x, gamma, beta, running_mean, running_var, eps, fuse_relu)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx, dgamma, dbeta = ext.bn_bwd(x, dy, gamma, save_mean, save_invstd,
+                                       ctx.fuse_relu, y)
+        return dx, dgamma, dbeta, None, None, None, None, None, None
+
+
+def batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
+               momentum: float = 0.9, eps: float = 1e-5, fuse_relu: bool = False):
+    """NHWC batch norm. `momentum` follows mxnet semantics:
+    moving = moving*momentum + batch*(1-momentum) (reference batch_norm-inl.h),
+    i.e. torch's momentum is (1 - mxnet momentum)."""
+    if _use_hip(x):
+        return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
+                                    training, momentum, eps, fuse_relu)
+    y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
+                     1.0 - momentum, eps)
+    return F.relu(y) if fuse_relu else y
+
+
+# ------------------------------------------------------------------ pooling
+
+class _MaxPoolNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        ext = require_ext()
+        y, idx = ext.maxpool_fwd(x, kernel, stride, padding)
+        ctx.save_for_backward(idx)
+        ctx.in_shape = x.shape
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        (idx,) = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        return ext.maxpool_bwd(dy, idx, ctx.in_shape[2], ctx.in_shape[3]), None, None, None
+
+
+def max_pool2d(x, kernel: int, stride: int, padding: int = 0):
+    if _use_hip(x):
+        return _MaxPoolNHWC.apply(x, kernel, stride, padding)
+    return F.max_pool2d(x, kernel, stride, padding)
+
+
+class _GlobalAvgPoolNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = require_ext()
+        ctx.in_shape = x.shape
+        return ext.global_avgpool_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        n, c, h, w = ctx.in_shape
+        return ext.global_avgpool_bwd(dy.contiguous(), h, w)
+
+
+def global_avg_pool(x):
+    """(N,C,H,W) -> (N,C)"""
+    if _use_hip(x):
+        return _GlobalAvgPoolNHWC.apply(x)
+    return x.mean(dim=(2, 3))
+
+
+# -------------------------------------------------------------- elementwise
+
+class _ReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = require_ext()
+        y = ext.relu_fwd(x)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        (y,) = ctx.saved_tensors
+        return ext.relu_bwd(dy.contiguous(memory_format=_mf(dy)), y)
+
+
+def relu(x):
+    if _use_hip(x):
+        return _ReLU.apply(x)
+    return F.relu(x)
+
+
+class _AddRelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        ext = require_ext()
+        y = ext.add_relu_fwd(a, b)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        (y,) = ctx.saved_tensors
+        g = ext.relu_bwd(dy.contiguous(memory_format=_mf(dy)), y)
+        return g, g
+
+
+def add_relu(a, b):
+    """Fused residual add + ReLU (resnet hot path: one HBM round trip)."""
+    if _use_hip(a):
+        return _AddRelu.apply(a, b)
+    return F.relu(a + b)
+
+
+def _mf(t):
+    return (
+        torch.channels_last
+        if t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last)
+        else torch.contiguous_format
+    )
+
+
+# ------------------------------------------------------------------- linear
+
+class _Linear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        ext = require_ext()
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        return ext.linear_fwd(x, w, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = ext.linear_dgrad(dy, w) if ctx.needs_input_grad[0] else None
+        dw = ext.linear_wgrad(dy, x) if ctx.needs_input_grad[1] else None
+        db = dy.float().sum(0).to(dy.dtype) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def linear(x, w, bias=None):
+    """x:(M,K) w:(N,K) -> (M,N)   (FullyConnected: X @ W^T + b)"""
+    if _use_hip(x):
+        return _Linear.apply(x, w, bias)
+    return F.linear(x, w, bias)
+
+
+# -------------------------------------------------------- fused softmax + CE
+
+class _SoftmaxCE(torch.autograd.Function):
+    """SoftmaxOutput semantics: forward returns sum-CE loss; backward emits
+    d_logits = (softmax(logits) - onehot(label)) — NOT scaled by 1/batch
+    (the optimizer's rescale_grad handles averaging; reference
+    softmax_output-inl.h)."""
+
+    @staticmethod
+    def forward(ctx, logits, label):
+        ext = require_ext()
+        loss, probs = ext.softmax_ce_fwd(logits, label)
+        ctx.save_for_backward(probs, label)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ext = require_ext()
+        probs, label = ctx.saved_tensors
+        return ext.softmax_ce_bwd(probs, label, dloss), None
+
+
+def softmax_cross_entropy_sum(logits, label):
+    if _use_hip(logits):
+        return _SoftmaxCE.apply(logits, label.to(torch.int32))
+    return F.cross_entropy(logits.float(), label.long(), reduction="sum")

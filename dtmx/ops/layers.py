@@ -1,0 +1,117 @@
+"""nn.Module layers over dtmx ops (the building blocks dtmx.models use).
+
+These hold torch Parameters (so autograd/bucketing/checkpointing work
+unchanged) and route compute through dtmx.ops.functional — HIP kernels on
+MI355X, torch reference on CPU.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from . import functional as DF
+
+
+class Conv2dNHWC(nn.Module):
+    def __init__(self, in_channels, out_channels, kernel_size, stride=1, padding=0,
+                 bias=False):
+        super().__init__()
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        self.kernel_size = kernel_size
+        self.stride = stride
+        self.padding = padding
+        w = torch.empty(out_channels, in_channels, kernel_size, kernel_size)
+        nn.init.kaiming_normal_(w, mode="fan_in", nonlinearity="relu")
+        self.weight = nn.Parameter(w)
+        self.bias = nn.Parameter(torch.zeros(out_channels)) if bias else None
+
+    def forward(self, x):
+        y = DF.conv2d(x, self.weight, self.stride, self.padding)
+        if self.bias is not None:
+            y = y + self.bias.reshape(1, -1, 1, 1).to(y.dtype)
+        return y
+
+    def extra_repr(self):
+        return (f"{self.in_channels}, {self.out_channels}, k={self.kernel_size}, "
+                f"s={self.stride}, p={self.padding}")
+
+
+class BatchNorm2dNHWC(nn.Module):
+    """mxnet-momentum BN (moving = m*moving + (1-m)*batch, m=0.9 default,
+    reference batch_norm-inl.h); running stats tracked in fp32 regardless of
+    compute dtype. Optional fused ReLU (the ResNet hot path)."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.9, fuse_relu=False):
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.fuse_relu = fuse_relu
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+
+    def _apply(self, fn, recurse=True):
+        # keep running stats fp32 when the module is cast to bf16/fp16
+        super()._apply(fn, recurse)
+        self.running_mean.data = self.running_mean.data.float()
+        self.running_var.data = self.running_var.data.float()
+        return self
+
+    def forward(self, x):
+        return DF.batch_norm(
+            x,
+            self.weight,
+            self.bias,
+            self.running_mean,
+            self.running_var,
+            self.training,
+            self.momentum,
+            self.eps,
+            self.fuse_relu,
+        )
+
+
+class ReLU(nn.Module):
+    def forward(self, x):
+        return DF.relu(x)
+
+
+class AddRelu(nn.Module):
+    def forward(self, a, b):
+        return DF.add_relu(a, b)
+
+
+class MaxPool2dNHWC(nn.Module):
+    def __init__(self, kernel_size, stride, padding=0):
+        super().__init__()
+        self.kernel_size = kernel_size
+        self.stride = stride
+        self.padding = padding
+
+    def forward(self, x):
+        return DF.max_pool2d(x, self.kernel_size, self.stride, self.padding)
+
+
+class GlobalAvgPool(nn.Module):
+    def forward(self, x):
+        return DF.global_avg_pool(x)
+
+
+class LinearBF16(nn.Module):
+    def __init__(self, in_features, out_features, bias=True):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        w = torch.empty(out_features, in_features)
+        nn.init.normal_(w, 0, 0.01)
+        self.weight = nn.Parameter(w)
+        self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+
+    def forward(self, x):
+        return DF.linear(x, self.weight, self.bias)

@@ -1,0 +1,100 @@
+"""Gradient bucketing with backward/communication overlap.
+
+The reference overlapped parameter-server pushes with backward through its
+dependency engine (per-key engine vars). The MI355X-native
+equivalent: gradients accumulate directly into flat per-bucket buffers
+(param.grad views), a post-accumulate hook fires an async RCCL all-reduce the
+moment a bucket's last gradient lands, and `finish()` joins the outstanding
+works before the optimizer step. Buckets are sized for xGMI ring collectives
+(7 point-to-point links/GPU, ~153 GB/s each): default 50 MiB so RCCL's
+multi-ring pipelining amortizes per-collective latency without delaying
+overlap (override with DTMX_BUCKET_MB).
+
+Buckets are filled in *reverse* parameter order — backward produces gradients
+roughly last-layer-first, so reverse order makes buckets complete early and
+overlap deepest (same reasoning as the reference's per-key ordering by sorted
+keys, kvstore_nccl.h:213-318, re-derived for collectives).
+"""
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+class GradBucketer:
+    def __init__(self, params: Sequence[torch.nn.Parameter],
+                 bucket_mb: Optional[float] = None, average: bool = False):
+        self.params = [p for p in params if p.requires_grad]
+        self.average = average
+        bucket_mb = bucket_mb or float(os.environ.get("DTMX_BUCKET_MB", "50"))
+        bucket_bytes = int(bucket_mb * 1024 * 1024)
+
+        # assign params to buckets in reverse order
+        self.buckets: List[List[torch.nn.Parameter]] = []
+        cur: List[torch.nn.Parameter] = []
+        cur_bytes = 0
+        for p in reversed(self.params):
+            sz = p.numel() * p.element_size()
+            if cur and cur_bytes + sz > bucket_bytes:
+                self.buckets.append(cur)
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += sz
+        if cur:
+            self.buckets.append(cur)
+
+        # flat buffers + grad views
+        self.flat: List[torch.Tensor] = []
+        self._param_bucket: Dict[int, int] = {}
+        self._pending: List[int] = []
+        self._works: List[Tuple[int, dist.Work]] = []
+        for bi, bucket in enumerate(self.buckets):
+            numel = sum(p.numel() for p in bucket)
+            buf = torch.zeros(numel, dtype=bucket[0].dtype, device=bucket[0].device)
+            off = 0
+            for p in bucket:
+                p.grad = buf[off : off + p.numel()].view_as(p)
+                off += p.numel()
+                self._param_bucket[id(p)] = bi
+            self.flat.append(buf)
+        self._pending = [len(b) for b in self.buckets]
+        self._hooks = [
+            p.register_post_accumulate_grad_hook(self._on_grad_ready) for p in self.params
+        ]
+        self._enabled = dist.is_initialized() and dist.get_world_size() > 1
+
+    # -- per-iteration lifecycle -------------------------------------------
+    def zero_grad(self):
+        for buf in self.flat:
+            buf.zero_()
+        self._pending = [len(b) for b in self.buckets]
+        self._works.clear()
+
+    def _on_grad_ready(self, param: torch.nn.Parameter):
+        bi = self._param_bucket[id(param)]
+        self._pending[bi] -= 1
+        if self._pending[bi] == 0 and self._enabled:
+            work = dist.all_reduce(self.flat[bi], op=dist.ReduceOp.SUM, async_op=True)
+            self._works.append((bi, work))
+
+    def finish(self):
+        """Join outstanding collectives; call before the optimizer step."""
+        for bi, work in self._works:
+            work.wait()
+            if self.average:
+                self.flat[bi].div_(dist.get_world_size())
+        self._works.clear()
+
+    def rebuild_after_membership_change(self):
+        """Communicator changed (elastic re-form): nothing to re-shard in the
+        replicated-DP layout; hooks and buffers stay valid, only the process
+        group changed. Re-read world size lazily in finish()."""
+        self._enabled = dist.is_initialized() and dist.get_world_size() > 1
+
+    def detach(self):
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()
