@@ -1,0 +1,137 @@
+#!/usr/bin/env python3
+"""Flagship benchmark (driver contract; BASELINE.json metric).
+
+Measures ResNet-50 ImageNet-shape training throughput (images/sec, bf16,
+NHWC, synthetic data, random init) through the full dtmx stack: HIP conv/BN/
+pool/GEMM kernels, fused softmax-CE, SGD-momentum multi-precision update, and
+— for N>1 ranks — bucketed RCCL all-reduce overlapped with backward
+(kvstore 'dist_sync' semantics).
+
+    python bench.py --gpus N --steps K --warmup W
+(N>1 is launched by the driver via torch.distributed.run, one rank per GPU.)
+
+Reference numbers this is measured against: BASELINE.md (ResNet-50 training
+images/sec, 1xV100 fp32: 298.51 @bs32, 363.69 @bs128).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch-size", type=int, default=128, help="per-GPU batch")
+    ap.add_argument("--network", type=str, default="resnet")
+    ap.add_argument("--num-layers", type=int, default=50)
+    ap.add_argument("--image-shape", type=str, default="3,224,224")
+    ap.add_argument("--dtype", type=str, default="bfloat16")
+    args = ap.parse_args()
+
+    import dtmx
+    from dtmx.io import SyntheticDataIter
+    from dtmx.models import get_symbol
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    dist_mode = world > 1
+
+    torch.manual_seed(1234)
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        ctx = dtmx.gpu(local_rank)
+        dtype = getattr(torch, args.dtype)
+    else:  # CPU smoke fallback (plumbing check only)
+        device = torch.device("cpu")
+        ctx = dtmx.cpu()
+        dtype = torch.float32
+
+    B = args.batch_size
+    shape = tuple(int(x) for x in args.image_shape.split(","))
+    data_shape = (B,) + shape
+
+    net = get_symbol(args.network, num_layers=args.num_layers, num_classes=1000,
+                     image_shape=args.image_shape)
+    mod = dtmx.Module(net, context=ctx)
+    mod.bind(data_shapes=[("data", data_shape)], label_shapes=[("softmax_label", (B,))],
+             dtype=dtype)
+    mod.init_params()
+    kv = dtmx.kvstore.create("dist_sync" if dist_mode else "device")
+    mod.init_optimizer(kvstore=kv,
+                       optimizer_params=(("learning_rate", 0.1), ("momentum", 0.9),
+                                         ("wd", 1e-4)))
+
+    it = SyntheticDataIter(1000, data_shape, max_iter=10 ** 9, dtype=dtype,
+                           device=device, layout="NHWC")
+
+    def step():
+        batch = it.next()
+        mod.forward_backward(batch)
+        mod.update()
+
+    for _ in range(args.warmup):
+        step()
+
+    import torch.distributed as dist
+
+    def barrier_sync():
+        if dist_mode:
+            dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks -> whole-job throughput
+    if dist_mode:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_gpus = world if dist_mode else 1
+    imgs = B * n_gpus * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    baseline = 363.69  # 1xV100 fp32 bs128; weak scaling vs 1 GPU
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec ResNet-50 ImageNet-shape",
+            "value": round(imgs, 2),
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(imgs / baseline, 3),
+            "dtype": str(dtype).replace("torch.", ""),
+            "data": "synthetic",
+            "config": {
+                "model": f"{args.network}-{args.num_layers}",
+                "global_batch": B * n_gpus,
+                "seq_len": None,
+                "image_shape": args.image_shape,
+                "parallelism": f"dp{n_gpus}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,60 @@
+// dtmx._C python bindings (torch extension).
+#include <torch/extension.h>
+
+namespace dtmx {
+// gemm_conv.hip
+at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&, const c10::optional<at::Tensor>&);
+at::Tensor linear_dgrad(const at::Tensor&, const at::Tensor&);
+at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&);
+at::Tensor conv_fwd(const at::Tensor&, const at::Tensor&, long, long);
+at::Tensor conv_dgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
+at::Tensor conv_wgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
+// batchnorm.hip
+std::vector<at::Tensor> bn_fwd_train(const at::Tensor&, const at::Tensor&,
+                                     const at::Tensor&, at::Tensor, at::Tensor,
+                                     double, double, bool);
+at::Tensor bn_fwd_infer(const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, const at::Tensor&, double, bool);
+std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, bool, const at::Tensor&);
+// pool_elem.hip
+std::vector<at::Tensor> maxpool_fwd(const at::Tensor&, long, long, long);
+at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, long, long, long,
+                       long, long);
+at::Tensor global_avgpool_fwd(const at::Tensor&);
+at::Tensor global_avgpool_bwd(const at::Tensor&, long, long);
+at::Tensor relu_fwd(const at::Tensor&);
+at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
+at::Tensor add_relu_fwd(const at::Tensor&, const at::Tensor&);
+// softmax_opt.hip
+std::vector<at::Tensor> softmax_ce_fwd(const at::Tensor&, const at::Tensor&);
+at::Tensor softmax_ce_bwd(const at::Tensor&, const at::Tensor&, const at::Tensor&);
+void sgd_mom_mp(at::Tensor, const at::Tensor&, at::Tensor, at::Tensor, double,
+                double, double, double, double);
+void sgd_mom_f32(at::Tensor, const at::Tensor&, at::Tensor, double, double,
+                 double, double, double);
+}  // namespace dtmx
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("linear_fwd", &dtmx::linear_fwd);
+  m.def("linear_dgrad", &dtmx::linear_dgrad);
+  m.def("linear_wgrad", &dtmx::linear_wgrad);
+  m.def("conv_fwd", &dtmx::conv_fwd);
+  m.def("conv_dgrad", &dtmx::conv_dgrad);
+  m.def("conv_wgrad", &dtmx::conv_wgrad);
+  m.def("bn_fwd_train", &dtmx::bn_fwd_train);
+  m.def("bn_fwd_infer", &dtmx::bn_fwd_infer);
+  m.def("bn_bwd", &dtmx::bn_bwd);
+  m.def("maxpool_fwd", &dtmx::maxpool_fwd);
+  m.def("maxpool_bwd", &dtmx::maxpool_bwd);
+  m.def("global_avgpool_fwd", &dtmx::global_avgpool_fwd);
+  m.def("global_avgpool_bwd", &dtmx::global_avgpool_bwd);
+  m.def("relu_fwd", &dtmx::relu_fwd);
+  m.def("relu_bwd", &dtmx::relu_bwd);
+  m.def("add_relu_fwd", &dtmx::add_relu_fwd);
+  m.def("softmax_ce_fwd", &dtmx::softmax_ce_fwd);
+  m.def("softmax_ce_bwd", &dtmx::softmax_ce_bwd);
+  m.def("sgd_mom_mp", &dtmx::sgd_mom_mp);
+  m.def("sgd_mom_f32", &dtmx::sgd_mom_f32);
+}

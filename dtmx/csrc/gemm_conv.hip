@@ -1,0 +1,552 @@
+// dtmx MFMA GEMM core + implicit-GEMM convolution for gfx950 (MI355X).
+//
+// One TN GEMM template serves FullyConnected fwd/dgrad/wgrad, conv fwd
+// (im2col gather on the A operand), conv dgrad (dy gather) and conv wgrad
+// (transposed operands + split-K) — the reference's cuBLAS/cuDNN/im2col stack
+// (src/operator/nn/convolution.cu, fully_connected-inl.h, linalg_impl.h)
+// redesigned as CDNA4 MFMA kernels.
+//
+// Structure (cdna_hip_programming.md §5 ladder step 3):
+//   - 128x128 block tile, BK=64, 256 threads (4 waves, 2x2 wave grid,
+//     64x64 per wave = 4x4 fragments of v_mfma_f32_16x16x32_bf16)
+//   - global->LDS staging via 16-B global_load_lds (async, no VGPR round trip)
+//   - LDS images [128 rows][64 k] bf16, XOR-swizzled byte^=((row&7)<<4) on the
+//     per-lane *source* address (rule 21) against ds_read_b128 bank conflicts
+//   - double-buffered; one vmcnt(0)+barrier per K-tile; s_setprio around MFMA
+//   - XCD-bijective block swizzle for L2 locality
+// fp32 accumulation; bf16 (or fp32-atomic split-K) epilogues.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "dtmx_common.h"
+
+namespace dtmx {
+
+// ---------------------------------------------------------------- providers
+// A provider returns the global address of the 16-B piece holding bf16
+// elements [k8*8, k8*8+8) of logical row m, or `zero` when out of range.
+
+struct DenseP {
+  const __bf16* base;
+  const __bf16* zero;
+  uint32_t M, K;  // rows, k extent (elements, multiple of 8)
+  uint32_t ld;    // row stride in elements
+  __device__ __forceinline__ const void* addr(uint32_t m, uint32_t k8) const {
+    uint32_t k = k8 * 8;
+    if (m >= M || k >= K) return zero;
+    return base + (size_t)m * ld + k;
+  }
+};
+
+// conv forward A: row m = (n,p,q) output pixel, k = (r,s,c), x is NHWC.
+struct ConvFwdA {
+  const __bf16* x;
+  const __bf16* zero;
+  uint32_t M, Ktot;              // M = N*P*Q, Ktot = R*S*C
+  uint32_t C, H, W, Q, S;
+  int u, v, ph, pw;              // stride, padding
+  FastDiv dQ, dPQ, dC, dS;
+  __device__ __forceinline__ const void* addr(uint32_t m, uint32_t k8) const {
+    uint32_t k = k8 * 8;
+    if (m >= M || k >= Ktot) return zero;
+    uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
+    uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
+    uint32_t rs = dC.div(k), c = dC.mod(k, rs);
+    uint32_t r = dS.div(rs), s = dS.mod(rs, r);
+    int ih = (int)(p * u) - ph + (int)r;
+    int iw = (int)(q * v) - pw + (int)s;
+    if ((uint32_t)ih >= H || (uint32_t)iw >= W) return zero;
+    return x + (((size_t)n * H + ih) * W + iw) * C + c;
+  }
+};
+
+// conv dgrad A: row m = (n,h,w) input pixel, k = (r,s,ko), dy is NHWC.
+// dx[n,h,w,c] = sum_{r,s,ko} dy[n,(h+ph-r)/u,(w+pw-s)/v,ko] * w[ko,r,s,c]
+struct ConvDgradA {
+  const __bf16* dy;
+  const __bf16* zero;
+  uint32_t M, Ktot;              // M = N*H*W, Ktot = R*S*Kout
+  uint32_t Ko, H, W, P, Q, S;
+  int u, v, ph, pw;
+  FastDiv dW_, dHW, dKo, dS;
+  __device__ __forceinline__ const void* addr(uint32_t m, uint32_t k8) const {
+    uint32_t k = k8 * 8;
+    if (m >= M || k >= Ktot) return zero;
+    uint32_t n = dHW.div(m), hw = dHW.mod(m, n);
+    uint32_t h = dW_.div(hw), w = dW_.mod(hw, h);
+    uint32_t rs = dKo.div(k), ko = dKo.mod(k, rs);
+    uint32_t r = dS.div(rs), s = dS.mod(rs, r);
+    int hp = (int)h + ph - (int)r;
+    int wp = (int)w + pw - (int)s;
+    if (hp < 0 || wp < 0) return zero;
+    uint32_t p = (uint32_t)hp, q = (uint32_t)wp;
+    if (u > 1) { if (hp % u) return zero; p = hp / u; }
+    if (v > 1) { if (wp % v) return zero; q = wp / v; }
+    if (p >= P || q >= Q) return zero;
+    return dy + (((size_t)n * P + p) * Q + q) * Ko + ko;
+  }
+};
+
+// ---------------------------------------------------------------- epilogues
+
+struct EpiBF16 {
+  __bf16* c;
+  const float* bias;  // nullable
+  uint32_t M, N;
+  int relu;
+  __device__ __forceinline__ void store(const f32x4 (&acc)[4][4], uint32_t m0,
+                                        uint32_t n0, uint32_t lane) const {
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        uint32_t n = n0 + j * 16 + (lane & 15);
+        if (n >= N) continue;
+        float b = bias ? bias[n] : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          uint32_t m = m0 + i * 16 + ((lane >> 4) << 2) + r;
+          if (m >= M) continue;
+          float v = acc[i][j][r] + b;
+          if (relu) v = fmaxf(v, 0.f);
+          c[(size_t)m * N + n] = (__bf16)v;
+        }
+      }
+  }
+};
+
+struct EpiAtomicF32 {  // split-K partial accumulation (conv wgrad)
+  float* c;
+  uint32_t M, N;
+  __device__ __forceinline__ void store(const f32x4 (&acc)[4][4], uint32_t m0,
+                                        uint32_t n0, uint32_t lane) const {
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        uint32_t n = n0 + j * 16 + (lane & 15);
+        if (n >= N) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          uint32_t m = m0 + i * 16 + ((lane >> 4) << 2) + r;
+          if (m >= M) continue;
+          atomicAdd(&c[(size_t)m * N + n], acc[i][j][r]);
+        }
+      }
+  }
+};
+
+// ------------------------------------------------------------------- kernel
+
+template <class PA, class PB, class EPI>
+__launch_bounds__(256, 2) __global__
+void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles, uint32_t tiles_n,
+                    uint32_t kt0) {
+  __shared__ __bf16 smem[2][2][128 * 64];  // [buf][A|B][row*64+col] 64 KiB
+  const uint32_t t = threadIdx.x;
+  const uint32_t wave = t >> 6, lane = t & 63;
+  const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const uint32_t bm = (bid / tiles_n) * 128, bn = (bid % tiles_n) * 128;
+
+  // staging geometry: iteration it covers rows it*32+(t>>3), byte col (t&7)*16;
+  // XOR swizzle applied to the SOURCE k-piece so the LDS image stays
+  // lane-linear for global_load_lds (rule 21).
+  uint32_t srow[4], sk8[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    srow[it] = it * 32 + (t >> 3);
+    uint32_t colb = (t & 7) * 16;
+    sk8[it] = (colb ^ ((srow[it] & 7) << 4)) >> 4;
+  }
+
+  auto stage = [&](int buf, uint32_t kt) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it)
+      glds16(pa.addr(bm + srow[it], kt * 8 + sk8[it]),
+             &smem[buf][0][it * 2048 + wave * 512]);
+#pragma unroll
+    for (int it = 0; it < 4; ++it)
+      glds16(pb.addr(bn + srow[it], kt * 8 + sk8[it]),
+             &smem[buf][1][it * 2048 + wave * 512]);
+  };
+
+  const uint32_t wr = (wave >> 1) * 64, wc = (wave & 1) * 64;
+  f32x4 acc[4][4] = {};
+
+  stage(0, kt0);
+  wait_vmcnt0();
+  __syncthreads();
+  int cur = 0;
+  for (uint32_t kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) stage(cur ^ 1, kt0 + kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        uint32_t row = wr + i * 16 + (lane & 15);
+        uint32_t colb = (kk * 64 + ((lane >> 4) << 4)) ^ ((row & 7) << 4);
+        af[i] = *(const bf16x8*)((const char*)&smem[cur][0][0] + row * 128 + colb);
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        uint32_t row = wc + j * 16 + (lane & 15);
+        uint32_t colb = (kk * 64 + ((lane >> 4) << 4)) ^ ((row & 7) << 4);
+        bfr[j] = *(const bf16x8*)((const char*)&smem[cur][1][0] + row * 128 + colb);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j],
+                                                              acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    wait_vmcnt0();
+    __syncthreads();
+    cur ^= 1;
+  }
+  epi.store(acc, bm + wr, bn + wc, lane);
+}
+
+// --------------------------------------------------- transpose (row-gather)
+// out[c][m] = row(m)[c]  for generic [M][C] -> [C][out_ld] bf16 transposes
+// (linear dgrad/wgrad operands, conv-wgrad dy^T and gathered im2col^T).
+// C % 8 == 0 required; rows beyond M (or gather-invalid) write zeros.
+
+struct IdentityRows {
+  const __bf16* base;
+  uint32_t M, ld;
+  __device__ __forceinline__ const __bf16* row(uint32_t m) const {
+    return m < M ? base + (size_t)m * ld : nullptr;
+  }
+};
+
+// rows of the im2col matrix for a FIXED (r,s): row m = (n,p,q) -> x pixel
+struct Im2colRows {
+  const __bf16* x;
+  uint32_t M, C, H, W, Q;
+  int u, v, ph, pw, r, s;
+  FastDiv dQ, dPQ;
+  __device__ __forceinline__ const __bf16* row(uint32_t m) const {
+    if (m >= M) return nullptr;
+    uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
+    uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
+    int ih = (int)(p * u) - ph + r;
+    int iw = (int)(q * v) - pw + s;
+    if ((uint32_t)ih >= H || (uint32_t)iw >= W) return nullptr;
+    return x + (((size_t)n * H + ih) * W + iw) * C;
+  }
+};
+
+template <class ROWS>
+__global__ void transpose_rowgather_kernel(ROWS rows, __bf16* out, uint32_t C,
+                                           uint32_t out_ld, uint32_t row_off,
+                                           uint32_t tiles_m) {
+  __shared__ __bf16 tile[64][72];  // +8 pad (16 B) against bank conflicts
+  const uint32_t t = threadIdx.x;
+  const uint32_t bm = (blockIdx.x % tiles_m) * 64;   // input row tile
+  const uint32_t bc = (blockIdx.x / tiles_m) * 64;   // input col tile
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    uint32_t rl = it * 32 + (t >> 3), cl = (t & 7) * 8;
+    const __bf16* p = rows.row(bm + rl);
+    bf16x8 v = {};
+    if (p && bc + cl < C) v = *(const bf16x8*)(p + bc + cl);
+    *(bf16x8*)&tile[rl][cl] = v;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    uint32_t cl = it * 32 + (t >> 3), ml = (t & 7) * 8;
+    if (bc + cl >= C || bm + ml >= out_ld) continue;
+    bf16x8 v;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) v[e] = tile[ml + e][cl];
+    *(bf16x8*)&out[(size_t)(row_off + bc + cl) * out_ld + bm + ml] = v;
+  }
+}
+
+// ------------------------------------------------------- im2col (small-C)
+// Explicit im2col for C % 8 != 0 (the 3-channel stem): out[m][k] with
+// k = (r*S+s)*C + c, zero-padded to Kpad (reference nn/im2col.cuh analog).
+__global__ void im2col_kernel(const __bf16* x, __bf16* out, uint32_t M,
+                              uint32_t Kpad, uint32_t Ktot, uint32_t C,
+                              uint32_t H, uint32_t W, uint32_t Q, uint32_t S,
+                              int u, int v, int ph, int pw, FastDiv dQ,
+                              FastDiv dPQ, FastDiv dC, FastDiv dS) {
+  size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)M * Kpad;
+  for (; idx < total; idx += (size_t)gridDim.x * blockDim.x) {
+    uint32_t m = idx / Kpad, k = idx % Kpad;
+    __bf16 val = (__bf16)0.f;
+    if (k < Ktot) {
+      uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
+      uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
+      uint32_t rs = dC.div(k), c = dC.mod(k, rs);
+      uint32_t r = dS.div(rs), s = dS.mod(rs, r);
+      int ih = (int)(p * u) - ph + (int)r;
+      int iw = (int)(q * v) - pw + (int)s;
+      if ((uint32_t)ih < H && (uint32_t)iw < W)
+        val = x[(((size_t)n * H + ih) * W + iw) * C + c];
+    }
+    out[idx] = val;
+  }
+}
+
+// ============================================================== host side ==
+
+static const __bf16* zero_page(const at::Tensor& like) {
+  static at::Tensor z;
+  if (!z.defined() || z.device() != like.device())
+    z = at::zeros({64}, like.options().dtype(at::kBFloat16));
+  return (const __bf16*)z.data_ptr();
+}
+
+static inline uint32_t ceil_div(uint32_t a, uint32_t b) { return (a + b - 1) / b; }
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_BF16_CUDA(t) \
+  TORCH_CHECK((t).is_cuda() && (t).scalar_type() == at::kBFloat16, #t " must be a CUDA bf16 tensor")
+
+template <class PA, class PB, class EPI>
+static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
+                        uint32_t N, uint32_t K, uint32_t splitk = 1) {
+  uint32_t tiles_m = ceil_div(M, 128), tiles_n = ceil_div(N, 128);
+  uint32_t ktiles_total = ceil_div(K, 64);
+  uint32_t kt_per = ceil_div(ktiles_total, splitk);
+  dim3 grid(tiles_m * tiles_n, splitk);
+  if (splitk == 1) {
+    gemm_tn_kernel<PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+        pa, pb, epi, ktiles_total, tiles_n, 0);
+  } else {
+    // launch one grid per slice on the same stream (kt0 varies); the
+    // fp32-atomic epilogue makes slices order-independent.
+    for (uint32_t s = 0; s < splitk; ++s) {
+      uint32_t kt0 = s * kt_per;
+      if (kt0 >= ktiles_total) break;
+      uint32_t kt = std::min(kt_per, ktiles_total - kt0);
+      gemm_tn_kernel<PA, PB, EPI><<<dim3(tiles_m * tiles_n), 256, 0, cur_stream()>>>(
+          pa, pb, epi, kt, tiles_n, kt0);
+    }
+  }
+}
+
+// -- plain GEMM entry points (FullyConnected; reference fully_connected-inl.h)
+
+at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  uint32_t M = xc.size(0), K = xc.size(1), N = wc.size(0);
+  TORCH_CHECK(K % 8 == 0, "linear: in_features must be a multiple of 8");
+  auto y = at::empty({(long)M, (long)N}, x.options());
+  at::Tensor bias_f;
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->to(at::kFloat).contiguous();
+    bp = bias_f.data_ptr<float>();
+  }
+  DenseP pa{(const __bf16*)xc.data_ptr(), zero_page(x), M, K, K};
+  DenseP pb{(const __bf16*)wc.data_ptr(), zero_page(x), N, K, K};
+  EpiBF16 epi{(__bf16*)y.data_ptr(), bp, M, N, 0};
+  launch_gemm(pa, pb, epi, M, N, K);
+  return y;
+}
+
+static at::Tensor transpose2d(const at::Tensor& in, uint32_t out_ld_pad = 0) {
+  // [M][C] -> [C][Mpad64?]: out_ld defaults to M rounded to 8
+  CHECK_BF16_CUDA(in);
+  auto inc = in.contiguous();
+  uint32_t M = inc.size(0), C = inc.size(1);
+  uint32_t out_ld = out_ld_pad ? out_ld_pad : ((M + 7) / 8) * 8;
+  TORCH_CHECK(C % 8 == 0, "transpose2d: cols must be a multiple of 8");
+  auto out = at::empty({(long)C, (long)out_ld}, in.options());
+  uint32_t tiles_m = ceil_div(out_ld, 64), tiles_c = ceil_div(C, 64);
+  IdentityRows rows{(const __bf16*)inc.data_ptr(), M, C};
+  transpose_rowgather_kernel<IdentityRows>
+      <<<tiles_m * tiles_c, 256, 0, cur_stream()>>>(
+          rows, (__bf16*)out.data_ptr(), C, out_ld, 0, tiles_m);
+  return out;
+}
+
+at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
+  CHECK_BF16_CUDA(dy);
+  uint32_t M = dy.size(0), N = dy.size(1), K = w.size(1);
+  TORCH_CHECK(N % 8 == 0, "linear dgrad: out_features must be a multiple of 8");
+  auto dyc = dy.contiguous();
+  auto wt = transpose2d(w.contiguous(), ((N + 7) / 8) * 8);  // [K][Npad]
+  auto dx = at::empty({(long)M, (long)K}, dy.options());
+  uint32_t Npad = wt.size(1);
+  DenseP pa{(const __bf16*)dyc.data_ptr(), zero_page(dy), M, N, N};
+  DenseP pb{(const __bf16*)wt.data_ptr(), zero_page(dy), K, N, Npad};
+  EpiBF16 epi{(__bf16*)dx.data_ptr(), nullptr, M, K, 0};
+  launch_gemm(pa, pb, epi, M, K, N);
+  return dx;
+}
+
+at::Tensor linear_wgrad(const at::Tensor& dy, const at::Tensor& x) {
+  CHECK_BF16_CUDA(dy);
+  uint32_t M = dy.size(0), N = dy.size(1), K = x.size(1);
+  uint32_t Mpad = ((M + 7) / 8) * 8;
+  auto dyt = transpose2d(dy.contiguous(), Mpad);  // [N][Mpad]
+  auto xt = transpose2d(x.contiguous(), Mpad);    // [K][Mpad]
+  auto dw = at::empty({(long)N, (long)K}, dy.options());
+  DenseP pa{(const __bf16*)dyt.data_ptr(), zero_page(dy), N, Mpad, Mpad};
+  DenseP pb{(const __bf16*)xt.data_ptr(), zero_page(dy), K, Mpad, Mpad};
+  EpiBF16 epi{(__bf16*)dw.data_ptr(), nullptr, N, K, 0};
+  launch_gemm(pa, pb, epi, N, K, Mpad);
+  return dw;
+}
+
+// ------------------------------------------------------------- conv fwd
+
+static void conv_out_dims(uint32_t H, uint32_t W, uint32_t R, uint32_t S,
+                          int stride, int pad, uint32_t& P, uint32_t& Q) {
+  P = (H + 2 * pad - R) / stride + 1;
+  Q = (W + 2 * pad - S) / stride + 1;
+}
+
+at::Tensor conv_fwd(const at::Tensor& x, const at::Tensor& w, long stride,
+                    long pad) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x must be NHWC");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast), "w must be KRSC");
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+  uint32_t Ko = w.size(0), R = w.size(2), S = w.size(3);
+  uint32_t P, Q;
+  conv_out_dims(H, W_, R, S, stride, pad, P, Q);
+  auto y = at::empty({(long)N, (long)Ko, (long)P, (long)Q},
+                     x.options(), at::MemoryFormat::ChannelsLast);
+  uint32_t M = N * P * Q, Ktot = R * S * C;
+  EpiBF16 epi{(__bf16*)y.data_ptr(), nullptr, M, Ko, 0};
+  if (C % 8 == 0) {
+    DenseP pb{(const __bf16*)w.data_ptr(), zero_page(x), Ko, Ktot, Ktot};
+    ConvFwdA pa;
+    pa.x = (const __bf16*)x.data_ptr();
+    pa.zero = zero_page(x);
+    pa.M = M; pa.Ktot = Ktot; pa.C = C; pa.H = H; pa.W = W_; pa.Q = Q; pa.S = S;
+    pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
+    pa.dQ.init(Q); pa.dPQ.init(P * Q); pa.dC.init(C); pa.dS.init(S);
+    launch_gemm(pa, pb, epi, M, Ko, Ktot);
+  } else {
+    // small-C path (3-channel stem): materialized im2col, then dense GEMM.
+    uint32_t Kpad = ((Ktot + 63) / 64) * 64;
+    auto col = at::empty({(long)M, (long)Kpad}, x.options());
+    FastDiv dQ, dPQ, dC, dS;
+    dQ.init(Q); dPQ.init(P * Q); dC.init(C); dS.init(S);
+    size_t total = (size_t)M * Kpad;
+    uint32_t blocks = std::min<size_t>((total + 255) / 256, 16384);
+    im2col_kernel<<<blocks, 256, 0, cur_stream()>>>(
+        (const __bf16*)x.data_ptr(), (__bf16*)col.data_ptr(), M, Kpad, Ktot, C,
+        H, W_, Q, S, stride, stride, pad, pad, dQ, dPQ, dC, dS);
+    // zero-pad the weight rows to Kpad so the k tail multiplies 0*0, not 0*NaN
+    auto wpad = at::constant_pad_nd(
+        w.permute({0, 2, 3, 1}).reshape({(long)Ko, (long)Ktot}),
+        {0, (long)(Kpad - Ktot)}, 0.0).contiguous();
+    DenseP pa{(const __bf16*)col.data_ptr(), zero_page(x), M, Kpad, Kpad};
+    DenseP pb{(const __bf16*)wpad.data_ptr(), zero_page(x), Ko, Kpad, Kpad};
+    launch_gemm(pa, pb, epi, M, Ko, Kpad);
+  }
+  return y;
+}
+
+// ------------------------------------------------------------- conv dgrad
+
+at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
+                      long pad, long H, long W_) {
+  CHECK_BF16_CUDA(dy);
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy must be NHWC");
+  uint32_t N = dy.size(0), Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
+  uint32_t C = w.size(1), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(Ko % 8 == 0, "conv dgrad: out channels must be a multiple of 8");
+  // B operand: W^T in (C,R,S,Ko) contiguous layout
+  auto wt = w.permute({1, 2, 3, 0}).contiguous();  // logical (C,R,S,K), dense
+  auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
+                      at::MemoryFormat::ChannelsLast);
+  uint32_t M = N * H * W_, Ktot = R * S * Ko;
+  ConvDgradA pa;
+  pa.dy = (const __bf16*)dy.data_ptr();
+  pa.zero = zero_page(dy);
+  pa.M = M; pa.Ktot = Ktot; pa.Ko = Ko; pa.H = H; pa.W = W_; pa.P = P; pa.Q = Q;
+  pa.S = S; pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
+  pa.dW_.init(W_); pa.dHW.init(H * W_); pa.dKo.init(Ko); pa.dS.init(S);
+  DenseP pb{(const __bf16*)wt.data_ptr(), zero_page(dy), C, Ktot, Ktot};
+  EpiBF16 epi{(__bf16*)dx.data_ptr(), nullptr, M, C, 0};
+  launch_gemm(pa, pb, epi, M, C, Ktot);
+  return dx;
+}
+
+// ------------------------------------------------------------- conv wgrad
+// dW[ko][(r,s,c)] = sum_{n,p,q} dy[n,p,q,ko] * x[n,pu-ph+r,qv-pw+s,c]
+// as TN GEMM over Kd = NPQ with transposed operands + split-K fp32 atomics.
+
+at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
+                      long stride, long pad) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(dy);
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+  uint32_t Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
+  uint32_t M = N * P * Q;               // the contraction length
+  uint32_t Mpad = ((M + 7) / 8) * 8;
+  uint32_t RSC = R * S * C;
+
+  // dy^T : [NPQ][Ko] -> [Ko][Mpad]
+  auto dyt = transpose2d(dy.reshape({(long)M, (long)Ko}), Mpad);
+
+  // gathered im2col^T: [RSC][Mpad] (one transpose-gather launch per (r,s))
+  at::Tensor xt;
+  if (C % 8 == 0) {
+    xt = at::empty({(long)RSC, (long)Mpad}, x.options());
+    uint32_t tiles_m = ceil_div(Mpad, 64), tiles_c = ceil_div(C, 64);
+    for (uint32_t r = 0; r < (uint32_t)R; ++r)
+      for (uint32_t s = 0; s < (uint32_t)S; ++s) {
+        Im2colRows rows;
+        rows.x = (const __bf16*)x.data_ptr();
+        rows.M = M; rows.C = C; rows.H = H; rows.W = W_; rows.Q = Q;
+        rows.u = stride; rows.v = stride; rows.ph = pad; rows.pw = pad;
+        rows.r = r; rows.s = s;
+        rows.dQ.init(Q); rows.dPQ.init(P * Q);
+        transpose_rowgather_kernel<Im2colRows>
+            <<<tiles_m * tiles_c, 256, 0, cur_stream()>>>(
+                rows, (__bf16*)xt.data_ptr(), C, Mpad, (r * S + s) * C, tiles_m);
+      }
+  } else {
+    // stem: materialize im2col then transpose
+    uint32_t Kpad = ((RSC + 7) / 8) * 8;
+    auto col = at::empty({(long)M, (long)Kpad}, x.options());
+    FastDiv dQ, dPQ, dC, dS;
+    dQ.init(Q); dPQ.init(P * Q); dC.init(C); dS.init(S);
+    size_t total = (size_t)M * Kpad;
+    uint32_t blocks = std::min<size_t>((total + 255) / 256, 16384);
+    im2col_kernel<<<blocks, 256, 0, cur_stream()>>>(
+        (const __bf16*)x.data_ptr(), (__bf16*)col.data_ptr(), M, Kpad, RSC, C,
+        H, W_, Q, S, stride, stride, pad, pad, dQ, dPQ, dC, dS);
+    xt = transpose2d(col, Mpad).narrow(0, 0, RSC).contiguous();
+  }
+
+  // choose split-K so the grid fills the chip (~>=512 blocks)
+  uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, 128);
+  uint32_t ktiles = ceil_div(Mpad, 64);
+  uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 512 / std::max(1u, tiles_mn)));
+
+  auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
+  DenseP pa{(const __bf16*)dyt.data_ptr(), zero_page(x), Ko, Mpad, Mpad};
+  DenseP pb{(const __bf16*)xt.data_ptr(), zero_page(x), RSC, Mpad, Mpad};
+  EpiAtomicF32 epi{dw32.data_ptr<float>(), Ko, RSC};
+  launch_gemm(pa, pb, epi, Ko, RSC, Mpad, splitk);
+  // (Ko, R, S, C) fp32 -> bf16, viewed back to logical (K,C,R,S) channels_last
+  auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
+                .to(at::kBFloat16)
+                .permute({0, 3, 1, 2});
+  return dw.contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+}  // namespace dtmx

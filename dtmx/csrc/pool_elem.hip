@@ -1,0 +1,244 @@
+// NHWC pooling + elementwise kernels for gfx950
+// (reference src/operator/nn/pool.cuh, mshadow_op elementwise kernels).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "dtmx_common.h"
+
+namespace dtmx {
+
+static hipStream_t pe_stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+// ------------------------------------------------------------- max pooling
+
+__global__ void maxpool_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+                                   uint8_t* __restrict__ idx, uint32_t N, uint32_t C,
+                                   uint32_t H, uint32_t W, uint32_t P, uint32_t Q,
+                                   uint32_t K, int u, int pad) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t total = (size_t)N * P * Q * C;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    uint32_t c = i % C;
+    size_t m = i / C;
+    uint32_t q = m % Q;
+    uint32_t p = (m / Q) % P;
+    uint32_t n = m / ((size_t)P * Q);
+    float best = -3.4e38f;
+    uint8_t besti = 0;
+    for (uint32_t kh = 0; kh < K; ++kh) {
+      int ih = (int)(p * u) - pad + (int)kh;
+      if ((uint32_t)ih >= H) continue;
+      for (uint32_t kw = 0; kw < K; ++kw) {
+        int iw = (int)(q * u) - pad + (int)kw;
+        if ((uint32_t)iw >= W) continue;
+        float v = (float)x[(((size_t)n * H + ih) * W + iw) * C + c];
+        if (v > best) { best = v; besti = kh * K + kw; }
+      }
+    }
+    y[i] = (__bf16)best;
+    idx[i] = besti;
+  }
+}
+
+__global__ void maxpool_bwd_kernel(const __bf16* __restrict__ dy,
+                                   const uint8_t* __restrict__ idx,
+                                   __bf16* __restrict__ dx, uint32_t N, uint32_t C,
+                                   uint32_t H, uint32_t W, uint32_t P, uint32_t Q,
+                                   uint32_t K, int u, int pad) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t total = (size_t)N * H * W * C;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    uint32_t c = i % C;
+    size_t m = i / C;
+    uint32_t w = m % W;
+    uint32_t h = (m / W) % H;
+    uint32_t n = m / ((size_t)H * W);
+    float acc = 0.f;
+    // windows (p,q) that contain (h,w): p*u - pad <= h < p*u - pad + K
+    int plo = ((int)h + pad - (int)K + u) / u;  // ceil((h+pad-K+1)/u)
+    if (plo < 0) plo = 0;
+    int phi = ((int)h + pad) / u;
+    if (phi >= (int)P) phi = P - 1;
+    int qlo = ((int)w + pad - (int)K + u) / u;
+    if (qlo < 0) qlo = 0;
+    int qhi = ((int)w + pad) / u;
+    if (qhi >= (int)Q) qhi = Q - 1;
+    for (int p = plo; p <= phi; ++p) {
+      uint32_t kh = (uint32_t)((int)h + pad - p * u);
+      if (kh >= K) continue;
+      for (int q = qlo; q <= qhi; ++q) {
+        uint32_t kw = (uint32_t)((int)w + pad - q * u);
+        if (kw >= K) continue;
+        size_t o = (((size_t)n * P + p) * Q + q) * C + c;
+        if (idx[o] == kh * K + kw) acc += (float)dy[o];
+      }
+    }
+    dx[i] = (__bf16)acc;
+  }
+}
+
+// ------------------------------------------------------- global average pool
+
+__global__ void gap_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+                               uint32_t N, uint32_t C, uint32_t HW) {
+  uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t n = blockIdx.y;
+  if (c >= C) return;
+  const __bf16* base = x + (size_t)n * HW * C + c;
+  float s = 0.f;
+  for (uint32_t i = 0; i < HW; ++i) s += (float)base[(size_t)i * C];
+  y[(size_t)n * C + c] = (__bf16)(s / HW);
+}
+
+__global__ void gap_bwd_kernel(const __bf16* __restrict__ dy, __bf16* __restrict__ dx,
+                               uint32_t N, uint32_t C, uint32_t HW) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t total = (size_t)N * HW * C;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  float inv = 1.f / HW;
+  for (; i < total; i += stride) {
+    uint32_t c = i % C;
+    uint32_t n = i / ((size_t)HW * C);
+    dx[i] = (__bf16)((float)dy[(size_t)n * C + c] * inv);
+  }
+}
+
+// -------------------------------------------------------------- elementwise
+
+__global__ void relu_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+                                size_t total8) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    bf16x8 v = *(const bf16x8*)(x + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)fmaxf((float)v[e], 0.f);
+    *(bf16x8*)(y + i * 8) = o;
+  }
+}
+
+__global__ void relu_bwd_kernel(const __bf16* __restrict__ dy,
+                                const __bf16* __restrict__ y,
+                                __bf16* __restrict__ dx, size_t total8) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    bf16x8 g = *(const bf16x8*)(dy + i * 8);
+    bf16x8 v = *(const bf16x8*)(y + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      o[e] = (float)v[e] > 0.f ? g[e] : (__bf16)0.f;
+    *(bf16x8*)(dx + i * 8) = o;
+  }
+}
+
+__global__ void add_relu_kernel(const __bf16* __restrict__ a,
+                                const __bf16* __restrict__ b,
+                                __bf16* __restrict__ y, size_t total8) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    bf16x8 va = *(const bf16x8*)(a + i * 8);
+    bf16x8 vb = *(const bf16x8*)(b + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      o[e] = (__bf16)fmaxf((float)va[e] + (float)vb[e], 0.f);
+    *(bf16x8*)(y + i * 8) = o;
+  }
+}
+
+// ============================================================== host side ==
+
+static uint32_t ew_blocks(size_t work) {
+  return std::min<size_t>((work + 255) / 256, 2048);
+}
+
+std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, long kernel, long stride,
+                                    long pad) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "maxpool: x must be NHWC");
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  uint32_t P = (H + 2 * pad - kernel) / stride + 1;
+  uint32_t Q = (W + 2 * pad - kernel) / stride + 1;
+  auto y = at::empty({(long)N, (long)C, (long)P, (long)Q}, x.options(),
+                     at::MemoryFormat::ChannelsLast);
+  auto idx = at::empty({(long)N, (long)C, (long)P, (long)Q},
+                       x.options().dtype(at::kByte), at::MemoryFormat::ChannelsLast);
+  size_t total = (size_t)N * P * Q * C;
+  maxpool_fwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
+      (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx, long H,
+                            long W, long kernel, long stride, long pad) {
+  uint32_t N = dy.size(0), C = dy.size(1), P = dy.size(2), Q = dy.size(3);
+  auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  auto dx = at::empty({(long)N, (long)C, (long)H, (long)W}, dy.options(),
+                      at::MemoryFormat::ChannelsLast);
+  size_t total = (size_t)N * H * W * C;
+  maxpool_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+      (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
+      (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+  return dx;
+}
+
+at::Tensor global_avgpool_fwd(const at::Tensor& x) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "gap: x must be NHWC");
+  uint32_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto y = at::empty({(long)N, (long)C}, x.options());
+  dim3 grid((C + 255) / 256, N);
+  gap_fwd_kernel<<<grid, 256, 0, pe_stream()>>>(
+      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(), N, C, HW);
+  return y;
+}
+
+at::Tensor global_avgpool_bwd(const at::Tensor& dy, long H, long W) {
+  uint32_t N = dy.size(0), C = dy.size(1), HW = H * W;
+  auto dx = at::empty({(long)N, (long)C, H, W}, dy.options(),
+                      at::MemoryFormat::ChannelsLast);
+  size_t total = (size_t)N * HW * C;
+  gap_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+      (const __bf16*)dy.data_ptr(), (__bf16*)dx.data_ptr(), N, C, HW);
+  return dx;
+}
+
+#define EW_CHECK(t)                                             \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, "bf16 only"); \
+  TORCH_CHECK((t).numel() % 8 == 0, "numel must be a multiple of 8")
+
+at::Tensor relu_fwd(const at::Tensor& x) {
+  EW_CHECK(x);
+  auto y = at::empty_like(x);
+  size_t t8 = x.numel() / 8;
+  relu_fwd_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
+      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(), t8);
+  return y;
+}
+
+at::Tensor relu_bwd(const at::Tensor& dy, const at::Tensor& y) {
+  EW_CHECK(dy);
+  auto dx = at::empty_like(dy);
+  size_t t8 = dy.numel() / 8;
+  relu_bwd_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
+      (const __bf16*)dy.data_ptr(), (const __bf16*)y.data_ptr(),
+      (__bf16*)dx.data_ptr(), t8);
+  return dx;
+}
+
+at::Tensor add_relu_fwd(const at::Tensor& a, const at::Tensor& b) {
+  EW_CHECK(a);
+  auto y = at::empty_like(a);
+  size_t t8 = a.numel() / 8;
+  add_relu_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
+      (const __bf16*)a.data_ptr(), (const __bf16*)b.data_ptr(),
+      (__bf16*)y.data_ptr(), t8);
+  return y;
+}
+
+}  // namespace dtmx

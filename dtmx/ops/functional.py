@@ -107,14 +107,19 @@ class _MaxPoolNHWC(torch.autograd.Function):
         y, idx = ext.maxpool_fwd(x, kernel, stride, padding)
         ctx.save_for_backward(idx)
         ctx.in_shape = x.shape
+        ctx.params = (kernel, stride, padding)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = require_ext()
         (idx,) = ctx.saved_tensors
+        k, s, p = ctx.params
         dy = dy.contiguous(memory_format=torch.channels_last)
-        return ext.maxpool_bwd(dy, idx, ctx.in_shape[2], ctx.in_shape[3]), None, None, None
+        return (
+            ext.maxpool_bwd(dy, idx, ctx.in_shape[2], ctx.in_shape[3], k, s, p),
+            None, None, None,
+        )
 
 
 def max_pool2d(x, kernel: int, stride: int, padding: int = 0):

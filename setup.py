@@ -1,0 +1,40 @@
+"""Build the dtmx HIP extension in-tree for gfx950 (MI355X):
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Produces dtmx/_C.*.so next to the package sources so the snapshot that
+travels to a GPU box carries the binary.
+"""
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "dtmx", "csrc")
+
+sources = [
+    os.path.join("dtmx", "csrc", f)
+    for f in sorted(os.listdir(CSRC))
+    if f.endswith((".cpp", ".hip"))
+]
+
+setup(
+    name="dtmx",
+    version="0.1.0",
+    packages=["dtmx"],
+    ext_modules=[
+        CUDAExtension(
+            name="dtmx._C",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)
