@@ -31,18 +31,22 @@ __device__ __forceinline__ void wait_vmcnt0() {
 // q = (n * magic) >> (32 + shift). Valid for n < 2^31, d >= 1.
 struct FastDiv {
   uint32_t d;
-  uint32_t magic;
+  uint32_t magic;   // 0 => power-of-two divisor, use shift only
   uint32_t shift;
   void init(uint32_t d_) {
     d = d_;
-    if (d == 1) { magic = 0; shift = 0; return; }
-    shift = 0;
-    while ((1u << shift) < d) ++shift;
-    uint64_t m = ((__uint128_t(1) << (32 + shift)) + d - 1) / d;
-    magic = (uint32_t)m;
+    if ((d & (d - 1)) == 0) {  // power of two (incl. 1)
+      magic = 0;
+      shift = __builtin_ctz(d);
+      return;
+    }
+    // round-up method, s = floor(log2 d): magic = ceil(2^(32+s)/d) < 2^32,
+    // exact for all n < 2^32 (Granlund-Montgomery).
+    shift = 31 - __builtin_clz(d);
+    magic = (uint32_t)(((__uint128_t(1) << (32 + shift)) + d - 1) / d);
   }
   __device__ __forceinline__ uint32_t div(uint32_t n) const {
-    if (d == 1) return n;
+    if (magic == 0) return n >> shift;
     return (uint32_t)((uint64_t(n) * magic) >> 32 >> shift);
   }
   __device__ __forceinline__ uint32_t mod(uint32_t n, uint32_t q) const {

@@ -338,14 +338,20 @@ static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
 
 // -- plain GEMM entry points (FullyConnected; reference fully_connected-inl.h)
 
+static at::Tensor pad_cols8(const at::Tensor& t) {
+  // zero-pad the last dim to a multiple of 8 (k-piece granularity)
+  long k = t.size(-1);
+  if (k % 8 == 0) return t.contiguous();
+  return at::constant_pad_nd(t, {0, 8 - (k % 8)}, 0.0).contiguous();
+}
+
 at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
                       const c10::optional<at::Tensor>& bias) {
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(w);
-  auto xc = x.contiguous();
-  auto wc = w.contiguous();
+  auto xc = pad_cols8(x);
+  auto wc = pad_cols8(w);
   uint32_t M = xc.size(0), K = xc.size(1), N = wc.size(0);
-  TORCH_CHECK(K % 8 == 0, "linear: in_features must be a multiple of 8");
   auto y = at::empty({(long)M, (long)N}, x.options());
   at::Tensor bias_f;
   const float* bp = nullptr;
@@ -379,15 +385,14 @@ static at::Tensor transpose2d(const at::Tensor& in, uint32_t out_ld_pad = 0) {
 at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
   CHECK_BF16_CUDA(dy);
   uint32_t M = dy.size(0), N = dy.size(1), K = w.size(1);
-  TORCH_CHECK(N % 8 == 0, "linear dgrad: out_features must be a multiple of 8");
-  auto dyc = dy.contiguous();
-  auto wt = transpose2d(w.contiguous(), ((N + 7) / 8) * 8);  // [K][Npad]
+  uint32_t Npad = ((N + 7) / 8) * 8;
+  auto dyc = pad_cols8(dy);                      // [M][Npad]
+  auto wt = transpose2d(pad_cols8(w), Npad);     // [Kpad8][Npad] (rows>N zero)
   auto dx = at::empty({(long)M, (long)K}, dy.options());
-  uint32_t Npad = wt.size(1);
-  DenseP pa{(const __bf16*)dyc.data_ptr(), zero_page(dy), M, N, N};
-  DenseP pb{(const __bf16*)wt.data_ptr(), zero_page(dy), K, N, Npad};
+  DenseP pa{(const __bf16*)dyc.data_ptr(), zero_page(dy), M, Npad, Npad};
+  DenseP pb{(const __bf16*)wt.data_ptr(), zero_page(dy), K, Npad, Npad};
   EpiBF16 epi{(__bf16*)dx.data_ptr(), nullptr, M, K, 0};
-  launch_gemm(pa, pb, epi, M, K, N);
+  launch_gemm(pa, pb, epi, M, K, Npad);
   return dx;
 }
 
@@ -395,8 +400,8 @@ at::Tensor linear_wgrad(const at::Tensor& dy, const at::Tensor& x) {
   CHECK_BF16_CUDA(dy);
   uint32_t M = dy.size(0), N = dy.size(1), K = x.size(1);
   uint32_t Mpad = ((M + 7) / 8) * 8;
-  auto dyt = transpose2d(dy.contiguous(), Mpad);  // [N][Mpad]
-  auto xt = transpose2d(x.contiguous(), Mpad);    // [K][Mpad]
+  auto dyt = transpose2d(pad_cols8(dy), Mpad);  // [Npad][Mpad]
+  auto xt = transpose2d(pad_cols8(x), Mpad);    // [Kpad][Mpad]
   auto dw = at::empty({(long)N, (long)K}, dy.options());
   DenseP pa{(const __bf16*)dyt.data_ptr(), zero_page(dy), N, Mpad, Mpad};
   DenseP pb{(const __bf16*)xt.data_ptr(), zero_page(dy), K, Mpad, Mpad};
@@ -466,14 +471,22 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy must be NHWC");
   uint32_t N = dy.size(0), Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
   uint32_t C = w.size(1), R = w.size(2), S = w.size(3);
-  TORCH_CHECK(Ko % 8 == 0, "conv dgrad: out channels must be a multiple of 8");
-  // B operand: W^T in (C,R,S,Ko) contiguous layout
-  auto wt = w.permute({1, 2, 3, 0}).contiguous();  // logical (C,R,S,K), dense
+  // pad the out-channel (k-piece) dim to a multiple of 8 when needed
+  at::Tensor dyk = dy.permute({0, 2, 3, 1});        // NHWC view (contiguous)
+  at::Tensor wtk = w.permute({1, 2, 3, 0});          // (C,R,S,K)
+  if (Ko % 8) {
+    long padk = 8 - (Ko % 8);
+    dyk = at::constant_pad_nd(dyk, {0, padk}, 0.0);
+    wtk = at::constant_pad_nd(wtk, {0, padk}, 0.0);
+    Ko += padk;
+  }
+  auto dyc = dyk.contiguous();
+  auto wt = wtk.contiguous();  // W^T in (C,R,S,Ko) dense layout
   auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
                       at::MemoryFormat::ChannelsLast);
   uint32_t M = N * H * W_, Ktot = R * S * Ko;
   ConvDgradA pa;
-  pa.dy = (const __bf16*)dy.data_ptr();
+  pa.dy = (const __bf16*)dyc.data_ptr();
   pa.zero = zero_page(dy);
   pa.M = M; pa.Ktot = Ktot; pa.Ko = Ko; pa.H = H; pa.W = W_; pa.P = P; pa.Q = Q;
   pa.S = S; pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
@@ -498,8 +511,9 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
   uint32_t Mpad = ((M + 7) / 8) * 8;
   uint32_t RSC = R * S * C;
 
-  // dy^T : [NPQ][Ko] -> [Ko][Mpad]
-  auto dyt = transpose2d(dy.reshape({(long)M, (long)Ko}), Mpad);
+  // dy^T : [NPQ][Ko] -> [Ko(+pad)][Mpad]
+  auto dyt = transpose2d(
+      pad_cols8(dy.permute({0, 2, 3, 1}).reshape({(long)M, (long)Ko})), Mpad);
 
   // gathered im2col^T: [RSC][Mpad] (one transpose-gather launch per (r,s))
   at::Tensor xt;
