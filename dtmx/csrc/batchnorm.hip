@@ -1,12 +1,12 @@
 // NHWC BatchNorm fwd/bwd for gfx950 (reference src/operator/nn/batch_norm.cu:
-// 208-360 — redesigned for NHWC/bf16: channel reductions are coalesced column
-// sums with fp32 atomically-merged partials; apply passes are vectorized
-// bf16x8 elementwise with optional fused ReLU).
+// 208-360 — redesigned for NHWC/bf16: channel reductions are bf16x8-vectorized
+// column sums with fp32 atomically-merged partials; apply passes are
+// vectorized elementwise with optional fused ReLU; all hot-loop indexing is
+// 32-bit with FastDiv (64-bit div/mod on the elementwise path measured ~5x).
 //
-// Pass structure (memory-bound; HBM-optimal would be 2 passes — fusing the
-// stats into the producing conv epilogue is a later-round optimization):
-//   fwd train: stats (x)  -> finalize (tiny) -> apply (x -> y)
-//   bwd:       grads-stats (x,dy,y) -> finalize (tiny) -> apply-dx
+// Pass structure (memory-bound):
+//   fwd train: stats (x) -> finalize (tiny) -> apply (x -> y)
+//   bwd:       grad-stats (x,dy[,y]) -> finalize (tiny) -> apply-dx
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -16,22 +16,61 @@ namespace dtmx {
 
 static hipStream_t bn_stream() { return at::hip::getCurrentHIPStream().stream(); }
 
-// ---- forward stats: partial sum / sumsq per channel ----------------------
+// tree-reduce 8 per-thread floats across the threads sharing a channel
+// vector (they sit at stride cvecs in the block), then one atomicAdd per
+// channel from the surviving thread.
+__device__ __forceinline__ void block_col_reduce(float* red, float (&v)[8],
+                                                 uint32_t cv, uint32_t cvecs,
+                                                 float* out) {
+  const uint32_t t = threadIdx.x;
+  __syncthreads();
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[t * 8 + e] = v[e];
+  __syncthreads();
+  for (uint32_t off = 128; off >= cvecs; off >>= 1) {
+    if (t < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (t < cvecs) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) atomicAdd(&out[t * 8 + e], red[t * 8 + e]);
+  }
+}
+
+// ---- forward stats: partial sum / sumsq per channel, vectorized ----------
+// thread t covers channel-vector cv = t % cvecs (8 channels), row stripe
+// r = r0 + t/cvecs, stepping by 256/cvecs. Requires cvecs <= 256 divisor of
+// 256 (C is a power-of-two multiple of 8 in practice); general C uses the
+// strided variant below.
 __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ psum,
                                 float* __restrict__ psumsq, uint32_t rows,
-                                uint32_t C, uint32_t rows_per_block) {
-  const uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;  // one channel / thread
-  if (c >= C) return;
-  const uint32_t r0 = blockIdx.y * rows_per_block;
-  const uint32_t r1 = min(r0 + rows_per_block, rows);
-  float s = 0.f, ss = 0.f;
-  for (uint32_t r = r0; r < r1; ++r) {
-    float v = (float)x[(size_t)r * C + c];
-    s += v;
-    ss += v * v;
+                                uint32_t cvecs, uint32_t rows_per_block) {
+  const uint32_t t = threadIdx.x;
+  const uint32_t cv = t % cvecs;
+  const uint32_t rstep = blockDim.x / cvecs;
+  const uint32_t r0 = blockIdx.y * rows_per_block + t / cvecs;
+  const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
+  const uint32_t C = cvecs * 8;
+  float s[8] = {}, ss[8] = {};
+  for (uint32_t r = r0; r < r1; r += rstep) {
+    bf16x8 v = *(const bf16x8*)(x + (size_t)r * C + cv * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float f = (float)v[e];
+      s[e] += f;
+      ss[e] += f * f;
+    }
   }
-  atomicAdd(&psum[c], s);
-  atomicAdd(&psumsq[c], ss);
+  // intra-block tree reduction over the threads sharing a channel vector
+  // (stride cvecs), then ONE atomicAdd per channel per block — without this
+  // the atomic contention on C words serializes the whole kernel (measured
+  // 30x slower at C=64).
+  __shared__ float red[256 * 8];
+  block_col_reduce(red, s, cv, cvecs, psum);
+  block_col_reduce(red, ss, cv, cvecs, psumsq);
 }
 
 // ---- finalize: mean/invstd, running stats, fused scale/shift -------------
@@ -53,8 +92,6 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,
   float invstd = rsqrtf(var + eps);
   save_mean[c] = mean;
   save_invstd[c] = invstd;
-  // torch-style running update with unbiased var; `momentum` is the mxnet
-  // moving fraction (moving = m*moving + (1-m)*batch)
   float unbiased = count > 1 ? var * count / (count - 1) : var;
   running_mean[c] = running_mean[c] * momentum + mean * (1.f - momentum);
   running_var[c] = running_var[c] * momentum + unbiased * (1.f - momentum);
@@ -78,59 +115,71 @@ __global__ void bn_infer_prep_kernel(const __bf16* __restrict__ gamma,
   shift[c] = (float)beta[c] - running_mean[c] * g * invstd;
 }
 
-// ---- apply: y = x*scale + shift (+relu), vectorized 8 --------------------
+// ---- apply: y = x*scale + shift (+relu), bf16x8, 32-bit indexing ---------
 __global__ void bn_apply_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
                                 const float* __restrict__ scale,
-                                const float* __restrict__ shift, size_t total,
-                                uint32_t C, int relu) {
-  size_t i8 = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  const size_t stride = (size_t)gridDim.x * blockDim.x * 8;
-  for (; i8 < total; i8 += stride) {
-    bf16x8 v = *(const bf16x8*)(x + i8);
-    uint32_t c0 = (uint32_t)(i8 % C);
+                                const float* __restrict__ shift, uint32_t total8,
+                                FastDiv dcv, int relu) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    bf16x8 v = *(const bf16x8*)(x + (size_t)i * 8);
+    uint32_t q = dcv.div(i);
+    uint32_t c0 = dcv.mod(i, q) * 8;
     bf16x8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      uint32_t c = c0 + e;  // C % 8 == 0 so the vector never crosses a row
-      float r = (float)v[e] * scale[c] + shift[c];
+      float r = (float)v[e] * scale[c0 + e] + shift[c0 + e];
       if (relu) r = fmaxf(r, 0.f);
       o[e] = (__bf16)r;
     }
-    *(bf16x8*)(y + i8) = o;
+    *(bf16x8*)(y + (size_t)i * 8) = o;
   }
 }
 
-// ---- backward stats: per-channel sum(dy), sum(dy*xhat) -------------------
+// ---- backward stats: per-channel sum(dy), sum(dy*xhat), vectorized -------
 __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
                                     const __bf16* __restrict__ dy,
-                                    const __bf16* __restrict__ y,  // for relu mask
+                                    const __bf16* __restrict__ y,  // relu mask
                                     const float* __restrict__ save_mean,
                                     const float* __restrict__ save_invstd,
                                     float* __restrict__ pdb, float* __restrict__ pdg,
-                                    uint32_t rows, uint32_t C,
+                                    uint32_t rows, uint32_t cvecs,
                                     uint32_t rows_per_block, int relu) {
-  const uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const uint32_t r0 = blockIdx.y * rows_per_block;
-  const uint32_t r1 = min(r0 + rows_per_block, rows);
-  const float mean = save_mean[c], invstd = save_invstd[c];
-  float db = 0.f, dg = 0.f;
-  for (uint32_t r = r0; r < r1; ++r) {
-    size_t i = (size_t)r * C + c;
-    float g = (float)dy[i];
-    if (relu && (float)y[i] <= 0.f) g = 0.f;
-    float xh = ((float)x[i] - mean) * invstd;
-    db += g;
-    dg += g * xh;
+  const uint32_t t = threadIdx.x;
+  const uint32_t cv = t % cvecs;
+  const uint32_t rstep = blockDim.x / cvecs;
+  const uint32_t r0 = blockIdx.y * rows_per_block + t / cvecs;
+  const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
+  const uint32_t C = cvecs * 8;
+  float mean[8], invstd[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    mean[e] = save_mean[cv * 8 + e];
+    invstd[e] = save_invstd[cv * 8 + e];
   }
-  atomicAdd(&pdb[c], db);
-  atomicAdd(&pdg[c], dg);
+  float db[8] = {}, dg[8] = {};
+  for (uint32_t r = r0; r < r1; r += rstep) {
+    size_t off = (size_t)r * C + cv * 8;
+    bf16x8 xv = *(const bf16x8*)(x + off);
+    bf16x8 gv = *(const bf16x8*)(dy + off);
+    bf16x8 yv;
+    if (relu) yv = *(const bf16x8*)(y + off);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float g = (float)gv[e];
+      if (relu && (float)yv[e] <= 0.f) g = 0.f;
+      db[e] += g;
+      dg[e] += g * ((float)xv[e] - mean[e]) * invstd[e];
+    }
+  }
+  __shared__ float red[256 * 8];
+  block_col_reduce(red, db, cv, cvecs, pdb);
+  block_col_reduce(red, dg, cv, cvecs, pdg);
 }
 
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ pdb,
                                        const float* __restrict__ pdg,
-                                       const __bf16* __restrict__ gamma,
-                                       const float* __restrict__ save_invstd,
                                        __bf16* __restrict__ dgamma,
                                        __bf16* __restrict__ dbeta, uint32_t C) {
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
@@ -148,43 +197,51 @@ __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ x,
                                  const __bf16* __restrict__ gamma,
                                  const float* __restrict__ pdb,
                                  const float* __restrict__ pdg,
-                                 __bf16* __restrict__ dx, size_t total,
-                                 uint32_t C, float inv_count, int relu) {
-  size_t i8 = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  const size_t stride = (size_t)gridDim.x * blockDim.x * 8;
-  for (; i8 < total; i8 += stride) {
-    bf16x8 xv = *(const bf16x8*)(x + i8);
-    bf16x8 gv = *(const bf16x8*)(dy + i8);
+                                 __bf16* __restrict__ dx, uint32_t total8,
+                                 FastDiv dcv, float inv_count, int relu) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    size_t off = (size_t)i * 8;
+    bf16x8 xv = *(const bf16x8*)(x + off);
+    bf16x8 gv = *(const bf16x8*)(dy + off);
     bf16x8 yv;
-    if (relu) yv = *(const bf16x8*)(y + i8);
-    uint32_t c0 = (uint32_t)(i8 % C);
+    if (relu) yv = *(const bf16x8*)(y + off);
+    uint32_t q = dcv.div(i);
+    uint32_t c0 = dcv.mod(i, q) * 8;
     bf16x8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       uint32_t c = c0 + e;
       float g = (float)gv[e];
       if (relu && (float)yv[e] <= 0.f) g = 0.f;
-      float mean = save_mean[c], invstd = save_invstd[c];
-      float xh = ((float)xv[e] - mean) * invstd;
-      float r = (float)gamma[c] * invstd *
-                (g - pdb[c] * inv_count - xh * pdg[c] * inv_count);
-      o[e] = (__bf16)r;
+      float invstd = save_invstd[c];
+      float xh = ((float)xv[e] - save_mean[c]) * invstd;
+      o[e] = (__bf16)((float)gamma[c] * invstd *
+                      (g - pdb[c] * inv_count - xh * pdg[c] * inv_count));
     }
-    *(bf16x8*)(dx + i8) = o;
+    *(bf16x8*)(dx + off) = o;
   }
 }
 
 // ============================================================== host side ==
 
-static void bn_grid(uint32_t rows, uint32_t C, dim3& grid, dim3& block,
+// vectorized stats geometry: cvecs must divide 256 (C = 8*cvecs); channel
+// counts in the model zoo are powers of two in [64, 2048].
+static bool stats_vec_ok(uint32_t C) {
+  uint32_t cv = C / 8;
+  return C % 8 == 0 && cv <= 256 && 256 % cv == 0;
+}
+
+static void bn_grid(uint32_t rows, uint32_t cvecs, dim3& grid,
                     uint32_t& rows_per_block) {
-  block = dim3(256);
-  uint32_t cb = (C + 255) / 256;
-  uint32_t target_blocks = 2048;
-  uint32_t rb = std::max<uint32_t>(1, target_blocks / cb);
+  uint32_t rstep = 256 / cvecs;
+  // >=16 strip iterations per thread so the per-block reduce+atomic tail
+  // amortizes; cap at 1024 blocks (4/CU) for bandwidth saturation.
+  uint32_t rb = std::min<uint32_t>(1024, std::max<uint32_t>(1, rows / (rstep * 16)));
   rows_per_block = (rows + rb - 1) / rb;
   rb = (rows + rows_per_block - 1) / rows_per_block;
-  grid = dim3(cb, rb);
+  grid = dim3(1, rb);
 }
 
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamma,
@@ -193,32 +250,34 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
                                      double eps, bool fuse_relu) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8");
-  uint32_t rows = N * H * W;
+  TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
+  uint32_t rows = N * H * W, cvecs = C / 8;
   auto opt_f = x.options().dtype(at::kFloat);
   auto psum = at::zeros({(long)C}, opt_f), psumsq = at::zeros({(long)C}, opt_f);
   auto save_mean = at::empty({(long)C}, opt_f), save_invstd = at::empty({(long)C}, opt_f);
   auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
   auto y = at::empty_like(x);
-  dim3 grid, block;
+  dim3 grid;
   uint32_t rpb;
-  bn_grid(rows, C, grid, block, rpb);
+  bn_grid(rows, cvecs, grid, rpb);
   auto s = bn_stream();
-  bn_stats_kernel<<<grid, block, 0, s>>>((const __bf16*)x.data_ptr(),
-                                         psum.data_ptr<float>(),
-                                         psumsq.data_ptr<float>(), rows, C, rpb);
+  bn_stats_kernel<<<grid, 256, 0, s>>>((const __bf16*)x.data_ptr(),
+                                       psum.data_ptr<float>(),
+                                       psumsq.data_ptr<float>(), rows, cvecs, rpb);
   bn_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
       psum.data_ptr<float>(), psumsq.data_ptr<float>(),
       (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
       scale.data_ptr<float>(), shift.data_ptr<float>(), C, rows, momentum, eps);
-  size_t total = (size_t)rows * C;
-  uint32_t blocks = std::min<size_t>((total / 8 + 255) / 256, 2048);
+  uint32_t total8 = rows * cvecs;
+  FastDiv dcv;
+  dcv.init(cvecs);
+  uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
   bn_apply_kernel<<<blocks, 256, 0, s>>>((const __bf16*)x.data_ptr(),
                                          (__bf16*)y.data_ptr(),
                                          scale.data_ptr<float>(),
-                                         shift.data_ptr<float>(), total, C,
+                                         shift.data_ptr<float>(), total8, dcv,
                                          fuse_relu ? 1 : 0);
   return {y, save_mean, save_invstd};
 }
@@ -228,7 +287,8 @@ at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
                         const at::Tensor& running_var, double eps, bool fuse_relu) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  uint32_t rows = N * H * W;
+  TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8");
+  uint32_t rows = N * H * W, cvecs = C / 8;
   auto opt_f = x.options().dtype(at::kFloat);
   auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
   auto y = at::empty_like(x);
@@ -237,12 +297,14 @@ at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
       (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       scale.data_ptr<float>(), shift.data_ptr<float>(), C, eps);
-  size_t total = (size_t)rows * C;
-  uint32_t blocks = std::min<size_t>((total / 8 + 255) / 256, 2048);
+  uint32_t total8 = rows * cvecs;
+  FastDiv dcv;
+  dcv.init(cvecs);
+  uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
   bn_apply_kernel<<<blocks, 256, 0, s>>>((const __bf16*)x.data_ptr(),
                                          (__bf16*)y.data_ptr(),
                                          scale.data_ptr<float>(),
-                                         shift.data_ptr<float>(), total, C,
+                                         shift.data_ptr<float>(), total8, dcv,
                                          fuse_relu ? 1 : 0);
   return y;
 }
@@ -252,33 +314,35 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
                                const at::Tensor& save_invstd, bool fuse_relu,
                                const at::Tensor& y) {
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  uint32_t rows = N * H * W;
+  TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
+  uint32_t rows = N * H * W, cvecs = C / 8;
   auto opt_f = x.options().dtype(at::kFloat);
   auto pdb = at::zeros({(long)C}, opt_f), pdg = at::zeros({(long)C}, opt_f);
   auto dgamma = at::empty({(long)C}, x.options());
   auto dbeta = at::empty({(long)C}, x.options());
   auto dx = at::empty_like(x);
-  dim3 grid, block;
+  dim3 grid;
   uint32_t rpb;
-  bn_grid(rows, C, grid, block, rpb);
+  bn_grid(rows, cvecs, grid, rpb);
   auto s = bn_stream();
-  bn_bwd_stats_kernel<<<grid, block, 0, s>>>(
+  bn_bwd_stats_kernel<<<grid, 256, 0, s>>>(
       (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
-      pdg.data_ptr<float>(), rows, C, rpb, fuse_relu ? 1 : 0);
+      pdg.data_ptr<float>(), rows, cvecs, rpb, fuse_relu ? 1 : 0);
   bn_bwd_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
-      pdb.data_ptr<float>(), pdg.data_ptr<float>(),
-      (const __bf16*)gamma.data_ptr(), save_invstd.data_ptr<float>(),
-      (__bf16*)dgamma.data_ptr(), (__bf16*)dbeta.data_ptr(), C);
-  size_t total = (size_t)rows * C;
-  uint32_t blocks = std::min<size_t>((total / 8 + 255) / 256, 2048);
+      pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
+      (__bf16*)dbeta.data_ptr(), C);
+  uint32_t total8 = rows * cvecs;
+  FastDiv dcv;
+  dcv.init(cvecs);
+  uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
   bn_bwd_dx_kernel<<<blocks, 256, 0, s>>>(
       (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), (const __bf16*)gamma.data_ptr(),
       pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dx.data_ptr(),
-      total, C, 1.f / rows, fuse_relu ? 1 : 0);
+      total8, dcv, 1.f / rows, fuse_relu ? 1 : 0);
   return {dx, dgamma, dbeta};
 }
 

@@ -140,8 +140,13 @@ struct EpiAtomicF32 {  // split-K partial accumulation (conv wgrad)
 
 template <class PA, class PB, class EPI>
 __launch_bounds__(256, 2) __global__
-void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles, uint32_t tiles_n,
-                    uint32_t kt0) {
+void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
+                    uint32_t tiles_n, uint32_t kt_per_slice) {
+  // split-K: blockIdx.y selects a K-slice (fp32-atomic epilogue makes the
+  // slices order-independent); single launch fills the chip.
+  const uint32_t kt0 = blockIdx.y * kt_per_slice;
+  const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
+  if (kt0 >= ktiles_total) return;
   __shared__ __bf16 smem[2][2][128 * 64];  // [buf][A|B][row*64+col] 64 KiB
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
@@ -318,22 +323,11 @@ static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
                         uint32_t N, uint32_t K, uint32_t splitk = 1) {
   uint32_t tiles_m = ceil_div(M, 128), tiles_n = ceil_div(N, 128);
   uint32_t ktiles_total = ceil_div(K, 64);
+  splitk = std::min(splitk, ktiles_total);
   uint32_t kt_per = ceil_div(ktiles_total, splitk);
-  dim3 grid(tiles_m * tiles_n, splitk);
-  if (splitk == 1) {
-    gemm_tn_kernel<PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
-        pa, pb, epi, ktiles_total, tiles_n, 0);
-  } else {
-    // launch one grid per slice on the same stream (kt0 varies); the
-    // fp32-atomic epilogue makes slices order-independent.
-    for (uint32_t s = 0; s < splitk; ++s) {
-      uint32_t kt0 = s * kt_per;
-      if (kt0 >= ktiles_total) break;
-      uint32_t kt = std::min(kt_per, ktiles_total - kt0);
-      gemm_tn_kernel<PA, PB, EPI><<<dim3(tiles_m * tiles_n), 256, 0, cur_stream()>>>(
-          pa, pb, epi, kt, tiles_n, kt0);
-    }
-  }
+  dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
+  gemm_tn_kernel<PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+      pa, pb, epi, ktiles_total, tiles_n, kt_per);
 }
 
 // -- plain GEMM entry points (FullyConnected; reference fully_connected-inl.h)
@@ -546,10 +540,11 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
     xt = transpose2d(col, Mpad).narrow(0, 0, RSC).contiguous();
   }
 
-  // choose split-K so the grid fills the chip (~>=512 blocks)
+  // choose split-K so the grid fills the chip (~1024 blocks; 2 blocks/CU on
+  // 256 CUs plus headroom for tail effects)
   uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, 128);
   uint32_t ktiles = ceil_div(Mpad, 64);
-  uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 512 / std::max(1u, tiles_mn)));
+  uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
 
   auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
   DenseP pa{(const __bf16*)dyt.data_ptr(), zero_page(x), Ko, Mpad, Mpad};
