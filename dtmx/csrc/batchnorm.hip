@@ -17,11 +17,12 @@ namespace dtmx {
 static hipStream_t bn_stream() { return at::hip::getCurrentHIPStream().stream(); }
 
 // tree-reduce 8 per-thread floats across the threads sharing a channel
-// vector (they sit at stride cvecs in the block), then one atomicAdd per
-// channel from the surviving thread.
+// vector (stride cvecs in the block), then write the block's partial row to
+// a [gridDim.y][C] slab. No atomics: same-address fp32 atomicAdd serializes
+// across blocks (measured ~0.2us PER BLOCK -> stats time linear in grid).
 __device__ __forceinline__ void block_col_reduce(float* red, float (&v)[8],
                                                  uint32_t cv, uint32_t cvecs,
-                                                 float* out) {
+                                                 float* slab_row) {
   const uint32_t t = threadIdx.x;
   __syncthreads();
 #pragma unroll
@@ -36,7 +37,7 @@ __device__ __forceinline__ void block_col_reduce(float* red, float (&v)[8],
   }
   if (t < cvecs) {
 #pragma unroll
-    for (int e = 0; e < 8; ++e) atomicAdd(&out[t * 8 + e], red[t * 8 + e]);
+    for (int e = 0; e < 8; ++e) slab_row[t * 8 + e] = red[t * 8 + e];
   }
 }
 
@@ -55,22 +56,92 @@ __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict_
   const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
   const uint32_t C = cvecs * 8;
   float s[8] = {}, ss[8] = {};
-  for (uint32_t r = r0; r < r1; r += rstep) {
-    bf16x8 v = *(const bf16x8*)(x + (size_t)r * C + cv * 8);
+  // 4x unrolled so four 16-B loads are in flight per wave (a single-buffer
+  // loop compiles to load -> vmcnt(0) -> use and runs HBM-latency-bound:
+  // measured ~20x off the bandwidth roofline).
+  auto accum = [&](bf16x8 v) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float f = (float)v[e];
       s[e] += f;
       ss[e] += f * f;
     }
+  };
+  uint32_t r = r0;
+  for (; r + 3 * rstep < r1; r += 4 * rstep) {
+    bf16x8 v0 = *(const bf16x8*)(x + (size_t)r * C + cv * 8);
+    bf16x8 v1 = *(const bf16x8*)(x + (size_t)(r + rstep) * C + cv * 8);
+    bf16x8 v2 = *(const bf16x8*)(x + (size_t)(r + 2 * rstep) * C + cv * 8);
+    bf16x8 v3 = *(const bf16x8*)(x + (size_t)(r + 3 * rstep) * C + cv * 8);
+    accum(v0); accum(v1); accum(v2); accum(v3);
   }
-  // intra-block tree reduction over the threads sharing a channel vector
-  // (stride cvecs), then ONE atomicAdd per channel per block — without this
-  // the atomic contention on C words serializes the whole kernel (measured
-  // 30x slower at C=64).
+  for (; r < r1; r += rstep)
+    accum(*(const bf16x8*)(x + (size_t)r * C + cv * 8));
+  // intra-block tree reduction, then a per-block slab row (see
+  // block_col_reduce).
   __shared__ float red[256 * 8];
-  block_col_reduce(red, s, cv, cvecs, psum);
-  block_col_reduce(red, ss, cv, cvecs, psumsq);
+  block_col_reduce(red, s, cv, cvecs, psum + (size_t)blockIdx.y * C);
+  block_col_reduce(red, ss, cv, cvecs, psumsq + (size_t)blockIdx.y * C);
+}
+
+// ---- parallel slab reduction: [nslabs][C] x2 -> [C] x2 -------------------
+// block covers ncv channel-vectors (<=8) x (256/ncv) slab lanes; lane-strided
+// accumulate then LDS tree. Replaces both the per-channel serial loop
+// (measured 190us/call at nslabs=2048) and cross-block atomics.
+__global__ void slab_reduce2_kernel(const float* __restrict__ a,
+                                    const float* __restrict__ b,
+                                    float* __restrict__ outa,
+                                    float* __restrict__ outb, uint32_t C,
+                                    uint32_t nslabs) {
+  const uint32_t cvecs = C / 8;
+  const uint32_t ncv = min(cvecs, 8u);
+  const uint32_t lanes = blockDim.x / ncv;
+  const uint32_t t = threadIdx.x;
+  const uint32_t cv = blockIdx.x * ncv + t % ncv;
+  const uint32_t lane = t / ncv;
+  float sa[8] = {}, sb[8] = {};
+  if (cv < cvecs) {
+    for (uint32_t sl = lane; sl < nslabs; sl += lanes) {
+      const float* pa = a + (size_t)sl * C + cv * 8;
+      const float* pb = b + (size_t)sl * C + cv * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        sa[e] += pa[e];
+        sb[e] += pb[e];
+      }
+    }
+  }
+  __shared__ float red[256 * 8];
+  // tree over the slab lanes (threads sharing cv sit at stride ncv)
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[t * 8 + e] = sa[e];
+  __syncthreads();
+  for (uint32_t off = 128; off >= ncv; off >>= 1) {
+    if (t < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (t < ncv && cv < cvecs) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) outa[cv * 8 + e] = red[t * 8 + e];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[t * 8 + e] = sb[e];
+  __syncthreads();
+  for (uint32_t off = 128; off >= ncv; off >>= 1) {
+    if (t < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (t < ncv && cv < cvecs) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) outb[cv * 8 + e] = red[t * 8 + e];
+  }
 }
 
 // ---- finalize: mean/invstd, running stats, fused scale/shift -------------
@@ -84,11 +155,17 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,
                                    float* __restrict__ save_invstd,
                                    float* __restrict__ scale,
                                    float* __restrict__ shift, uint32_t C,
-                                   uint32_t count, float momentum, float eps) {
+                                   uint32_t nslabs, uint32_t count,
+                                   float momentum, float eps) {
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = psum[c] / count;
-  float var = fmaxf(psumsq[c] / count - mean * mean, 0.f);
+  float fsum = 0.f, fsumsq = 0.f;
+  for (uint32_t b = 0; b < nslabs; ++b) {
+    fsum += psum[(size_t)b * C + c];
+    fsumsq += psumsq[(size_t)b * C + c];
+  }
+  float mean = fsum / count;
+  float var = fmaxf(fsumsq / count - mean * mean, 0.f);
   float invstd = rsqrtf(var + eps);
   save_mean[c] = mean;
   save_invstd[c] = invstd;
@@ -159,12 +236,7 @@ __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
     invstd[e] = save_invstd[cv * 8 + e];
   }
   float db[8] = {}, dg[8] = {};
-  for (uint32_t r = r0; r < r1; r += rstep) {
-    size_t off = (size_t)r * C + cv * 8;
-    bf16x8 xv = *(const bf16x8*)(x + off);
-    bf16x8 gv = *(const bf16x8*)(dy + off);
-    bf16x8 yv;
-    if (relu) yv = *(const bf16x8*)(y + off);
+  auto accum = [&](bf16x8 xv, bf16x8 gv, bf16x8 yv) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float g = (float)gv[e];
@@ -172,20 +244,46 @@ __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
       db[e] += g;
       dg[e] += g * ((float)xv[e] - mean[e]) * invstd[e];
     }
+  };
+  bf16x8 zed = {};
+  uint32_t r = r0;
+  for (; r + rstep < r1; r += 2 * rstep) {  // 2x unroll: 4-6 loads in flight
+    size_t o0 = (size_t)r * C + cv * 8, o1 = (size_t)(r + rstep) * C + cv * 8;
+    bf16x8 x0 = *(const bf16x8*)(x + o0), x1 = *(const bf16x8*)(x + o1);
+    bf16x8 g0 = *(const bf16x8*)(dy + o0), g1 = *(const bf16x8*)(dy + o1);
+    bf16x8 y0 = relu ? *(const bf16x8*)(y + o0) : zed;
+    bf16x8 y1 = relu ? *(const bf16x8*)(y + o1) : zed;
+    accum(x0, g0, y0);
+    accum(x1, g1, y1);
+  }
+  for (; r < r1; r += rstep) {
+    size_t o0 = (size_t)r * C + cv * 8;
+    accum(*(const bf16x8*)(x + o0), *(const bf16x8*)(dy + o0),
+          relu ? *(const bf16x8*)(y + o0) : zed);
   }
   __shared__ float red[256 * 8];
-  block_col_reduce(red, db, cv, cvecs, pdb);
-  block_col_reduce(red, dg, cv, cvecs, pdg);
+  block_col_reduce(red, db, cv, cvecs, pdb + (size_t)blockIdx.y * C);
+  block_col_reduce(red, dg, cv, cvecs, pdg + (size_t)blockIdx.y * C);
 }
 
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ pdb,
                                        const float* __restrict__ pdg,
                                        __bf16* __restrict__ dgamma,
-                                       __bf16* __restrict__ dbeta, uint32_t C) {
+                                       __bf16* __restrict__ dbeta,
+                                       float* __restrict__ tdb,
+                                       float* __restrict__ tdg, uint32_t C,
+                                       uint32_t nslabs) {
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  dgamma[c] = (__bf16)pdg[c];
-  dbeta[c] = (__bf16)pdb[c];
+  float db = 0.f, dg = 0.f;
+  for (uint32_t b = 0; b < nslabs; ++b) {
+    db += pdb[(size_t)b * C + c];
+    dg += pdg[(size_t)b * C + c];
+  }
+  dgamma[c] = (__bf16)dg;
+  dbeta[c] = (__bf16)db;
+  tdb[c] = db;
+  tdg[c] = dg;
 }
 
 // dx = gamma*invstd * (dy - dbeta/M - xhat * dgamma/M)
@@ -237,8 +335,8 @@ static void bn_grid(uint32_t rows, uint32_t cvecs, dim3& grid,
                     uint32_t& rows_per_block) {
   uint32_t rstep = 256 / cvecs;
   // >=16 strip iterations per thread so the per-block reduce+atomic tail
-  // amortizes; cap at 1024 blocks (4/CU) for bandwidth saturation.
-  uint32_t rb = std::min<uint32_t>(1024, std::max<uint32_t>(1, rows / (rstep * 16)));
+  // amortizes; cap at 2048 blocks (8/CU) for latency hiding + BW saturation.
+  uint32_t rb = std::min<uint32_t>(2048, std::max<uint32_t>(1, rows / (rstep * 16)));
   rows_per_block = (rows + rb - 1) / rb;
   rb = (rows + rows_per_block - 1) / rows_per_block;
   grid = dim3(1, rb);
@@ -253,23 +351,31 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
   TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
   uint32_t rows = N * H * W, cvecs = C / 8;
   auto opt_f = x.options().dtype(at::kFloat);
-  auto psum = at::zeros({(long)C}, opt_f), psumsq = at::zeros({(long)C}, opt_f);
-  auto save_mean = at::empty({(long)C}, opt_f), save_invstd = at::empty({(long)C}, opt_f);
-  auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
-  auto y = at::empty_like(x);
   dim3 grid;
   uint32_t rpb;
   bn_grid(rows, cvecs, grid, rpb);
+  uint32_t nslabs = grid.y;
+  auto psum = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto psumsq = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto save_mean = at::empty({(long)C}, opt_f), save_invstd = at::empty({(long)C}, opt_f);
+  auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
+  auto y = at::empty_like(x);
   auto s = bn_stream();
   bn_stats_kernel<<<grid, 256, 0, s>>>((const __bf16*)x.data_ptr(),
                                        psum.data_ptr<float>(),
                                        psumsq.data_ptr<float>(), rows, cvecs, rpb);
+  auto tsum = at::empty({(long)C}, opt_f), tsumsq = at::empty({(long)C}, opt_f);
+  uint32_t ncv = std::min(cvecs, 8u);
+  slab_reduce2_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+      psum.data_ptr<float>(), psumsq.data_ptr<float>(), tsum.data_ptr<float>(),
+      tsumsq.data_ptr<float>(), C, nslabs);
   bn_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
-      psum.data_ptr<float>(), psumsq.data_ptr<float>(),
+      tsum.data_ptr<float>(), tsumsq.data_ptr<float>(),
       (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(), C, rows, momentum, eps);
+      scale.data_ptr<float>(), shift.data_ptr<float>(), C, 1, rows,
+      momentum, eps);
   uint32_t total8 = rows * cvecs;
   FastDiv dcv;
   dcv.init(cvecs);
@@ -317,22 +423,31 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
   uint32_t rows = N * H * W, cvecs = C / 8;
   auto opt_f = x.options().dtype(at::kFloat);
-  auto pdb = at::zeros({(long)C}, opt_f), pdg = at::zeros({(long)C}, opt_f);
-  auto dgamma = at::empty({(long)C}, x.options());
-  auto dbeta = at::empty({(long)C}, x.options());
-  auto dx = at::empty_like(x);
   dim3 grid;
   uint32_t rpb;
   bn_grid(rows, cvecs, grid, rpb);
+  uint32_t nslabs = grid.y;
+  auto pdb = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto pdg = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto tdb = at::empty({(long)C}, opt_f), tdg = at::empty({(long)C}, opt_f);
+  auto dgamma = at::empty({(long)C}, x.options());
+  auto dbeta = at::empty({(long)C}, x.options());
+  auto dx = at::empty_like(x);
   auto s = bn_stream();
   bn_bwd_stats_kernel<<<grid, 256, 0, s>>>(
       (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
       pdg.data_ptr<float>(), rows, cvecs, rpb, fuse_relu ? 1 : 0);
+  auto rdb = at::empty({(long)C}, opt_f), rdg = at::empty({(long)C}, opt_f);
+  uint32_t ncv = std::min(cvecs, 8u);
+  slab_reduce2_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+      pdb.data_ptr<float>(), pdg.data_ptr<float>(), rdb.data_ptr<float>(),
+      rdg.data_ptr<float>(), C, nslabs);
   bn_bwd_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
-      pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
-      (__bf16*)dbeta.data_ptr(), C);
+      rdb.data_ptr<float>(), rdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
+      (__bf16*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
+      C, 1);
   uint32_t total8 = rows * cvecs;
   FastDiv dcv;
   dcv.init(cvecs);
@@ -341,7 +456,7 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
       (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), (const __bf16*)gamma.data_ptr(),
-      pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dx.data_ptr(),
+      tdb.data_ptr<float>(), tdg.data_ptr<float>(), (__bf16*)dx.data_ptr(),
       total8, dcv, 1.f / rows, fuse_relu ? 1 : 0);
   return {dx, dgamma, dbeta};
 }

@@ -80,6 +80,7 @@ class Module:
         self._optimizer: Optional[Optimizer] = None
         self._update_on_kvstore = True
         self._bucketer: Optional[GradBucketer] = None
+        self._use_fused_sgd = False
         self._batch_size = 0
 
         self._outputs: List[torch.Tensor] = []
@@ -233,16 +234,33 @@ class Module:
                     kv.init(name, p.data)
             for name, b in auxs.items():
                 kv.init(name, b.data, exclude_update=True)
-        # eagerly create optimizer state so elastic state broadcasts are
-        # well-defined on every rank (momenta exist before any update)
-        for i, (name, p) in enumerate(args.items()):
-            st = self._updater.states.get(i)
-            if st is None:
-                self._updater.states[i] = optimizer.create_state_multi_precision(i, p.data)
-        # gradient bucketing + overlap for the distributed path
-        if isinstance(kv, DistKVStore):
-            self._bucketer = GradBucketer(list(args.values()))
-            if initialize_from_kvstore:
+        # gradient bucketing (+ comm overlap when distributed). On GPU with
+        # plain SGD+bf16, parameters are flattened so the whole optimizer
+        # step is one fused HIP kernel per bucket.
+        from ..optimizer.optimizer import SGD as _SGD
+
+        on_gpu = self.device.type == "cuda"
+        self._use_fused_sgd = (
+            on_gpu
+            and type(optimizer) is _SGD
+            and self._dtype == torch.bfloat16
+            and not self.fixed_param_names
+        )
+        if self._use_fused_sgd:
+            from ..ops.hip import has_ext
+
+            self._use_fused_sgd = has_ext()
+        if not self._use_fused_sgd:
+            # eagerly create optimizer state so elastic state broadcasts are
+            # well-defined on every rank (momenta exist before any update)
+            for i, (name, p) in enumerate(args.items()):
+                st = self._updater.states.get(i)
+                if st is None:
+                    self._updater.states[i] = optimizer.create_state_multi_precision(i, p.data)
+        if isinstance(kv, DistKVStore) or on_gpu:
+            self._bucketer = GradBucketer(list(args.values()),
+                                          flatten_params=self._use_fused_sgd)
+            if isinstance(kv, DistKVStore) and initialize_from_kvstore:
                 self._sync_full_state()
         self.optimizer_initialized = True
 
@@ -252,10 +270,15 @@ class Module:
         if not dist.is_initialized() or dist.get_world_size() == 1:
             return
         args, auxs = self._classified_named_tensors()
-        tensors: List[torch.Tensor] = [p.data for p in args.values()]
-        tensors += [b.data for b in auxs.values()]
-        for i in sorted(self._updater.states.keys()):
-            tensors += _flatten_state(self._updater.states[i])
+        if self._use_fused_sgd and self._bucketer is not None:
+            # params are views of the flat buffers: one broadcast per bucket
+            tensors: List[torch.Tensor] = list(self._bucketer.state_tensors())
+            tensors += [b.data for b in auxs.values()]
+        else:
+            tensors = [p.data for p in args.values()]
+            tensors += [b.data for b in auxs.values()]
+            for i in sorted(self._updater.states.keys()):
+                tensors += _flatten_state(self._updater.states[i])
         with torch.no_grad():
             for t in tensors:
                 dist.broadcast(t, src=0)
@@ -314,10 +337,19 @@ class Module:
         args, auxs = self._classified_named_tensors()
         if self._bucketer is not None:
             self._bucketer.finish()  # join async all-reduces (sum over workers)
-            for i, (name, p) in enumerate(args.items()):
-                if name in self.fixed_param_names or p.grad is None:
-                    continue
-                self._updater(i, p.grad, p.data)
+            if self._use_fused_sgd:
+                opt = self._optimizer
+                opt.num_update += 1
+                lr = opt.lr_scheduler(opt.num_update) if opt.lr_scheduler else opt.lr
+                self._bucketer.fused_sgd_step(
+                    lr, getattr(opt, "momentum", 0.0), opt.wd, opt.rescale_grad,
+                    opt.clip_gradient or 0.0,
+                )
+            else:
+                for i, (name, p) in enumerate(args.items()):
+                    if name in self.fixed_param_names or p.grad is None:
+                        continue
+                    self._updater(i, p.grad, p.data)
         elif isinstance(self._kvstore, LocalKVStore) or self._kvstore is None:
             for i, (name, p) in enumerate(args.items()):
                 if name in self.fixed_param_names or p.grad is None:

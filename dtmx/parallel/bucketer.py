@@ -25,10 +25,22 @@ import torch.distributed as dist
 
 
 class GradBucketer:
+    """With `flatten_params=True` the parameters themselves are also moved
+    into per-bucket flat bf16 buffers (p.data becomes a [strided] view), plus
+    fp32 master and momentum flats — so the whole optimizer step is ONE fused
+    HIP kernel per bucket (dtmx/csrc/softmax_opt.hip sgd_mom_mp) instead of
+    2-4 torch ops per tensor (reference optimizer_op-inl.h MP_SGDMom redesign
+    as a multi-tensor kernel)."""
+
     def __init__(self, params: Sequence[torch.nn.Parameter],
-                 bucket_mb: Optional[float] = None, average: bool = False):
+                 bucket_mb: Optional[float] = None, average: bool = False,
+                 flatten_params: bool = False):
         self.params = [p for p in params if p.requires_grad]
         self.average = average
+        self.flatten_params = flatten_params
+        self.flat_w: List[torch.Tensor] = []
+        self.flat_master: List[torch.Tensor] = []
+        self.flat_mom: List[torch.Tensor] = []
         bucket_mb = bucket_mb or float(os.environ.get("DTMX_BUCKET_MB", "50"))
         bucket_bytes = int(bucket_mb * 1024 * 1024)
 
@@ -53,18 +65,58 @@ class GradBucketer:
         self._works: List[Tuple[int, dist.Work]] = []
         for bi, bucket in enumerate(self.buckets):
             numel = sum(p.numel() for p in bucket)
-            buf = torch.zeros(numel, dtype=bucket[0].dtype, device=bucket[0].device)
+            dev = bucket[0].device
+            buf = torch.zeros(numel, dtype=bucket[0].dtype, device=dev)
+            wbuf = None
+            if self.flatten_params:
+                wbuf = torch.empty(numel, dtype=bucket[0].dtype, device=dev)
             off = 0
             for p in bucket:
-                p.grad = buf[off : off + p.numel()].view_as(p)
-                off += p.numel()
+                n = p.numel()
+                p.grad = self._shaped_view(buf, off, p)
+                if wbuf is not None:
+                    with torch.no_grad():
+                        src = (
+                            p.detach().permute(0, 2, 3, 1).reshape(-1)  # KRSC order
+                            if p.dim() == 4 else p.detach().reshape(-1)
+                        )
+                        wbuf[off : off + n].copy_(src)
+                        p.data = self._shaped_view(wbuf, off, p)
+                off += n
                 self._param_bucket[id(p)] = bi
             self.flat.append(buf)
+            if wbuf is not None:
+                self.flat_w.append(wbuf)
+                self.flat_master.append(wbuf.detach().float())
+                self.flat_mom.append(torch.zeros(numel, dtype=torch.float32, device=dev))
         self._pending = [len(b) for b in self.buckets]
         self._hooks = [
             p.register_post_accumulate_grad_hook(self._on_grad_ready) for p in self.params
         ]
         self._enabled = dist.is_initialized() and dist.get_world_size() > 1
+
+    @staticmethod
+    def _shaped_view(buf: torch.Tensor, off: int, p: torch.Tensor) -> torch.Tensor:
+        """View matching p's logical shape AND memory layout (channels_last
+        for 4D conv weights) so flat bytes line up between w/grad/master."""
+        v = buf[off : off + p.numel()]
+        if p.dim() == 4:
+            K, C, R, S = p.shape
+            return v.view(K, R, S, C).permute(0, 3, 1, 2)
+        return v.view(p.shape)
+
+    # -- fused optimizer step ----------------------------------------------
+    def fused_sgd_step(self, lr: float, momentum: float, wd: float,
+                       rescale: float, clip: float = 0.0):
+        from ..ops.hip import require_ext
+
+        ext = require_ext()
+        for bi in range(len(self.flat)):
+            ext.sgd_mom_mp(self.flat_w[bi], self.flat[bi], self.flat_master[bi],
+                           self.flat_mom[bi], lr, momentum, wd, rescale, clip)
+
+    def state_tensors(self) -> List[torch.Tensor]:
+        return self.flat_w + self.flat_master + self.flat_mom
 
     # -- per-iteration lifecycle -------------------------------------------
     def zero_grad(self):
