@@ -94,12 +94,13 @@ struct EpiBF16 {
   const float* bias;  // nullable
   uint32_t M, N;
   int relu;
-  __device__ __forceinline__ void store(const f32x4 (&acc)[4][4], uint32_t m0,
+  template <int NJ>
+  __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,
                                         uint32_t n0, uint32_t lane) const {
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < NJ; ++j) {
         uint32_t n = n0 + j * 16 + (lane & 15);
         if (n >= N) continue;
         float b = bias ? bias[n] : 0.f;
@@ -118,12 +119,13 @@ struct EpiBF16 {
 struct EpiAtomicF32 {  // split-K partial accumulation (conv wgrad)
   float* c;
   uint32_t M, N;
-  __device__ __forceinline__ void store(const f32x4 (&acc)[4][4], uint32_t m0,
+  template <int NJ>
+  __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,
                                         uint32_t n0, uint32_t lane) const {
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < NJ; ++j) {
         uint32_t n = n0 + j * 16 + (lane & 15);
         if (n >= N) continue;
 #pragma unroll
@@ -138,20 +140,25 @@ struct EpiAtomicF32 {  // split-K partial accumulation (conv wgrad)
 
 // ------------------------------------------------------------------- kernel
 
-template <class PA, class PB, class EPI>
+template <int NJ, class PA, class PB, class EPI>
 __launch_bounds__(256, 2) __global__
 void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
                     uint32_t tiles_n, uint32_t kt_per_slice) {
+  // NJ = 16-col fragments per wave: BN tile = NJ*32 (128 for square work,
+  // 64 for narrow-N layers like resnet's K_out=64 convs where a 128 tile
+  // wastes half the MFMA work).
   // split-K: blockIdx.y selects a K-slice (fp32-atomic epilogue makes the
   // slices order-independent); single launch fills the chip.
+  constexpr uint32_t BN = NJ * 32;
   const uint32_t kt0 = blockIdx.y * kt_per_slice;
   const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
   if (kt0 >= ktiles_total) return;
-  __shared__ __bf16 smem[2][2][128 * 64];  // [buf][A|B][row*64+col] 64 KiB
+  __shared__ __bf16 smem[2][(128 + BN) * 64];  // A then B images per buffer
+  constexpr uint32_t B_OFF = 128 * 64;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
   const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  const uint32_t bm = (bid / tiles_n) * 128, bn = (bid % tiles_n) * 128;
+  const uint32_t bm = (bid / tiles_n) * 128, bn = (bid % tiles_n) * BN;
 
   // staging geometry: iteration it covers rows it*32+(t>>3), byte col (t&7)*16;
   // XOR swizzle applied to the SOURCE k-piece so the LDS image stays
@@ -168,15 +175,15 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
 #pragma unroll
     for (int it = 0; it < 4; ++it)
       glds16(pa.addr(bm + srow[it], kt * 8 + sk8[it]),
-             &smem[buf][0][it * 2048 + wave * 512]);
+             &smem[buf][it * 2048 + wave * 512]);
 #pragma unroll
-    for (int it = 0; it < 4; ++it)
+    for (int it = 0; it < NJ; ++it)
       glds16(pb.addr(bn + srow[it], kt * 8 + sk8[it]),
-             &smem[buf][1][it * 2048 + wave * 512]);
+             &smem[buf][B_OFF + it * 2048 + wave * 512]);
   };
 
-  const uint32_t wr = (wave >> 1) * 64, wc = (wave & 1) * 64;
-  f32x4 acc[4][4] = {};
+  const uint32_t wr = (wave >> 1) * 64, wc = (wave & 1) * NJ * 16;
+  f32x4 acc[4][NJ] = {};
 
   stage(0, kt0);
   wait_vmcnt0();
@@ -186,24 +193,24 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     if (kt + 1 < ktiles) stage(cur ^ 1, kt0 + kt + 1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[4], bfr[NJ];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         uint32_t row = wr + i * 16 + (lane & 15);
         uint32_t colb = (kk * 64 + ((lane >> 4) << 4)) ^ ((row & 7) << 4);
-        af[i] = *(const bf16x8*)((const char*)&smem[cur][0][0] + row * 128 + colb);
+        af[i] = *(const bf16x8*)((const char*)&smem[cur][0] + row * 128 + colb);
       }
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < NJ; ++j) {
         uint32_t row = wc + j * 16 + (lane & 15);
         uint32_t colb = (kk * 64 + ((lane >> 4) << 4)) ^ ((row & 7) << 4);
-        bfr[j] = *(const bf16x8*)((const char*)&smem[cur][1][0] + row * 128 + colb);
+        bfr[j] = *(const bf16x8*)((const char*)&smem[cur][B_OFF] + row * 128 + colb);
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NJ; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j],
                                                               acc[i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
@@ -212,7 +219,7 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     __syncthreads();
     cur ^= 1;
   }
-  epi.store(acc, bm + wr, bn + wc, lane);
+  epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
 }
 
 // --------------------------------------------------- transpose (row-gather)
@@ -321,13 +328,21 @@ static hipStream_t cur_stream() {
 template <class PA, class PB, class EPI>
 static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
                         uint32_t N, uint32_t K, uint32_t splitk = 1) {
-  uint32_t tiles_m = ceil_div(M, 128), tiles_n = ceil_div(N, 128);
   uint32_t ktiles_total = ceil_div(K, 64);
   splitk = std::min(splitk, ktiles_total);
   uint32_t kt_per = ceil_div(ktiles_total, splitk);
-  dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-  gemm_tn_kernel<PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
-      pa, pb, epi, ktiles_total, tiles_n, kt_per);
+  uint32_t tiles_m = ceil_div(M, 128);
+  if (N <= 64) {  // narrow-N tile: BN=64
+    uint32_t tiles_n = ceil_div(N, 64);
+    dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
+    gemm_tn_kernel<2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+  } else {
+    uint32_t tiles_n = ceil_div(N, 128);
+    dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
+    gemm_tn_kernel<4, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+  }
 }
 
 // -- plain GEMM entry points (FullyConnected; reference fully_connected-inl.h)
