@@ -350,7 +350,11 @@ def create(name: str = "local") -> KVStore:
         return LocalKVStore(name)
     if name in ("dist_sync", "dist_device_sync", "dist_async", "dist_sync_device", "nccl", "dist"):
         # single-process fallback: behave like local when no dist env is set
-        if "RANK" not in os.environ and "DMLC_ROLE" not in os.environ and "WORLD_SIZE" not in os.environ:
+        dist_env = any(
+            k in os.environ
+            for k in ("RANK", "WORLD_SIZE", "DMLC_ROLE", "DMLC_PS_ROOT_URI", "DMLC_WORKER_ID")
+        ) or os.environ.get("ELASTIC_TRAINING_ENABLED", "0").lower() in ("1", "true")
+        if not dist_env:
             logging.warning("kvstore '%s' requested without a distributed launcher; using local", name)
             return LocalKVStore(name)
         return DistKVStore(name)

@@ -221,19 +221,22 @@ class Module:
         self._updater = get_updater(optimizer)
 
         # register params with the kvstore; the store aliases the live
-        # parameter tensors (device-resident, zero-copy — "device" comm)
+        # parameter tensors (device-resident, zero-copy — "device" comm).
+        # A joiner (initialize_from_kvstore) skips the per-key init
+        # broadcasts: survivors don't re-issue them after a re-form, so the
+        # joiner adopts cluster state solely via _sync_full_state below —
+        # both sides then run the identical collective sequence
+        # (reference semantics: joiner pulls instead of pushing init,
+        # kvstore_dist.h:205-224 / model.py:116-133).
         if kv is not None:
-            kv.set_optimizer(optimizer) if self._update_on_kvstore else None
+            if self._update_on_kvstore:
+                kv.set_optimizer(optimizer)
             self._updater = kv._updater or self._updater
-            for i, (name, p) in enumerate(args.items()):
-                if isinstance(kv, DistKVStore) and not initialize_from_kvstore:
-                    kv.init(name, p.data)  # broadcast rank0 init
-                elif isinstance(kv, DistKVStore):
-                    kv.init(name, p.data)  # joiner: receives cluster weights
-                else:
+            if not (isinstance(kv, DistKVStore) and initialize_from_kvstore):
+                for i, (name, p) in enumerate(args.items()):
                     kv.init(name, p.data)
-            for name, b in auxs.items():
-                kv.init(name, b.data, exclude_update=True)
+                for name, b in auxs.items():
+                    kv.init(name, b.data, exclude_update=True)
         # gradient bucketing (+ comm overlap when distributed). On GPU with
         # plain SGD+bf16, parameters are flattened so the whole optimizer
         # step is one fused HIP kernel per bucket.

@@ -251,6 +251,13 @@ class ElasticContext:
         # rank 0 reads the roster version; everyone agrees via broadcast
         decision = [0, None]
         if dist.get_rank() == 0:
+            # publish the barrier's env payload (EPOCH_BEGIN etc.) so the
+            # launcher can hand it to ssh-launched joiners (reference
+            # ETNodeManager launchCommandOnNewWorker, elastic_training.cc:26-62)
+            for ek, ev in (env or {}).items():
+                self.store.set(f"cluster/env/{ek}", str(ev))
+            if env and "EPOCH_BEGIN" in env:
+                self.store.set("cluster/epoch", str(env["EPOCH_BEGIN"]))
             v = self.store.add("roster/version", 0)
             decision = [v, self._read_members(v) if v != self.version else None]
         obj = [decision]

@@ -1,0 +1,134 @@
+"""Elastic membership integration tests (the tests the reference LACKS —
+SURVEY.md §4 'No elastic tests exist'). Simulates the full add/remove
+protocol on localhost with gloo workers, mirroring the reference's
+WORKER_HOST_FILE rewrite workflow (elastic_training.cc:135-157).
+
+Scenario: start 2 workers -> a third joins mid-training (NEW_WORKER
+bootstrap, cluster-state adoption) -> the third is removed (exits cleanly
+at the next epoch barrier) -> survivors finish. Asserts:
+  - every surviving worker saw the worker count go 2 -> 3 -> 2
+  - final parameters are bit-identical across survivors (replicated-DP
+    invariant survives two re-forms)
+  - the joiner exited 0 after its removal
+  - the scheduler's audit log has the ADDED/REMOVED lines
+"""
+import json
+import os
+import subprocess
+import sys
+import time
+
+import pytest
+import torch
+
+from dtmx.parallel.rendezvous import Scheduler
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+WORKER = os.path.join(ROOT, "tests", "elastic_worker.py")
+
+
+def _free_port():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _spawn(wid, port, out, extra=None, num_epoch=8):
+    env = dict(os.environ)
+    env.update(
+        DMLC_WORKER_ID=wid,
+        DMLC_PS_ROOT_URI="127.0.0.1",
+        DMLC_PS_ROOT_PORT=str(port),
+        ELASTIC_TRAINING_ENABLED="1",
+        DTMX_BACKEND="gloo",
+        ELASTIC_TEST_OUT=out,
+        EPOCH_SLEEP="0.3",
+        NUM_EPOCH=str(num_epoch),
+        PYTHONPATH=ROOT,
+    )
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    env.update(extra or {})
+    return subprocess.Popen([sys.executable, WORKER], env=env,
+                            stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+
+
+def _wait_epoch(sched, at_least, timeout=60):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        try:
+            e = int(sched.store.get("cluster/epoch"))
+            if e >= at_least:
+                return e
+        except Exception:
+            pass
+        time.sleep(0.05)
+    raise TimeoutError(f"epoch {at_least} not reached")
+
+
+@pytest.mark.timeout(240)
+def test_elastic_join_then_leave(tmp_path):
+    port = _free_port()
+    hostfile = str(tmp_path / "hosts")
+    sched = Scheduler("127.0.0.1", port, ["127.0.0.1#0", "127.0.0.1#1"],
+                      hostfile=hostfile)
+    outs = {i: str(tmp_path / f"out{i}.json") for i in range(3)}
+    w0 = _spawn("127.0.0.1#0", port, outs[0])
+    w1 = _spawn("127.0.0.1#1", port, outs[1])
+    try:
+        # let a couple of epochs run, then add a worker
+        _wait_epoch(sched, 2)
+        sched.publish(["127.0.0.1#0", "127.0.0.1#1", "127.0.0.1#2"])
+        epoch_b = sched.store.get("cluster/env/EPOCH_BEGIN").decode()
+        w2 = _spawn("127.0.0.1#2", port, outs[2],
+                    extra={"NEW_WORKER": "1", "EPOCH_BEGIN": epoch_b})
+        # later, remove it again (removal of a non-initial worker)
+        _wait_epoch(sched, 5)
+        sched.publish(["127.0.0.1#0", "127.0.0.1#1"])
+
+        for name, p in (("w0", w0), ("w1", w1), ("w2", w2)):
+            rc = p.wait(timeout=150)
+            if rc != 0:
+                out, err = p.communicate()
+                raise AssertionError(f"{name} rc={rc}\n{err.decode()[-3000:]}")
+    finally:
+        for p in (w0, w1):
+            if p.poll() is None:
+                p.kill()
+
+    r0 = json.load(open(outs[0]))
+    r1 = json.load(open(outs[1]))
+    assert r0["final_workers"] == 2 and r1["final_workers"] == 2
+    assert 3 in r0["worker_counts"], r0  # saw the expansion
+    assert r0["worker_counts"][0] == 2
+    # replicated-DP invariant: identical params after two re-forms
+    assert r0["param_sum"] == pytest.approx(r1["param_sum"], rel=0, abs=0)
+    # audit log (reference elastic_training.cc:108-126)
+    log = open(hostfile + "_log").read()
+    assert "ADDED 127.0.0.1#2" in log and "REMOVED 127.0.0.1#2" in log
+
+
+@pytest.mark.timeout(240)
+def test_initial_workers_not_removable(tmp_path):
+    port = _free_port()
+    sched = Scheduler("127.0.0.1", port, ["a#0", "b#0"])
+    gens = sched.publish(["a#0"])  # try to remove an initial worker
+    assert gens == []  # refused
+    assert sched.members == ["a#0", "b#0"]
+
+
+def test_removal_beats_addition(tmp_path):
+    port = _free_port()
+    sched = Scheduler("127.0.0.1", port, ["a#0", "b#0", "c#0"])
+    # c is not initial? it IS initial here; use a non-initial member
+    sched.publish(["a#0", "b#0", "c#0", "d#0"])  # add d
+    gens = sched.publish(["a#0", "b#0", "e#0"])  # remove c(init, blocked)+d, add e
+    # removal generation first, addition second (elastic_training.cc:66-77)
+    assert len(gens) == 2
+    assert "d#0" not in gens[0] and "e#0" not in gens[0]
+    assert "e#0" in gens[1]
+    assert "c#0" in sched.members  # initial worker retained
