@@ -185,12 +185,29 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   const uint32_t wr = (wave >> 1) * 64, wc = (wave & 1) * NJ * 16;
   f32x4 acc[4][NJ] = {};
 
+  // 2-deep glds pipeline with COUNTED vmcnt + raw barriers (cdna_hip_
+  // programming.md §5 'Pipelining across barriers'): tile t+1's LDS-DMA
+  // stays in flight across the barrier while tile t computes — a vmcnt(0)
+  // drain per tile exposes the full HBM round trip at every K-step
+  // (measured ~4x off on resnet's short-K conv shapes).
+  constexpr int G = 4 + NJ;  // glds issued per wave per tile
+  auto wait_tile = [&](bool one_in_flight) {
+    if (one_in_flight) {
+      if constexpr (NJ == 4)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+  };
+
   stage(0, kt0);
-  wait_vmcnt0();
-  __syncthreads();
-  int cur = 0;
+  if (ktiles > 1) stage(1, kt0 + 1);
   for (uint32_t kt = 0; kt < ktiles; ++kt) {
-    if (kt + 1 < ktiles) stage(cur ^ 1, kt0 + kt + 1);
+    const uint32_t cur = kt & 1;
+    wait_tile(kt + 1 < ktiles);          // tile kt landed (kt+1 may fly on)
+    __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 af[4], bfr[NJ];
@@ -215,9 +232,8 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
                                                               acc[i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    wait_vmcnt0();
-    __syncthreads();
-    cur ^= 1;
+    __builtin_amdgcn_s_barrier();        // every wave done reading smem[cur]
+    if (kt + 2 < ktiles) stage(cur, kt0 + kt + 2);
   }
   epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
 }
