@@ -74,10 +74,19 @@ def main():
     it = SyntheticDataIter(1000, data_shape, max_iter=10 ** 9, dtype=dtype,
                            device=device, layout="NHWC")
 
+    use_graph = (
+        device.type == "cuda"
+        and os.environ.get("DTMX_HIPGRAPH", "0") == "1"
+        and not dist_mode  # multi-rank capture of RCCL: enable after validation
+    )
+
     def step():
         batch = it.next()
-        mod.forward_backward(batch)
-        mod.update()
+        if use_graph:
+            mod.graphed_step(batch)
+        else:
+            mod.forward_backward(batch)
+            mod.update()
 
     for _ in range(args.warmup):
         step()

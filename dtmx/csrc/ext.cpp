@@ -32,6 +32,8 @@ std::vector<at::Tensor> softmax_ce_fwd(const at::Tensor&, const at::Tensor&);
 at::Tensor softmax_ce_bwd(const at::Tensor&, const at::Tensor&, const at::Tensor&);
 void sgd_mom_mp(at::Tensor, const at::Tensor&, at::Tensor, at::Tensor, double,
                 double, double, double, double);
+void sgd_mom_mp_dev(at::Tensor, const at::Tensor&, at::Tensor, at::Tensor,
+                    const at::Tensor&);
 void sgd_mom_f32(at::Tensor, const at::Tensor&, at::Tensor, double, double,
                  double, double, double);
 }  // namespace dtmx
@@ -56,5 +58,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_fwd", &dtmx::softmax_ce_fwd);
   m.def("softmax_ce_bwd", &dtmx::softmax_ce_bwd);
   m.def("sgd_mom_mp", &dtmx::sgd_mom_mp);
+  m.def("sgd_mom_mp_dev", &dtmx::sgd_mom_mp_dev);
   m.def("sgd_mom_f32", &dtmx::sgd_mom_f32);
 }

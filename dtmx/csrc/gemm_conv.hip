@@ -90,10 +90,25 @@ struct ConvDgradA {
 // ---------------------------------------------------------------- epilogues
 
 struct EpiBF16 {
+  static constexpr bool kLdsStage = true;
   __bf16* c;
   const float* bias;  // nullable
   uint32_t M, N;
   int relu;
+  // coalesced row-chunk store used by the kernel's LDS-staged epilogue
+  __device__ __forceinline__ void store_chunk(uint32_t m, uint32_t n0,
+                                              bf16x8 v) const {
+    if (m >= M || n0 >= N) return;  // N % 8 == 0: chunks never straddle
+    if (bias || relu) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float f = (float)v[e] + (bias ? bias[n0 + e] : 0.f);
+        if (relu) f = fmaxf(f, 0.f);
+        v[e] = (__bf16)f;
+      }
+    }
+    *(bf16x8*)(c + (size_t)m * N + n0) = v;
+  }
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,
                                         uint32_t n0, uint32_t lane) const {
@@ -117,8 +132,10 @@ struct EpiBF16 {
 };
 
 struct EpiAtomicF32 {  // split-K partial accumulation (conv wgrad)
+  static constexpr bool kLdsStage = false;
   float* c;
   uint32_t M, N;
+  __device__ __forceinline__ void store_chunk(uint32_t, uint32_t, bf16x8) const {}
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,
                                         uint32_t n0, uint32_t lane) const {
@@ -235,7 +252,32 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     __builtin_amdgcn_s_barrier();        // every wave done reading smem[cur]
     if (kt + 2 < ktiles) stage(cur, kt0 + kt + 2);
   }
-  epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
+  if constexpr (EPI::kLdsStage) {
+    // stage the C tile through LDS so global stores are 16-B row chunks:
+    // the direct fragment store is 16-64 half-coalesced 2-B stores per lane
+    // (store-issue-bound; cdna_hip_programming.md T21 diagnosis).
+    __syncthreads();  // nothing in flight; reuse smem[0] as [128][BN] bf16
+    __bf16* ct = &smem[0][0];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < NJ; ++j) {
+        uint32_t col = wc + j * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          uint32_t row = wr + i * 16 + ((lane >> 4) << 2) + r;
+          ct[row * BN + col] = (__bf16)acc[i][j][r];
+        }
+      }
+    __syncthreads();
+    constexpr uint32_t CHUNKS = 128 * BN / 8;
+    for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
+      uint32_t row = idx / (BN / 8), nc = idx % (BN / 8);
+      epi.store_chunk(bm + row, bn + nc * 8, *(const bf16x8*)(ct + row * BN + nc * 8));
+    }
+  } else {
+    epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
+  }
 }
 
 // --------------------------------------------------- transpose (row-gather)
@@ -457,7 +499,12 @@ at::Tensor conv_fwd(const at::Tensor& x, const at::Tensor& w, long stride,
                      x.options(), at::MemoryFormat::ChannelsLast);
   uint32_t M = N * P * Q, Ktot = R * S * C;
   EpiBF16 epi{(__bf16*)y.data_ptr(), nullptr, M, Ko, 0};
-  if (C % 8 == 0) {
+  if (R == 1 && S == 1 && stride == 1 && pad == 0 && C % 8 == 0) {
+    // 1x1/s1: the im2col matrix IS x — pure dense GEMM, no gather decode
+    DenseP pa{(const __bf16*)x.data_ptr(), zero_page(x), M, C, C};
+    DenseP pb{(const __bf16*)w.data_ptr(), zero_page(x), Ko, C, C};
+    launch_gemm(pa, pb, epi, M, Ko, C);
+  } else if (C % 8 == 0) {
     DenseP pb{(const __bf16*)w.data_ptr(), zero_page(x), Ko, Ktot, Ktot};
     ConvFwdA pa;
     pa.x = (const __bf16*)x.data_ptr();
@@ -510,6 +557,14 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
   auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
                       at::MemoryFormat::ChannelsLast);
   uint32_t M = N * H * W_, Ktot = R * S * Ko;
+  if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+    // 1x1/s1 dgrad: dx = dy @ W^T — dense, no gather
+    DenseP pad_{(const __bf16*)dyc.data_ptr(), zero_page(dy), M, Ko, Ko};
+    DenseP pbd{(const __bf16*)wt.data_ptr(), zero_page(dy), C, Ko, Ko};
+    EpiBF16 epid{(__bf16*)dx.data_ptr(), nullptr, M, C, 0};
+    launch_gemm(pad_, pbd, epid, M, C, Ko);
+    return dx;
+  }
   ConvDgradA pa;
   pa.dy = (const __bf16*)dyc.data_ptr();
   pa.zero = zero_page(dy);

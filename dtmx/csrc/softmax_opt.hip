@@ -87,6 +87,30 @@ __global__ void sgd_mom_mp_kernel(__bf16* __restrict__ w, const __bf16* __restri
   }
 }
 
+// device-hyperparameter variant (hipGraph-capturable: lr/rescale live in a
+// device buffer updated between replays, not baked into kernel args).
+// hyper = {lr, momentum, wd, rescale, clip}
+__global__ void sgd_mom_mp_dev_kernel(__bf16* __restrict__ w,
+                                      const __bf16* __restrict__ g,
+                                      float* __restrict__ master,
+                                      float* __restrict__ mom, size_t total,
+                                      const float* __restrict__ hyper) {
+  const float lr = hyper[0], momentum = hyper[1], wd = hyper[2],
+              rescale = hyper[3], clip = hyper[4];
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    float gv = (float)g[i] * rescale;
+    if (clip > 0.f) gv = fminf(fmaxf(gv, -clip), clip);
+    gv += wd * master[i];
+    float m = momentum * mom[i] - lr * gv;
+    mom[i] = m;
+    float nw = master[i] + m;
+    master[i] = nw;
+    w[i] = (__bf16)nw;
+  }
+}
+
 __global__ void sgd_mom_f32_kernel(float* __restrict__ w, const float* __restrict__ g,
                                    float* __restrict__ mom, size_t total, float lr,
                                    float momentum, float wd, float rescale,
@@ -142,6 +166,16 @@ void sgd_mom_mp(at::Tensor w, const at::Tensor& g, at::Tensor master,
       (__bf16*)w.data_ptr(), (const __bf16*)g.data_ptr(),
       master.data_ptr<float>(), mom.data_ptr<float>(), total, lr, momentum, wd,
       rescale, clip);
+}
+
+void sgd_mom_mp_dev(at::Tensor w, const at::Tensor& g, at::Tensor master,
+                    at::Tensor mom, const at::Tensor& hyper) {
+  size_t total = w.numel();
+  uint32_t blocks = std::min<size_t>((total + 255) / 256, 4096);
+  sgd_mom_mp_dev_kernel<<<blocks, 256, 0, so_stream()>>>(
+      (__bf16*)w.data_ptr(), (const __bf16*)g.data_ptr(),
+      master.data_ptr<float>(), mom.data_ptr<float>(), total,
+      hyper.data_ptr<float>());
 }
 
 void sgd_mom_f32(at::Tensor w, const at::Tensor& g, at::Tensor mom, double lr,

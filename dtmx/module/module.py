@@ -82,6 +82,12 @@ class Module:
         self._bucketer: Optional[GradBucketer] = None
         self._use_fused_sgd = False
         self._batch_size = 0
+        # hipGraph step capture (HIP graphs in place of the reference's
+        # op-segment bulking, graph_executor.cc:1318)
+        self._capture_mode = False
+        self._graph = None
+        self._static_inputs = None
+        self._hyper_dev = None
 
         self._outputs: List[torch.Tensor] = []
         self._loss: Optional[torch.Tensor] = None
@@ -344,9 +350,12 @@ class Module:
                 opt = self._optimizer
                 opt.num_update += 1
                 lr = opt.lr_scheduler(opt.num_update) if opt.lr_scheduler else opt.lr
+                hyper = None
+                if self._capture_mode:
+                    hyper = self._update_hyper(lr)
                 self._bucketer.fused_sgd_step(
                     lr, getattr(opt, "momentum", 0.0), opt.wd, opt.rescale_grad,
-                    opt.clip_gradient or 0.0,
+                    opt.clip_gradient or 0.0, hyper=hyper,
                 )
             else:
                 for i, (name, p) in enumerate(args.items()):
@@ -366,6 +375,68 @@ class Module:
                 self._kvstore.pull(name, out=p.data)
         if update_aux_params:
             self.store_aux_params()
+
+    # -------------------------------------------------- hipGraph step replay
+    def _update_hyper(self, lr: float) -> torch.Tensor:
+        opt = self._optimizer
+        if self._hyper_dev is None:
+            self._hyper_dev = torch.zeros(5, dtype=torch.float32, device=self.device)
+            self._hyper_host = torch.zeros(5, dtype=torch.float32, pin_memory=True)
+        self._hyper_host[0] = lr
+        self._hyper_host[1] = getattr(opt, "momentum", 0.0)
+        self._hyper_host[2] = opt.wd
+        self._hyper_host[3] = opt.rescale_grad
+        self._hyper_host[4] = opt.clip_gradient or 0.0
+        self._hyper_dev.copy_(self._hyper_host, non_blocking=True)
+        return self._hyper_dev
+
+    def graphed_step(self, data_batch: DataBatch):
+        """One fwd+bwd+update captured in a hipGraph and replayed (launch-
+        bound inner loops belong in hipGraphs — the MI355X-native replacement
+        for the reference's executor op-segment bulking). Requires static
+        shapes; falls back to eager when capture is unavailable.
+
+        Multi-rank: the bucketer's RCCL all-reduces are captured into the
+        graph (ProcessGroupNCCL supports hipGraph capture)."""
+        if not (self.device.type == "cuda" and self._use_fused_sgd):
+            self.forward_backward(data_batch)
+            self.update()
+            return
+        if self._graph is None:
+            self._capture_mode = True
+            data = [self._to_device(d, True) for d in data_batch.data]
+            labels = [self._to_device(l, False) for l in (data_batch.label or [])]
+            self._static_inputs = (data, labels)
+            static_batch = DataBatch(data=data, label=labels)
+            # warmup on a side stream (allocator + autograd graph priming)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    self.forward_backward(static_batch)
+                    self.update()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self.forward_backward(static_batch)
+                self.update()
+            self._graph = g
+            return
+        data, labels = self._static_inputs
+        with torch.no_grad():
+            for dst, src in zip(data, data_batch.data):
+                if dst.data_ptr() != src.data_ptr():
+                    dst.copy_(self._to_device(src, True), non_blocking=True)
+            for dst, src in zip(labels, data_batch.label or []):
+                if dst.data_ptr() != src.data_ptr():
+                    dst.copy_(self._to_device(src, False), non_blocking=True)
+        self._update_hyper(
+            self._optimizer.lr_scheduler(self._optimizer.num_update)
+            if self._optimizer.lr_scheduler else self._optimizer.lr
+        )
+        self._optimizer.num_update += 1
+        self._graph.replay()
 
     def store_aux_params(self):
         """Cluster-average BN running stats (reference module.py:668-683 ->

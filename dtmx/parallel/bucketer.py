@@ -107,13 +107,21 @@ class GradBucketer:
 
     # -- fused optimizer step ----------------------------------------------
     def fused_sgd_step(self, lr: float, momentum: float, wd: float,
-                       rescale: float, clip: float = 0.0):
+                       rescale: float, clip: float = 0.0,
+                       hyper: Optional[torch.Tensor] = None):
+        """hyper: optional device float32[5] {lr,mom,wd,rescale,clip} — when
+        given, the kernel reads hyperparameters from it (hipGraph-capturable:
+        update the buffer between replays instead of re-recording)."""
         from ..ops.hip import require_ext
 
         ext = require_ext()
         for bi in range(len(self.flat)):
-            ext.sgd_mom_mp(self.flat_w[bi], self.flat[bi], self.flat_master[bi],
-                           self.flat_mom[bi], lr, momentum, wd, rescale, clip)
+            if hyper is not None:
+                ext.sgd_mom_mp_dev(self.flat_w[bi], self.flat[bi],
+                                   self.flat_master[bi], self.flat_mom[bi], hyper)
+            else:
+                ext.sgd_mom_mp(self.flat_w[bi], self.flat[bi], self.flat_master[bi],
+                               self.flat_mom[bi], lr, momentum, wd, rescale, clip)
 
     def state_tensors(self) -> List[torch.Tensor]:
         return self.flat_w + self.flat_master + self.flat_mom

@@ -1,0 +1,79 @@
+#!/usr/bin/env python3
+"""Kernel microbenchmarks: per-shape TF/s for the dtmx GEMM/conv kernels
+(within-run A/B harness for kernel tuning; cdna_hip_programming.md §5.4)."""
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from dtmx.ops.hip import require_ext
+
+ext = require_ext()
+DEV = "cuda:0"
+
+
+def timeit(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def tf(flops, sec):
+    return flops / sec / 1e12
+
+
+def nhwc(t):
+    return t.to(DEV).contiguous(memory_format=torch.channels_last)
+
+
+def main():
+    torch.manual_seed(0)
+    # 1) square dense GEMM peak of the kernel
+    for M, N, K in [(4096, 4096, 4096), (8192, 8192, 8192)]:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+        b = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+        s = timeit(lambda: ext.linear_fwd(a, b, None))
+        print(f"dense {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
+
+    # 2) resnet-shaped dense (same GEMM the conv would do, no gather)
+    for M, N, K in [(401408, 64, 576), (401408, 128, 256), (100352, 256, 1152),
+                    (25088, 512, 2304), (6272, 2048, 512)]:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+        b = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+        s = timeit(lambda: ext.linear_fwd(a, b, None))
+        print(f"skinny {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
+
+    # 3) conv fwd/dgrad/wgrad on key resnet-50 bs128 layers
+    cases = [
+        ("l1.conv2 3x3 64->64 56^2 s1", 128, 64, 56, 64, 3, 1, 1),
+        ("l1.conv3 1x1 64->256 56^2", 128, 64, 56, 256, 1, 1, 0),
+        ("l2.conv2 3x3 128->128 28^2", 128, 128, 28, 128, 3, 1, 1),
+        ("l3.conv2 3x3 256->256 14^2", 128, 256, 14, 256, 3, 1, 1),
+        ("l4.conv2 3x3 512->512 7^2", 128, 512, 7, 512, 3, 1, 1),
+        ("l2.ds 1x1 256->512 s2 28^2", 128, 256, 56, 512, 1, 2, 0),
+    ]
+    for name, Nb, C, H, K, R, stride, pad in cases:
+        x = torch.randn(Nb, C, H, H, dtype=torch.bfloat16)
+        w = torch.randn(K, C, R, R, dtype=torch.bfloat16) * 0.1
+        xd, wd = nhwc(x), nhwc(w)
+        P = (H + 2 * pad - R) // stride + 1
+        dy = nhwc(torch.randn(Nb, K, P, P, dtype=torch.bfloat16))
+        flops = 2.0 * Nb * P * P * K * R * R * C
+        sf = timeit(lambda: ext.conv_fwd(xd, wd, stride, pad))
+        sd = timeit(lambda: ext.conv_dgrad(dy, wd, stride, pad, H, H))
+        sw = timeit(lambda: ext.conv_wgrad(xd, dy, R, R, stride, pad))
+        print(f"{name}: fwd {tf(flops,sf):6.1f} TF ({sf*1e3:.3f} ms)  "
+              f"dgrad {tf(flops,sd):6.1f} TF ({sd*1e3:.3f})  "
+              f"wgrad {tf(flops,sw):6.1f} TF ({sw*1e3:.3f})")
+
+
+if __name__ == "__main__":
+    main()
