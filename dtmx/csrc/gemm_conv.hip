@@ -280,6 +280,132 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   }
 }
 
+// -------------------------------------------- NT kernel (contraction-major)
+// Both operands stored [Kd][cols] (contraction-major) — conv wgrad consumes
+// dy and the im2col view of x NHWC-NATIVE, eliminating the dy^T/im2col^T
+// materialization passes (they were ~2x the wgrad GEMM's own HBM traffic).
+// LDS images [64 kd][BM|BN]; fragments gathered by scalar u16 column reads
+// (8 per fragment), hidden under the MFMA stream.
+
+struct WgradDyA {  // A: [kd = npq][m = ko] = dy NHWC as-is
+  const __bf16* dy;
+  const __bf16* zero;
+  uint32_t Kd, Mdim;
+  __device__ __forceinline__ const void* addr(uint32_t kd, uint32_t c8) const {
+    if (kd >= Kd || c8 * 8 >= Mdim) return zero;
+    return dy + (size_t)kd * Mdim + c8 * 8;
+  }
+};
+
+struct WgradXcolB {  // B: [kd = npq][n = (r,s,c)] gathered from x NHWC
+  const __bf16* x;
+  const __bf16* zero;
+  uint32_t Kd, Ndim;  // Kd = NPQ, Ndim = R*S*C
+  uint32_t C, H, W, Q, S;
+  int u, v, ph, pw;
+  FastDiv dQ, dPQ, dC, dS;
+  __device__ __forceinline__ const void* addr(uint32_t kd, uint32_t c8) const {
+    uint32_t nn = c8 * 8;
+    if (kd >= Kd || nn >= Ndim) return zero;
+    uint32_t n = dPQ.div(kd), pq = dPQ.mod(kd, n);
+    uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
+    uint32_t rs = dC.div(nn), c = dC.mod(nn, rs);
+    uint32_t r = dS.div(rs), s = dS.mod(rs, r);
+    int ih = (int)(p * u) - ph + (int)r;
+    int iw = (int)(q * v) - pw + (int)s;
+    if ((uint32_t)ih >= H || (uint32_t)iw >= W) return zero;
+    return x + (((size_t)n * H + ih) * W + iw) * C + c;
+  }
+};
+
+template <int NJ, class PA, class PB, class EPI>
+__launch_bounds__(256, 2) __global__
+void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
+                    uint32_t tiles_n, uint32_t kt_per_slice) {
+  constexpr uint32_t BN = NJ * 32;
+  const uint32_t kt0 = blockIdx.y * kt_per_slice;
+  const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
+  if (kt0 >= ktiles_total) return;
+  __shared__ __bf16 smem[2][64 * (128 + BN)];  // [kd][m] then [kd][n]
+  constexpr uint32_t B_OFF = 64 * 128;
+  const uint32_t t = threadIdx.x;
+  const uint32_t wave = t >> 6, lane = t & 63;
+  const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const uint32_t bm = (bid / tiles_n) * 128, bn = (bid % tiles_n) * BN;
+
+  // staging: A rows are 256 B (128 m), chunk = 16 kd-rows per 4 KiB;
+  //          B rows are BN*2 B, NJ 4-KiB chunks total.
+  auto stage = [&](int buf, uint32_t kt) {
+    const uint32_t kd0 = kt * 64;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      uint32_t krow = it * 16 + (t >> 4), piece = t & 15;
+      glds16(pa.addr(kd0 + krow, bm / 8 + piece),
+             &smem[buf][it * 2048 + wave * 512]);
+    }
+#pragma unroll
+    for (int it = 0; it < NJ; ++it) {
+      uint32_t krow = it * (256 / (NJ * 4)) + t / (NJ * 4);
+      uint32_t piece = t % (NJ * 4);
+      glds16(pb.addr(kd0 + krow, bn / 8 + piece),
+             &smem[buf][B_OFF + it * 2048 + wave * 512]);
+    }
+  };
+
+  const uint32_t wr = (wave >> 1) * 64, wc = (wave & 1) * NJ * 16;
+  f32x4 acc[4][NJ] = {};
+
+  constexpr int G = 4 + NJ;
+  auto wait_tile = [&](bool one_in_flight) {
+    if (one_in_flight) {
+      if constexpr (NJ == 4)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+  };
+
+  stage(0, kt0);
+  if (ktiles > 1) stage(1, kt0 + 1);
+  for (uint32_t kt = 0; kt < ktiles; ++kt) {
+    const uint32_t cur = kt & 1;
+    wait_tile(kt + 1 < ktiles);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bfr[NJ];
+      const uint32_t kbase = kk * 32 + ((lane >> 4) << 3);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        uint32_t m = wr + i * 16 + (lane & 15);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          af[i][e] = smem[cur][(kbase + e) * 128 + m];
+      }
+#pragma unroll
+      for (int j = 0; j < NJ; ++j) {
+        uint32_t n = wc + j * 16 + (lane & 15);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          bfr[j][e] = smem[cur][B_OFF + (kbase + e) * BN + n];
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j],
+                                                              acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_s_barrier();
+    if (kt + 2 < ktiles) stage(cur, kt0 + kt + 2);
+  }
+  epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
+}
+
 // --------------------------------------------------- transpose (row-gather)
 // out[c][m] = row(m)[c]  for generic [M][C] -> [C][out_ld] bf16 transposes
 // (linear dgrad/wgrad operands, conv-wgrad dy^T and gathered im2col^T).
@@ -399,6 +525,27 @@ static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
     uint32_t tiles_n = ceil_div(N, 128);
     dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
     gemm_tn_kernel<4, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+  }
+}
+
+template <class PA, class PB, class EPI>
+static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
+                           uint32_t M, uint32_t N, uint32_t K,
+                           uint32_t splitk = 1) {
+  uint32_t ktiles_total = ceil_div(K, 64);
+  splitk = std::min(splitk, ktiles_total);
+  uint32_t kt_per = ceil_div(ktiles_total, splitk);
+  uint32_t tiles_m = ceil_div(M, 128);
+  if (N <= 64) {
+    uint32_t tiles_n = ceil_div(N, 64);
+    dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
+    gemm_nt_kernel<2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+  } else {
+    uint32_t tiles_n = ceil_div(N, 128);
+    dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
+    gemm_nt_kernel<4, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
         pa, pb, epi, ktiles_total, tiles_n, kt_per);
   }
 }
@@ -590,6 +737,33 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
   uint32_t M = N * P * Q;               // the contraction length
   uint32_t Mpad = ((M + 7) / 8) * 8;
   uint32_t RSC = R * S * C;
+
+  if (C % 8 == 0 && Ko % 8 == 0) {
+    // transpose-free: dy and the im2col view of x are read NHWC-native by
+    // the contraction-major (NT) kernel
+    uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, RSC <= 64 ? 64 : 128);
+    uint32_t ktiles = ceil_div(M, 64);
+    uint32_t splitk = std::max<uint32_t>(
+        1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
+    auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
+    WgradDyA pa;
+    pa.dy = (const __bf16*)dy.data_ptr();
+    pa.zero = zero_page(x);
+    pa.Kd = M;
+    pa.Mdim = Ko;
+    WgradXcolB pb;
+    pb.x = (const __bf16*)x.data_ptr();
+    pb.zero = zero_page(x);
+    pb.Kd = M; pb.Ndim = RSC; pb.C = C; pb.H = H; pb.W = W_; pb.Q = Q; pb.S = S;
+    pb.u = stride; pb.v = stride; pb.ph = pad; pb.pw = pad;
+    pb.dQ.init(Q); pb.dPQ.init(P * Q); pb.dC.init(C); pb.dS.init(S);
+    EpiAtomicF32 epi{dw32.data_ptr<float>(), Ko, RSC};
+    launch_gemm_nt(pa, pb, epi, Ko, RSC, M, splitk);
+    auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
+                  .to(at::kBFloat16)
+                  .permute({0, 3, 1, 2});
+    return dw.contiguous(at::MemoryFormat::ChannelsLast);
+  }
 
   // dy^T : [NPQ][Ko] -> [Ko(+pad)][Mpad]
   auto dyt = transpose2d(
