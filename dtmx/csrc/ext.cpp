@@ -36,6 +36,9 @@ void sgd_mom_mp_dev(at::Tensor, const at::Tensor&, at::Tensor, at::Tensor,
                     const at::Tensor&);
 void sgd_mom_f32(at::Tensor, const at::Tensor&, at::Tensor, double, double,
                  double, double, double);
+// compress.hip
+at::Tensor quantize_2bit(const at::Tensor&, at::Tensor, double);
+at::Tensor dequantize_2bit(const at::Tensor&, long, double);
 }  // namespace dtmx
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -60,4 +63,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_mom_mp", &dtmx::sgd_mom_mp);
   m.def("sgd_mom_mp_dev", &dtmx::sgd_mom_mp_dev);
   m.def("sgd_mom_f32", &dtmx::sgd_mom_f32);
+  m.def("quantize_2bit", &dtmx::quantize_2bit);
+  m.def("dequantize_2bit", &dtmx::dequantize_2bit);
 }

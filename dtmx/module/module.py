@@ -586,13 +586,48 @@ class Module:
         return torch.cat(outs, dim=0)
 
     # ----------------------------------------------------------- checkpoints
+    def save_optimizer_states(self, fname: str):
+        """reference module.py:839-868 save_optimizer_states."""
+        import pickle
+
+        if self._use_fused_sgd and self._bucketer is not None:
+            payload = {
+                "fused": True,
+                "master": [t.cpu() for t in self._bucketer.flat_master],
+                "mom": [t.cpu() for t in self._bucketer.flat_mom],
+                "num_update": self._optimizer.num_update,
+            }
+            with open(fname, "wb") as f:
+                pickle.dump(payload, f)
+        elif self._kvstore is not None:
+            self._kvstore._updater = self._updater
+            self._kvstore._optimizer = self._optimizer
+            self._kvstore.save_optimizer_states(fname)
+
+    def load_optimizer_states(self, fname: str):
+        import pickle
+
+        with open(fname, "rb") as f:
+            payload = pickle.load(f)
+        if isinstance(payload, dict) and payload.get("fused"):
+            assert self._use_fused_sgd and self._bucketer is not None
+            with torch.no_grad():
+                for dst, src in zip(self._bucketer.flat_master, payload["master"]):
+                    dst.copy_(src.to(dst.device))
+                for dst, src in zip(self._bucketer.flat_mom, payload["mom"]):
+                    dst.copy_(src.to(dst.device))
+                for wb, mb in zip(self._bucketer.flat_w, self._bucketer.flat_master):
+                    wb.copy_(mb.to(wb.dtype))
+            self._optimizer.num_update = payload.get("num_update", 0)
+        elif self._kvstore is not None:
+            self._kvstore._updater = self._updater
+            self._kvstore.load_optimizer_states(fname)
+
     def save_checkpoint(self, prefix: str, epoch: int, save_optimizer_states=False):
         arg_p, aux_p = self.get_params()
         model_mod.save_checkpoint(prefix, epoch, self.symbol, arg_p, aux_p)
-        if save_optimizer_states and self._kvstore is not None:
-            self._kvstore._updater = self._updater
-            self._kvstore._optimizer = self._optimizer
-            self._kvstore.save_optimizer_states("%s-%04d.states" % (prefix, epoch))
+        if save_optimizer_states:
+            self.save_optimizer_states("%s-%04d.states" % (prefix, epoch))
 
     @staticmethod
     def load(prefix: str, epoch: int, symbol_builder, **kwargs) -> "Module":

@@ -36,6 +36,15 @@ class TwoBitCompression:
         if res is None or res.shape != grad.shape:
             res = torch.zeros_like(grad, dtype=torch.float32)
             self._residual[key] = res
+        if grad.is_cuda and grad.dtype == torch.bfloat16:
+            from ..ops.hip import require_ext
+
+            ext = require_ext()
+            packed = ext.quantize_2bit(grad.contiguous().view(-1), res.view(-1),
+                                       self.threshold)
+            return ext.dequantize_2bit(packed, grad.numel(), self.threshold).view(
+                grad.shape
+            )
         g = grad.float() + res
         q = torch.where(
             g >= self.threshold,
