@@ -335,18 +335,24 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
 
   // staging: A rows are 256 B (128 m), chunk = 16 kd-rows per 4 KiB;
   //          B rows are BN*2 B, NJ 4-KiB chunks total.
+  // column-XOR swizzle by kd-row bit 3: the two 16-lane clusters of a
+  // 32-lane read half sit at kd and kd+8, which land on identical banks in
+  // a linear image (2-way conflict, SQ_LDS_BANK_CONFLICT ~3e9/step measured)
+  // — flipping 16 columns on odd (kd>>3) separates them. Applied on the
+  // glds SOURCE (lane-linear dest, rule 21) and un-applied on the reads.
   auto stage = [&](int buf, uint32_t kt) {
     const uint32_t kd0 = kt * 64;
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
-      uint32_t krow = it * 16 + (t >> 4), piece = t & 15;
+      uint32_t krow = it * 16 + (t >> 4);
+      uint32_t piece = (t & 15) ^ (((krow >> 3) & 1) << 1);
       glds16(pa.addr(kd0 + krow, bm / 8 + piece),
              &smem[buf][it * 2048 + wave * 512]);
     }
 #pragma unroll
     for (int it = 0; it < NJ; ++it) {
       uint32_t krow = it * (256 / (NJ * 4)) + t / (NJ * 4);
-      uint32_t piece = t % (NJ * 4);
+      uint32_t piece = (t % (NJ * 4)) ^ (((krow >> 3) & 1) << 1);
       glds16(pb.addr(kd0 + krow, bn / 8 + piece),
              &smem[buf][B_OFF + it * 2048 + wave * 512]);
     }
@@ -381,15 +387,19 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
       for (int i = 0; i < 4; ++i) {
         uint32_t m = wr + i * 16 + (lane & 15);
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          af[i][e] = smem[cur][(kbase + e) * 128 + m];
+        for (int e = 0; e < 8; ++e) {
+          uint32_t kr = kbase + e;
+          af[i][e] = smem[cur][kr * 128 + (m ^ (((kr >> 3) & 1) << 4))];
+        }
       }
 #pragma unroll
       for (int j = 0; j < NJ; ++j) {
         uint32_t n = wc + j * 16 + (lane & 15);
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          bfr[j][e] = smem[cur][B_OFF + (kbase + e) * BN + n];
+        for (int e = 0; e < 8; ++e) {
+          uint32_t kr = kbase + e;
+          bfr[j][e] = smem[cur][B_OFF + kr * BN + (n ^ (((kr >> 3) & 1) << 4))];
+        }
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
