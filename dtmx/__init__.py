@@ -10,8 +10,17 @@ Usage mirrors `import mxnet as mx`:
     mod = mx.mod.Module(net, context=mx.gpu(0))
     mod.bind(...); mod.fit(train_iter, kvstore=kv, ...)
 """
+import os as _os
+
+# NaiveEngine analog (reference MXNET_ENGINE_TYPE=NaiveEngine, engine.cc:32-48):
+# serialize every kernel launch for race bisection. Must be set before HIP
+# initializes, hence here at package import.
+if _os.environ.get("DTMX_BLOCKING", "0") == "1":
+    _os.environ.setdefault("AMD_SERIALIZE_KERNEL", "3")
+    _os.environ.setdefault("HIP_LAUNCH_BLOCKING", "1")
+
 from . import callback, context, initializer, io, lr_scheduler, metric, model
-from . import kvstore, models, ndarray, optimizer, parallel, profiler
+from . import gluon, kvstore, models, ndarray, optimizer, parallel, profiler
 from . import module as mod
 from .context import Context, cpu, gpu, num_gpus
 from .module import Module
