@@ -54,6 +54,19 @@ class _ConvNHWC(torch.autograd.Function):
 
 def conv2d(x, w, stride: int = 1, padding: int = 0):
     if _use_hip(x):
+        C = x.shape[1]
+        if C % 8 != 0:
+            # zero-pad channels to 8 (the 3-channel stem) so the implicit-GEMM
+            # gather path applies — cheaper than materializing im2col
+            # (~0.5 GB/step on resnet-50 bs128). F.pad is differentiable, so
+            # dw slices back through autograd.
+            padc = 8 - (C % 8)
+            x = F.pad(x, (0, 0, 0, 0, 0, padc)).contiguous(
+                memory_format=torch.channels_last
+            )
+            w = F.pad(w, (0, 0, 0, 0, 0, padc)).contiguous(
+                memory_format=torch.channels_last
+            )
         return _ConvNHWC.apply(x, w, stride, padding)
     return F.conv2d(x, w, None, stride, padding)
 
