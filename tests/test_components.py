@@ -1,0 +1,186 @@
+"""Per-layer component tests (reference tests/python/unittest: test_optimizer,
+test_io, test_metric, test_lr_scheduler, test_initializer subsets)."""
+import math
+import os
+
+import numpy as np
+import pytest
+import torch
+
+import dtmx
+from dtmx import initializer, lr_scheduler, metric
+from dtmx.io import CSVIter, DataBatch, NDArrayIter, ResizeIter, SyntheticDataIter
+from dtmx.optimizer import SGD, Adam, LBSGD, create as opt_create, get_updater
+
+
+# ------------------------------------------------------------- optimizer
+
+def test_sgd_momentum_reference_math():
+    w = torch.ones(4)
+    g = torch.full((4,), 2.0)
+    opt = SGD(learning_rate=0.1, momentum=0.9, rescale_grad=0.5, wd=0.01)
+    st = opt.create_state(0, w)
+    opt.update(0, w, g, st)
+    # g' = 2*0.5 + 0.01*1 = 1.01; mom = -0.1*1.01; w = 1 - 0.101
+    assert torch.allclose(w, torch.full((4,), 1 - 0.101), atol=1e-6)
+    opt.update(0, w, g, st)
+    mom2 = 0.9 * -0.101 - 0.1 * (1.0 + 0.01 * w[0].item() + 0.0)
+    # just sanity: momentum accumulates (larger step)
+    assert w[0] < 1 - 2 * 0.101
+
+
+def test_sgd_multi_precision_master_weights():
+    w = torch.randn(8).bfloat16()
+    opt = SGD(learning_rate=0.1, momentum=0.9, multi_precision=True)
+    st = opt.create_state_multi_precision(0, w)
+    master, _ = st
+    assert master.dtype == torch.float32
+    g = torch.randn(8).bfloat16()
+    opt.update_multi_precision(0, w, g, st)
+    assert torch.allclose(w.float(), master, atol=0.01)
+
+
+def test_adam_step():
+    w = torch.ones(4)
+    opt = Adam(learning_rate=0.01)
+    st = opt.create_state(0, w)
+    opt.update(0, w, torch.ones(4), st)
+    assert (w < 1).all()
+
+
+def test_clip_gradient():
+    w = torch.zeros(2)
+    opt = SGD(learning_rate=1.0, clip_gradient=0.1)
+    opt.update(0, w, torch.tensor([10.0, -10.0]), None)
+    assert torch.allclose(w, torch.tensor([-0.1, 0.1]))
+
+
+def test_create_registry():
+    assert isinstance(opt_create("sgd"), SGD)
+    assert isinstance(opt_create("adam"), Adam)
+    assert isinstance(opt_create("lbsgd"), LBSGD)
+
+
+def test_updater_state_per_index():
+    opt = SGD(learning_rate=0.1, momentum=0.9)
+    upd = get_updater(opt)
+    w0, w1 = torch.ones(2), torch.ones(3)
+    upd(0, torch.ones(2), w0)
+    upd(1, torch.ones(3), w1)
+    assert set(upd.get_states().keys()) == {0, 1}
+
+
+# ------------------------------------------------------------------- io
+
+def test_ndarrayiter_sharding():
+    data = np.arange(100, dtype=np.float32).reshape(100, 1)
+    it0 = NDArrayIter({"data": data}, batch_size=10, part_index=0, num_parts=2,
+                      shuffle=False)
+    it1 = NDArrayIter({"data": data}, batch_size=10, part_index=1, num_parts=2,
+                      shuffle=False)
+    b0 = it0.next().data[0]
+    b1 = it1.next().data[0]
+    assert b0[0, 0] == 0 and b1[0, 0] == 50  # disjoint shards
+
+
+def test_ndarrayiter_pad_last_batch():
+    data = np.arange(25, dtype=np.float32).reshape(25, 1)
+    it = NDArrayIter({"data": data}, batch_size=10, shuffle=False)
+    batches = list(it)
+    assert len(batches) == 3
+    assert batches[2].pad == 5
+
+
+def test_resize_iter_caps_epoch():
+    it = ResizeIter(SyntheticDataIter(10, (4, 8), max_iter=100), size=7)
+    assert len(list(it)) == 7
+    it.reset()
+    assert len(list(it)) == 7
+
+
+def test_csv_iter(tmp_path):
+    f = str(tmp_path / "d.csv")
+    np.savetxt(f, np.arange(12).reshape(4, 3), delimiter=",")
+    lf = str(tmp_path / "l.csv")
+    np.savetxt(lf, np.arange(4), delimiter=",")
+    it = CSVIter(f, (3,), lf, batch_size=2)
+    b = it.next()
+    assert b.data[0].shape == (2, 3)
+
+
+# --------------------------------------------------------------- metric
+
+def test_accuracy_and_topk():
+    preds = torch.tensor([[0.1, 0.9], [0.8, 0.2]])
+    labels = torch.tensor([1.0, 1.0])
+    acc = metric.create("acc")
+    acc.update([labels], [preds])
+    assert dict(acc.get_name_value())["accuracy"] == 0.5
+    topk = metric.TopKAccuracy(2)
+    topk.update([labels], [preds])
+    assert dict(topk.get_name_value())["top_k_accuracy_2"] == 1.0
+
+
+def test_composite_metric():
+    m = metric.create(["acc", "ce"])
+    preds = torch.tensor([[0.2, 0.8]])
+    m.update([torch.tensor([1.0])], [preds])
+    names, values = m.get()
+    assert "accuracy" in names and "cross-entropy" in names
+
+
+# ---------------------------------------------------------- lr scheduler
+
+def test_multifactor():
+    s = lr_scheduler.MultiFactorScheduler([10, 20], factor=0.1)
+    s.base_lr = 1.0
+    assert s(5) == 1.0
+    assert s(15) == pytest.approx(0.1)
+    assert s(25) == pytest.approx(0.01)
+
+
+def test_warmup_rearm():
+    s = lr_scheduler.WarmupScheduler(1.0, warmup_steps=10)
+    assert s(0) == 0.0
+    assert s(5) == pytest.approx(0.5)
+    assert s(10) == 1.0
+    s.rearm(100)  # worker join: ramp again (dynamic minibatch paper)
+    assert s(105) == pytest.approx(0.5)
+
+
+# ----------------------------------------------------------- initializer
+
+def test_xavier_scale():
+    w = torch.empty(64, 64)
+    initializer.Xavier(rnd_type="uniform", factor_type="avg", magnitude=3)("fc_weight", w)
+    bound = math.sqrt(3.0 / 64)
+    assert w.abs().max() <= bound + 1e-6
+
+
+def test_name_based_dispatch():
+    ini = initializer.create("default")
+    b = torch.randn(4)
+    ini("fc1_bias", b)
+    assert torch.equal(b, torch.zeros(4))
+    g = torch.randn(4)
+    ini("bn0_gamma", g)
+    assert torch.equal(g, torch.ones(4))
+
+
+# ----------------------------------------------------- optimizer states io
+
+def test_kvstore_optimizer_states_roundtrip(tmp_path):
+    kv = dtmx.kvstore.create("local")
+    kv.set_optimizer(SGD(learning_rate=0.1, momentum=0.9))
+    kv.init("w", torch.ones(4))
+    kv.push("w", torch.ones(4))
+    f = str(tmp_path / "opt.states")
+    kv.save_optimizer_states(f)
+    kv2 = dtmx.kvstore.create("local")
+    kv2.set_optimizer(SGD(learning_rate=0.1, momentum=0.9))
+    kv2.load_optimizer_states(f)
+    s1 = kv._updater.get_states()
+    s2 = kv2._updater.get_states()
+    assert set(s1) == set(s2)
+    for k in s1:
+        assert torch.allclose(s1[k], s2[k])
