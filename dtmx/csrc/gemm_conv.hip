@@ -26,15 +26,25 @@ namespace dtmx {
 // A provider returns the global address of the 16-B piece holding bf16
 // elements [k8*8, k8*8+8) of logical row m, or `zero` when out of range.
 
+// Providers expose a per-row context (Row) computed ONCE per block — the
+// block's rows are fixed across the whole K loop, so the (n,p,q)/(n,h,w)
+// FastDiv decode does not belong in the per-piece staging path (measured
+// gather overhead vs same-shape dense GEMM ~30-40%).
 struct DenseP {
   const __bf16* base;
   const __bf16* zero;
   uint32_t M, K;  // rows, k extent (elements, multiple of 8)
   uint32_t ld;    // row stride in elements
-  __device__ __forceinline__ const void* addr(uint32_t m, uint32_t k8) const {
+  struct Row {
+    const __bf16* p;  // nullptr = out-of-range row
+  };
+  __device__ __forceinline__ Row row(uint32_t m) const {
+    return {m < M ? base + (size_t)m * ld : nullptr};
+  }
+  __device__ __forceinline__ const void* addr(const Row& r, uint32_t k8) const {
     uint32_t k = k8 * 8;
-    if (m >= M || k >= K) return zero;
-    return base + (size_t)m * ld + k;
+    if (!r.p || k >= K) return zero;
+    return r.p + k;
   }
 };
 
@@ -46,17 +56,30 @@ struct ConvFwdA {
   uint32_t C, H, W, Q, S;
   int u, v, ph, pw;              // stride, padding
   FastDiv dQ, dPQ, dC, dS;
-  __device__ __forceinline__ const void* addr(uint32_t m, uint32_t k8) const {
-    uint32_t k = k8 * 8;
-    if (m >= M || k >= Ktot) return zero;
+  struct Row {
+    const __bf16* pixel0;  // &x[n][p*u-ph][q*v-pw][0] (may point out of range)
+    int ih0, iw0;          // p*u-ph, q*v-pw
+    uint32_t valid;
+  };
+  __device__ __forceinline__ Row row(uint32_t m) const {
+    Row r{};
+    if (m >= M) return r;
     uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
     uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
+    r.ih0 = (int)(p * u) - ph;
+    r.iw0 = (int)(q * v) - pw;
+    r.pixel0 = x + (((int64_t)n * H + r.ih0) * W + r.iw0) * (int64_t)C;
+    r.valid = 1;
+    return r;
+  }
+  __device__ __forceinline__ const void* addr(const Row& rc, uint32_t k8) const {
+    uint32_t k = k8 * 8;
+    if (!rc.valid || k >= Ktot) return zero;
     uint32_t rs = dC.div(k), c = dC.mod(k, rs);
     uint32_t r = dS.div(rs), s = dS.mod(rs, r);
-    int ih = (int)(p * u) - ph + (int)r;
-    int iw = (int)(q * v) - pw + (int)s;
-    if ((uint32_t)ih >= H || (uint32_t)iw >= W) return zero;
-    return x + (((size_t)n * H + ih) * W + iw) * C + c;
+    if ((uint32_t)(rc.ih0 + (int)r) >= H || (uint32_t)(rc.iw0 + (int)s) >= W)
+      return zero;
+    return rc.pixel0 + ((int64_t)r * W + s) * C + c;
   }
 };
 
@@ -69,21 +92,35 @@ struct ConvDgradA {
   uint32_t Ko, H, W, P, Q, S;
   int u, v, ph, pw;
   FastDiv dW_, dHW, dKo, dS;
-  __device__ __forceinline__ const void* addr(uint32_t m, uint32_t k8) const {
-    uint32_t k = k8 * 8;
-    if (m >= M || k >= Ktot) return zero;
+  struct Row {
+    const __bf16* base_n;  // &dy[n][0][0][0]
+    int hp, wp;            // h+ph, w+pw
+    uint32_t valid;
+  };
+  __device__ __forceinline__ Row row(uint32_t m) const {
+    Row r{};
+    if (m >= M) return r;
     uint32_t n = dHW.div(m), hw = dHW.mod(m, n);
     uint32_t h = dW_.div(hw), w = dW_.mod(hw, h);
+    r.hp = (int)h + ph;
+    r.wp = (int)w + pw;
+    r.base_n = dy + (size_t)n * P * Q * Ko;
+    r.valid = 1;
+    return r;
+  }
+  __device__ __forceinline__ const void* addr(const Row& rc, uint32_t k8) const {
+    uint32_t k = k8 * 8;
+    if (!rc.valid || k >= Ktot) return zero;
     uint32_t rs = dKo.div(k), ko = dKo.mod(k, rs);
     uint32_t r = dS.div(rs), s = dS.mod(rs, r);
-    int hp = (int)h + ph - (int)r;
-    int wp = (int)w + pw - (int)s;
+    int hp = rc.hp - (int)r;
+    int wp = rc.wp - (int)s;
     if (hp < 0 || wp < 0) return zero;
     uint32_t p = (uint32_t)hp, q = (uint32_t)wp;
     if (u > 1) { if (hp % u) return zero; p = hp / u; }
     if (v > 1) { if (wp % v) return zero; q = wp / v; }
     if (p >= P || q >= Q) return zero;
-    return dy + (((size_t)n * P + p) * Q + q) * Ko + ko;
+    return rc.base_n + ((size_t)p * Q + q) * Ko + ko;
   }
 };
 
@@ -209,15 +246,22 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     uint32_t colb = (t & 7) * 16;
     sk8[it] = (colb ^ ((srow[it] & 7) << 4)) >> 4;
   }
+  // per-block row contexts: the expensive (n,p,q)/(n,h,w) decode runs once
+  typename PA::Row arow[4];
+  typename PB::Row brow[NJ];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) arow[it] = pa.row(bm + srow[it]);
+#pragma unroll
+  for (int it = 0; it < NJ; ++it) brow[it] = pb.row(bn + srow[it]);
 
   auto stage = [&](int buf, uint32_t kt) {
 #pragma unroll
     for (int it = 0; it < 4; ++it)
-      glds16(pa.addr(bm + srow[it], kt * 8 + sk8[it]),
+      glds16(pa.addr(arow[it], kt * 8 + sk8[it]),
              &smem[buf][it * 2048 + wave * 512]);
 #pragma unroll
     for (int it = 0; it < NJ; ++it)
-      glds16(pb.addr(bn + srow[it], kt * 8 + sk8[it]),
+      glds16(pb.addr(brow[it], kt * 8 + sk8[it]),
              &smem[buf][B_OFF + it * 2048 + wave * 512]);
   };
 
@@ -685,8 +729,11 @@ static bool smallgrid_splitk(const PA& pa, const PB& pb, at::Tensor& out_bf16,
                              uint32_t M, uint32_t N, uint32_t K) {
   uint32_t tiles = ceil_div(M, 128) * ceil_div(N, N <= 64 ? 64 : 128);
   uint32_t ktiles = ceil_div(K, 64);
-  if (tiles >= 512 || ktiles < 2) return false;
-  uint32_t splitk = std::min(ktiles, std::max<uint32_t>(2, 1024 / tiles));
+  // only for severely underfilled grids: the fp32-atomic + cast overhead and
+  // the shortened per-block K loop beat plain launch only below ~160 blocks
+  // (measured: 392-block l3 conv lost 40%, 196-block l4 was a wash).
+  if (tiles >= 160 || ktiles < 8) return false;
+  uint32_t splitk = std::min(ktiles / 4, std::max<uint32_t>(2, 1024 / tiles));
   auto acc = at::zeros({(long)M, (long)N}, out_bf16.options().dtype(at::kFloat));
   EpiAtomicF32 epi{acc.data_ptr<float>(), M, N};
   launch_gemm(pa, pb, epi, M, N, K, splitk);
