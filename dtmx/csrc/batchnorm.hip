@@ -195,18 +195,22 @@ __global__ void bn_infer_prep_kernel(const __bf16* __restrict__ gamma,
 // ---- apply: y = x*scale + shift (+relu), bf16x8, 32-bit indexing ---------
 __global__ void bn_apply_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
                                 const float* __restrict__ scale,
-                                const float* __restrict__ shift, uint32_t total8,
-                                FastDiv dcv, int relu) {
+                                const float* __restrict__ shift,
+                                const __bf16* __restrict__ residual,  // nullable
+                                uint32_t total8, FastDiv dcv, int relu) {
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t stride = gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
     bf16x8 v = *(const bf16x8*)(x + (size_t)i * 8);
+    bf16x8 res;
+    if (residual) res = *(const bf16x8*)(residual + (size_t)i * 8);
     uint32_t q = dcv.div(i);
     uint32_t c0 = dcv.mod(i, q) * 8;
     bf16x8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float r = (float)v[e] * scale[c0 + e] + shift[c0 + e];
+      if (residual) r += (float)res[e];
       if (relu) r = fmaxf(r, 0.f);
       o[e] = (__bf16)r;
     }
@@ -295,8 +299,10 @@ __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ x,
                                  const __bf16* __restrict__ gamma,
                                  const float* __restrict__ pdb,
                                  const float* __restrict__ pdg,
-                                 __bf16* __restrict__ dx, uint32_t total8,
-                                 FastDiv dcv, float inv_count, int relu) {
+                                 __bf16* __restrict__ dx,
+                                 __bf16* __restrict__ dres,  // nullable
+                                 uint32_t total8, FastDiv dcv, float inv_count,
+                                 int relu) {
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t stride = gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
@@ -307,18 +313,20 @@ __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ x,
     if (relu) yv = *(const bf16x8*)(y + off);
     uint32_t q = dcv.div(i);
     uint32_t c0 = dcv.mod(i, q) * 8;
-    bf16x8 o;
+    bf16x8 o, om;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       uint32_t c = c0 + e;
       float g = (float)gv[e];
       if (relu && (float)yv[e] <= 0.f) g = 0.f;
+      if (dres) om[e] = (__bf16)g;
       float invstd = save_invstd[c];
       float xh = ((float)xv[e] - save_mean[c]) * invstd;
       o[e] = (__bf16)((float)gamma[c] * invstd *
                       (g - pdb[c] * inv_count - xh * pdg[c] * inv_count));
     }
     *(bf16x8*)(dx + off) = o;
+    if (dres) *(bf16x8*)(dres + off) = om;
   }
 }
 
@@ -345,7 +353,8 @@ static void bn_grid(uint32_t rows, uint32_t cvecs, dim3& grid,
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamma,
                                      const at::Tensor& beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
-                                     double eps, bool fuse_relu) {
+                                     double eps, bool fuse_relu,
+                                     const c10::optional<at::Tensor>& residual) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
@@ -380,17 +389,18 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
   FastDiv dcv;
   dcv.init(cvecs);
   uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-  bn_apply_kernel<<<blocks, 256, 0, s>>>((const __bf16*)x.data_ptr(),
-                                         (__bf16*)y.data_ptr(),
-                                         scale.data_ptr<float>(),
-                                         shift.data_ptr<float>(), total8, dcv,
-                                         fuse_relu ? 1 : 0);
+  bn_apply_kernel<<<blocks, 256, 0, s>>>(
+      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
+      scale.data_ptr<float>(), shift.data_ptr<float>(),
+      residual.has_value() ? (const __bf16*)residual->data_ptr() : nullptr,
+      total8, dcv, fuse_relu ? 1 : 0);
   return {y, save_mean, save_invstd};
 }
 
 at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
                         const at::Tensor& beta, const at::Tensor& running_mean,
-                        const at::Tensor& running_var, double eps, bool fuse_relu) {
+                        const at::Tensor& running_var, double eps, bool fuse_relu,
+                        const c10::optional<at::Tensor>& residual) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8");
@@ -407,18 +417,18 @@ at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
   FastDiv dcv;
   dcv.init(cvecs);
   uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-  bn_apply_kernel<<<blocks, 256, 0, s>>>((const __bf16*)x.data_ptr(),
-                                         (__bf16*)y.data_ptr(),
-                                         scale.data_ptr<float>(),
-                                         shift.data_ptr<float>(), total8, dcv,
-                                         fuse_relu ? 1 : 0);
+  bn_apply_kernel<<<blocks, 256, 0, s>>>(
+      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
+      scale.data_ptr<float>(), shift.data_ptr<float>(),
+      residual.has_value() ? (const __bf16*)residual->data_ptr() : nullptr,
+      total8, dcv, fuse_relu ? 1 : 0);
   return y;
 }
 
 std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
                                const at::Tensor& gamma, const at::Tensor& save_mean,
                                const at::Tensor& save_invstd, bool fuse_relu,
-                               const at::Tensor& y) {
+                               const at::Tensor& y, bool want_dres) {
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
   uint32_t rows = N * H * W, cvecs = C / 8;
@@ -433,6 +443,8 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   auto dgamma = at::empty({(long)C}, x.options());
   auto dbeta = at::empty({(long)C}, x.options());
   auto dx = at::empty_like(x);
+  at::Tensor dres;
+  if (want_dres) dres = at::empty_like(x);
   auto s = bn_stream();
   bn_bwd_stats_kernel<<<grid, 256, 0, s>>>(
       (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
@@ -457,7 +469,9 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), (const __bf16*)gamma.data_ptr(),
       tdb.data_ptr<float>(), tdg.data_ptr<float>(), (__bf16*)dx.data_ptr(),
-      total8, dcv, 1.f / rows, fuse_relu ? 1 : 0);
+      want_dres ? (__bf16*)dres.data_ptr() : nullptr, total8, dcv, 1.f / rows,
+      fuse_relu ? 1 : 0);
+  if (want_dres) return {dx, dgamma, dbeta, dres};
   return {dx, dgamma, dbeta};
 }
 

@@ -12,12 +12,14 @@ at::Tensor conv_wgrad(const at::Tensor&, const at::Tensor&, long, long, long, lo
 // batchnorm.hip
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor&, const at::Tensor&,
                                      const at::Tensor&, at::Tensor, at::Tensor,
-                                     double, double, bool);
+                                     double, double, bool,
+                                     const c10::optional<at::Tensor>&);
 at::Tensor bn_fwd_infer(const at::Tensor&, const at::Tensor&, const at::Tensor&,
-                        const at::Tensor&, const at::Tensor&, double, bool);
+                        const at::Tensor&, const at::Tensor&, double, bool,
+                        const c10::optional<at::Tensor>&);
 std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&,
-                               const at::Tensor&, bool, const at::Tensor&);
+                               const at::Tensor&, bool, const at::Tensor&, bool);
 // pool_elem.hip
 std::vector<at::Tensor> maxpool_fwd(const at::Tensor&, long, long, long);
 at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, long, long, long,

@@ -28,19 +28,17 @@ class BasicBlock(nn.Module):
         self.conv1 = Conv2dNHWC(in_ch, ch, 3, stride, 1)
         self.bn1 = BatchNorm2dNHWC(ch, fuse_relu=True)
         self.conv2 = Conv2dNHWC(ch, ch, 3, 1, 1)
-        self.bn2 = BatchNorm2dNHWC(ch)
+        self.bn2 = BatchNorm2dNHWC(ch, fuse_relu=True)  # fused +residual +relu
         self.downsample = None
         if stride != 1 or in_ch != ch:
             self.downsample = nn.Sequential(
                 Conv2dNHWC(in_ch, ch, 1, stride, 0), BatchNorm2dNHWC(ch)
             )
-        self.add_relu = AddRelu()
 
     def forward(self, x):
         sc = x if self.downsample is None else self.downsample(x)
         y = self.bn1(self.conv1(x))
-        y = self.bn2(self.conv2(y))
-        return self.add_relu(y, sc)
+        return self.bn2(self.conv2(y), sc)
 
 
 class Bottleneck(nn.Module):
@@ -54,20 +52,18 @@ class Bottleneck(nn.Module):
         self.conv2 = Conv2dNHWC(ch, ch, 3, stride, 1)  # v1.5: stride on the 3x3
         self.bn2 = BatchNorm2dNHWC(ch, fuse_relu=True)
         self.conv3 = Conv2dNHWC(ch, out_ch, 1, 1, 0)
-        self.bn3 = BatchNorm2dNHWC(out_ch)
+        self.bn3 = BatchNorm2dNHWC(out_ch, fuse_relu=True)  # fused +residual +relu
         self.downsample = None
         if stride != 1 or in_ch != out_ch:
             self.downsample = nn.Sequential(
                 Conv2dNHWC(in_ch, out_ch, 1, stride, 0), BatchNorm2dNHWC(out_ch)
             )
-        self.add_relu = AddRelu()
 
     def forward(self, x):
         sc = x if self.downsample is None else self.downsample(x)
         y = self.bn1(self.conv1(x))
         y = self.bn2(self.conv2(y))
-        y = self.bn3(self.conv3(y))
-        return self.add_relu(y, sc)
+        return self.bn3(self.conv3(y), sc)
 
 
 _CONFIGS = {

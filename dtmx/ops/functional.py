@@ -76,16 +76,19 @@ def conv2d(x, w, stride: int = 1, padding: int = 0):
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum,
-                eps, fuse_relu):
+                eps, fuse_relu, residual):
         ext = require_ext()
         if training:
             y, save_mean, save_invstd = ext.bn_fwd_train(
-                x, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu
+                x, gamma, beta, running_mean, running_var, momentum, eps,
+                fuse_relu, residual
             )
             ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
             ctx.fuse_relu = fuse_relu
+            ctx.has_residual = residual is not None
         else:
-            y = ext.bn_fwd_infer(x, gamma, beta, running_mean, running_var, eps, fuse_relu)
+            y = ext.bn_fwd_infer(x, gamma, beta, running_mean, running_var, eps,
+                                 fuse_relu, residual)
         return y
 
     @staticmethod
@@ -93,21 +96,28 @@ class _BatchNormNHWC(torch.autograd.Function):
         ext = require_ext()
         x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
-        dx, dgamma, dbeta = ext.bn_bwd(x, dy, gamma, save_mean, save_invstd,
-                                       ctx.fuse_relu, y)
-        return dx, dgamma, dbeta, None, None, None, None, None, None
+        out = ext.bn_bwd(x, dy, gamma, save_mean, save_invstd,
+                         ctx.fuse_relu, y, ctx.has_residual)
+        dres = out[3] if ctx.has_residual else None
+        return (out[0], out[1], out[2], None, None, None, None, None, None, dres)
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
-               momentum: float = 0.9, eps: float = 1e-5, fuse_relu: bool = False):
+               momentum: float = 0.9, eps: float = 1e-5, fuse_relu: bool = False,
+               residual=None):
     """NHWC batch norm. `momentum` follows mxnet semantics:
     moving = moving*momentum + batch*(1-momentum) (reference batch_norm-inl.h),
-    i.e. torch's momentum is (1 - mxnet momentum)."""
+    i.e. torch's momentum is (1 - mxnet momentum).
+
+    `residual`: fused y = [relu](bn(x) + residual) — the resnet block tail in
+    one pass (bn_apply's extra read beats a separate add_relu round trip)."""
     if _use_hip(x):
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
-                                    training, momentum, eps, fuse_relu)
+                                    training, momentum, eps, fuse_relu, residual)
     y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
                      1.0 - momentum, eps)
+    if residual is not None:
+        y = y + residual
     return F.relu(y) if fuse_relu else y
 
 

@@ -63,7 +63,7 @@ class BatchNorm2dNHWC(nn.Module):
         self.running_var.data = self.running_var.data.float()
         return self
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         return DF.batch_norm(
             x,
             self.weight,
@@ -74,6 +74,7 @@ class BatchNorm2dNHWC(nn.Module):
             self.momentum,
             self.eps,
             self.fuse_relu,
+            residual,
         )
 
 

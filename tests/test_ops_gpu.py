@@ -132,7 +132,7 @@ def test_bn_fwd_train_and_bwd():
     rm = torch.zeros(C, dtype=torch.float32, device=DEV)
     rv = torch.ones(C, dtype=torch.float32, device=DEV)
     y, sm, si = ext.bn_fwd_train(nhwc(x), gamma.to(DEV), beta.to(DEV), rm, rv,
-                                 0.9, 1e-5, False)
+                                 0.9, 1e-5, False, None)
     xf = x.float().requires_grad_(True)
     ref = F.batch_norm(xf, None, None, gamma.float(), beta.float(), True, 0.1, 1e-5)
     assert_close(y.contiguous(), ref, name="bn_fwd")
@@ -142,7 +142,7 @@ def test_bn_fwd_train_and_bwd():
     dy = mk((N, C, H, W), seed=8)
     ref.backward(dy.float())
     dx, dgamma, dbeta = ext.bn_bwd(nhwc(x), nhwc(dy), gamma.to(DEV), sm, si,
-                                   False, y)
+                                   False, y, False)
     assert_close(dx.contiguous(), xf.grad, rtol=0.05, name="bn_dx")
     xhat = (x.float() - x.float().mean(dim=(0, 2, 3), keepdim=True)) / (
         x.float().var(dim=(0, 2, 3), unbiased=False, keepdim=True) + 1e-5).sqrt()
@@ -160,10 +160,45 @@ def test_bn_fused_relu():
     rm = torch.zeros(C, dtype=torch.float32, device=DEV)
     rv = torch.ones(C, dtype=torch.float32, device=DEV)
     y, sm, si = ext.bn_fwd_train(nhwc(x), gamma.to(DEV), beta.to(DEV), rm, rv,
-                                 0.9, 1e-5, True)
+                                 0.9, 1e-5, True, None)
     ref = F.relu(F.batch_norm(x.float(), None, None, gamma.float(), beta.float(),
                               True, 0.1, 1e-5))
     assert_close(y.contiguous(), ref, name="bn_relu")
+
+
+def test_bn_fused_residual_full_autograd():
+    """y = relu(bn(x) + res): forward + all grads vs torch fp32 reference."""
+    from dtmx.ops import functional as DF
+
+    N, C, H, W = 4, 64, 14, 14
+    x = mk((N, C, H, W), seed=21)
+    res = mk((N, C, H, W), seed=22)
+    gamma = (torch.randn(C) * 0.1 + 1).bfloat16()
+    beta = (torch.randn(C) * 0.1).bfloat16()
+    dy = mk((N, C, H, W), seed=23)
+
+    xd = nhwc(x).requires_grad_(True)
+    resd = nhwc(res).requires_grad_(True)
+    gd = gamma.to(DEV).requires_grad_(True)
+    bd = beta.to(DEV).requires_grad_(True)
+    rm = torch.zeros(C, dtype=torch.float32, device=DEV)
+    rv = torch.ones(C, dtype=torch.float32, device=DEV)
+    y = DF.batch_norm(xd, gd, bd, rm, rv, True, 0.9, 1e-5, True, resd)
+    y.backward(nhwc(dy))
+
+    xf = x.float().requires_grad_(True)
+    rf = res.float().requires_grad_(True)
+    gf = gamma.float().requires_grad_(True)
+    bf_ = beta.float().requires_grad_(True)
+    ref = torch.nn.functional.relu(
+        torch.nn.functional.batch_norm(xf, None, None, gf, bf_, True, 0.1, 1e-5) + rf
+    )
+    ref.backward(dy.float())
+    assert_close(y.contiguous(), ref, name="bnres_y")
+    assert_close(xd.grad.contiguous(), xf.grad, rtol=0.05, name="bnres_dx")
+    assert_close(resd.grad.contiguous(), rf.grad, rtol=0.05, name="bnres_dres")
+    assert_close(gd.grad, gf.grad, rtol=0.05, name="bnres_dgamma")
+    assert_close(bd.grad, bf_.grad, rtol=0.05, name="bnres_dbeta")
 
 
 def test_bn_infer():
@@ -176,7 +211,7 @@ def test_bn_infer():
     rm = torch.randn(C, dtype=torch.float32)
     rv = torch.rand(C, dtype=torch.float32) + 0.5
     y = ext.bn_fwd_infer(nhwc(x), gamma.to(DEV), beta.to(DEV), rm.to(DEV),
-                         rv.to(DEV), 1e-5, False)
+                         rv.to(DEV), 1e-5, False, None)
     ref = F.batch_norm(x.float(), rm, rv, gamma.float(), beta.float(), False, 0.1, 1e-5)
     assert_close(y.contiguous(), ref, name="bn_infer")
 
