@@ -84,6 +84,116 @@ __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict_
   block_col_reduce(red, ss, cv, cvecs, psumsq + (size_t)blockIdx.y * C);
 }
 
+// device-side body shared by the fused reduce+finalize kernels: reduces the
+// two [nslabs][C] slabs for this thread's channel-vector into sa/sb (valid on
+// threads t < ncv after the final barrier).
+__device__ __forceinline__ void slab_reduce2_body(const float* a, const float* b,
+                                                  uint32_t C, uint32_t nslabs,
+                                                  float (&outa)[8], float (&outb)[8],
+                                                  uint32_t& cv_out, uint32_t& ncv_out) {
+  const uint32_t cvecs = C / 8;
+  const uint32_t ncv = min(cvecs, 8u);
+  const uint32_t lanes = blockDim.x / ncv;
+  const uint32_t t = threadIdx.x;
+  const uint32_t cv = blockIdx.x * ncv + t % ncv;
+  const uint32_t lane = t / ncv;
+  cv_out = cv;
+  ncv_out = ncv;
+  float sa[8] = {}, sb[8] = {};
+  if (cv < cvecs) {
+    for (uint32_t sl = lane; sl < nslabs; sl += lanes) {
+      const float* pa = a + (size_t)sl * C + cv * 8;
+      const float* pb = b + (size_t)sl * C + cv * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        sa[e] += pa[e];
+        sb[e] += pb[e];
+      }
+    }
+  }
+  __shared__ float red[256 * 8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[t * 8 + e] = sa[e];
+  __syncthreads();
+  for (uint32_t off = 128; off >= ncv; off >>= 1) {
+    if (t < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (t < ncv) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) outa[e] = red[t * 8 + e];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[t * 8 + e] = sb[e];
+  __syncthreads();
+  for (uint32_t off = 128; off >= ncv; off >>= 1) {
+    if (t < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (t < ncv) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) outb[e] = red[t * 8 + e];
+  }
+}
+
+// fused: slab reduce + forward finalize (mean/invstd, running stats, scale/shift)
+__global__ void bn_reduce_finalize_kernel(
+    const float* __restrict__ psum, const float* __restrict__ psumsq,
+    const __bf16* __restrict__ gamma, const __bf16* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ save_mean, float* __restrict__ save_invstd,
+    float* __restrict__ scale, float* __restrict__ shift, uint32_t C,
+    uint32_t nslabs, uint32_t count, float momentum, float eps) {
+  float fsum[8], fsumsq[8];
+  uint32_t cv, ncv;
+  slab_reduce2_body(psum, psumsq, C, nslabs, fsum, fsumsq, cv, ncv);
+  if (threadIdx.x < ncv && cv < C / 8) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      uint32_t c = cv * 8 + e;
+      float mean = fsum[e] / count;
+      float var = fmaxf(fsumsq[e] / count - mean * mean, 0.f);
+      float invstd = rsqrtf(var + eps);
+      save_mean[c] = mean;
+      save_invstd[c] = invstd;
+      float unbiased = count > 1 ? var * count / (count - 1) : var;
+      running_mean[c] = running_mean[c] * momentum + mean * (1.f - momentum);
+      running_var[c] = running_var[c] * momentum + unbiased * (1.f - momentum);
+      float g = (float)gamma[c];
+      scale[c] = g * invstd;
+      shift[c] = (float)beta[c] - mean * g * invstd;
+    }
+  }
+}
+
+// fused: slab reduce + backward finalize (dgamma/dbeta + totals for dx)
+__global__ void bn_bwd_reduce_finalize_kernel(
+    const float* __restrict__ pdb, const float* __restrict__ pdg,
+    __bf16* __restrict__ dgamma, __bf16* __restrict__ dbeta,
+    float* __restrict__ tdb, float* __restrict__ tdg, uint32_t C,
+    uint32_t nslabs) {
+  float db[8], dg[8];
+  uint32_t cv, ncv;
+  slab_reduce2_body(pdb, pdg, C, nslabs, db, dg, cv, ncv);
+  if (threadIdx.x < ncv && cv < C / 8) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      uint32_t c = cv * 8 + e;
+      dgamma[c] = (__bf16)dg[e];
+      dbeta[c] = (__bf16)db[e];
+      tdb[c] = db[e];
+      tdg[c] = dg[e];
+    }
+  }
+}
+
 // ---- parallel slab reduction: [nslabs][C] x2 -> [C] x2 -------------------
 // block covers ncv channel-vectors (<=8) x (256/ncv) slab lanes; lane-strided
 // accumulate then LDS tree. Replaces both the per-channel serial loop
@@ -373,17 +483,13 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
   bn_stats_kernel<<<grid, 256, 0, s>>>((const __bf16*)x.data_ptr(),
                                        psum.data_ptr<float>(),
                                        psumsq.data_ptr<float>(), rows, cvecs, rpb);
-  auto tsum = at::empty({(long)C}, opt_f), tsumsq = at::empty({(long)C}, opt_f);
   uint32_t ncv = std::min(cvecs, 8u);
-  slab_reduce2_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
-      psum.data_ptr<float>(), psumsq.data_ptr<float>(), tsum.data_ptr<float>(),
-      tsumsq.data_ptr<float>(), C, nslabs);
-  bn_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
-      tsum.data_ptr<float>(), tsumsq.data_ptr<float>(),
+  bn_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+      psum.data_ptr<float>(), psumsq.data_ptr<float>(),
       (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(), C, 1, rows,
+      scale.data_ptr<float>(), shift.data_ptr<float>(), C, nslabs, rows,
       momentum, eps);
   uint32_t total8 = rows * cvecs;
   FastDiv dcv;
@@ -451,15 +557,11 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
       pdg.data_ptr<float>(), rows, cvecs, rpb, fuse_relu ? 1 : 0);
-  auto rdb = at::empty({(long)C}, opt_f), rdg = at::empty({(long)C}, opt_f);
   uint32_t ncv = std::min(cvecs, 8u);
-  slab_reduce2_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
-      pdb.data_ptr<float>(), pdg.data_ptr<float>(), rdb.data_ptr<float>(),
-      rdg.data_ptr<float>(), C, nslabs);
-  bn_bwd_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
-      rdb.data_ptr<float>(), rdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
+  bn_bwd_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+      pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
       (__bf16*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
-      C, 1);
+      C, nslabs);
   uint32_t total8 = rows * cvecs;
   FastDiv dcv;
   dcv.init(cvecs);

@@ -11,33 +11,46 @@ static hipStream_t pe_stream() { return at::hip::getCurrentHIPStream().stream();
 
 // ------------------------------------------------------------- max pooling
 
+// one thread = 8 channels (bf16x8 loads/stores; scalar 2-B accesses measured
+// ~10x off the bandwidth roofline on the stem maxpool)
 __global__ void maxpool_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
                                    uint8_t* __restrict__ idx, uint32_t N, uint32_t C,
                                    uint32_t H, uint32_t W, uint32_t P, uint32_t Q,
-                                   uint32_t K, int u, int pad) {
-  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-  const size_t total = (size_t)N * P * Q * C;
-  const size_t stride = (size_t)gridDim.x * blockDim.x;
+                                   uint32_t K, int u, int pad, FastDiv dCv,
+                                   FastDiv dQ_, FastDiv dPQ) {
+  const uint32_t cvecs = C / 8;
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t total = (uint32_t)(N * P * Q) * cvecs;
+  const uint32_t stride = gridDim.x * blockDim.x;
+  typedef __attribute__((ext_vector_type(8))) uint8_t u8x8;
   for (; i < total; i += stride) {
-    uint32_t c = i % C;
-    size_t m = i / C;
-    uint32_t q = m % Q;
-    uint32_t p = (m / Q) % P;
-    uint32_t n = m / ((size_t)P * Q);
-    float best = -3.4e38f;
-    uint8_t besti = 0;
+    uint32_t m = dCv.div(i), cv = dCv.mod(i, m);
+    uint32_t np = dQ_.div(m), q = dQ_.mod(m, np);
+    uint32_t n = dPQ.div(m), p = dQ_.div(dPQ.mod(m, n));
+    float best[8];
+    u8x8 besti;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) { best[e] = -3.4e38f; besti[e] = 0; }
     for (uint32_t kh = 0; kh < K; ++kh) {
       int ih = (int)(p * u) - pad + (int)kh;
       if ((uint32_t)ih >= H) continue;
       for (uint32_t kw = 0; kw < K; ++kw) {
         int iw = (int)(q * u) - pad + (int)kw;
         if ((uint32_t)iw >= W) continue;
-        float v = (float)x[(((size_t)n * H + ih) * W + iw) * C + c];
-        if (v > best) { best = v; besti = kh * K + kw; }
+        bf16x8 v = *(const bf16x8*)(x + (((size_t)n * H + ih) * W + iw) * C + cv * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = (float)v[e];
+          if (f > best[e]) { best[e] = f; besti[e] = kh * K + kw; }
+        }
       }
     }
-    y[i] = (__bf16)best;
-    idx[i] = besti;
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)best[e];
+    size_t off = (size_t)m * C + cv * 8;
+    *(bf16x8*)(y + off) = o;
+    *(u8x8*)(idx + off) = besti;
   }
 }
 
@@ -45,17 +58,17 @@ __global__ void maxpool_bwd_kernel(const __bf16* __restrict__ dy,
                                    const uint8_t* __restrict__ idx,
                                    __bf16* __restrict__ dx, uint32_t N, uint32_t C,
                                    uint32_t H, uint32_t W, uint32_t P, uint32_t Q,
-                                   uint32_t K, int u, int pad) {
-  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-  const size_t total = (size_t)N * H * W * C;
-  const size_t stride = (size_t)gridDim.x * blockDim.x;
+                                   uint32_t K, int u, int pad, FastDiv dCv,
+                                   FastDiv dW2, FastDiv dHW) {
+  const uint32_t cvecs = C / 8;
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t total = (uint32_t)(N * H * W) * cvecs;
+  const uint32_t stride = gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
-    uint32_t c = i % C;
-    size_t m = i / C;
-    uint32_t w = m % W;
-    uint32_t h = (m / W) % H;
-    uint32_t n = m / ((size_t)H * W);
-    float acc = 0.f;
+    uint32_t m = dCv.div(i), cv = dCv.mod(i, m);
+    uint32_t hw_ = dHW.mod(m, dHW.div(m));
+    uint32_t n = dHW.div(m);
+    uint32_t h = dW2.div(hw_), w = dW2.mod(hw_, h);
     // windows (p,q) that contain (h,w): p*u - pad <= h < p*u - pad + K
     int plo = ((int)h + pad - (int)K + u) / u;  // ceil((h+pad-K+1)/u)
     if (plo < 0) plo = 0;
@@ -65,17 +78,27 @@ __global__ void maxpool_bwd_kernel(const __bf16* __restrict__ dy,
     if (qlo < 0) qlo = 0;
     int qhi = ((int)w + pad) / u;
     if (qhi >= (int)Q) qhi = Q - 1;
+    typedef __attribute__((ext_vector_type(8))) uint8_t u8x8;
+    float acc8[8] = {};
     for (int p = plo; p <= phi; ++p) {
       uint32_t kh = (uint32_t)((int)h + pad - p * u);
       if (kh >= K) continue;
       for (int q = qlo; q <= qhi; ++q) {
         uint32_t kw = (uint32_t)((int)w + pad - q * u);
         if (kw >= K) continue;
-        size_t o = (((size_t)n * P + p) * Q + q) * C + c;
-        if (idx[o] == kh * K + kw) acc += (float)dy[o];
+        size_t o = (((size_t)n * P + p) * Q + q) * C + cv * 8;
+        u8x8 iv = *(const u8x8*)(idx + o);
+        bf16x8 gv = *(const bf16x8*)(dy + o);
+        uint8_t want = (uint8_t)(kh * K + kw);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (iv[e] == want) acc8[e] += (float)gv[e];
       }
     }
-    dx[i] = (__bf16)acc;
+    bf16x8 o8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o8[e] = (__bf16)acc8[e];
+    *(bf16x8*)(dx + (size_t)m * C + cv * 8) = o8;
   }
 }
 
@@ -168,10 +191,14 @@ std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, long kernel, long strid
                      at::MemoryFormat::ChannelsLast);
   auto idx = at::empty({(long)N, (long)C, (long)P, (long)Q},
                        x.options().dtype(at::kByte), at::MemoryFormat::ChannelsLast);
-  size_t total = (size_t)N * P * Q * C;
+  TORCH_CHECK(C % 8 == 0, "maxpool: C must be a multiple of 8");
+  FastDiv dCv, dQ_, dPQ;
+  dCv.init(C / 8); dQ_.init(Q); dPQ.init(P * Q);
+  size_t total = (size_t)N * P * Q * (C / 8);
   maxpool_fwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
       (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
-      (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+      (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
+      dCv, dQ_, dPQ);
   return {y, idx};
 }
 
@@ -181,10 +208,13 @@ at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx, long H,
   auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
   auto dx = at::empty({(long)N, (long)C, (long)H, (long)W}, dy.options(),
                       at::MemoryFormat::ChannelsLast);
-  size_t total = (size_t)N * H * W * C;
+  FastDiv dCv, dW2, dHW;
+  dCv.init(C / 8); dW2.init(W); dHW.init(H * W);
+  size_t total = (size_t)N * H * W * (C / 8);
   maxpool_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
       (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
-      (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+      (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
+      dCv, dW2, dHW);
   return dx;
 }
 
