@@ -392,7 +392,9 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   const uint32_t kt0 = blockIdx.y * kt_per_slice;
   const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
   if (kt0 >= ktiles_total) return;
-  __shared__ __bf16 smem[2][64 * (128 + BN)];  // [kd][m] then [kd][n]
+  // 16-B alignment: ds_read_b64_tr_b16 at a misaligned address silently
+  // returns the 8-aligned address's data (G17)
+  __shared__ __attribute__((aligned(16))) __bf16 smem[2][64 * (128 + BN)];
   constexpr uint32_t B_OFF = 64 * 128;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
@@ -448,23 +450,95 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 af[4], bfr[NJ];
-      const uint32_t kbase = kk * 32 + ((lane >> 4) << 3);
+      // hardware transpose reads: each 16-lane group reads a [4 kd][16 col]
+      // block of the contraction-major image and receives it column-wise —
+      // 2 ds_read_b64_tr_b16 per fragment instead of 8 scalar u16 reads
+      // (cdna_hip_programming.md T10; source lane r supplies the 8-B run at
+      // row kbase+r/4, cols colbase+(r%4)*4).
+      const uint32_t r_ = lane & 15;
+      const uint32_t kq = r_ >> 2, cq = (r_ & 3) * 4;
+      const uint32_t kg = kk * 32 + ((lane >> 4) << 3) + kq;
+      typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_;
+      union Frag {
+        bf16x8 f;
+        u32x2_ h[2];
+      };
+      {
+        unsigned a0, a1, a2, a3, a4, a5, a6, a7;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        uint32_t m = wr + i * 16 + (lane & 15);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          uint32_t kr = kbase + e;
-          af[i][e] = smem[cur][kr * 128 + (m ^ (((kr >> 3) & 1) << 4))];
+        for (int i = 0; i < 4; ++i) {
+          uint32_t lo = (kg)*128 + ((wr + i * 16 + cq) ^ (((kg >> 3) & 1) << 4));
+          uint32_t hi = (kg + 4) * 128 +
+                        ((wr + i * 16 + cq) ^ ((((kg + 4) >> 3) & 1) << 4));
+          unsigned alo = (unsigned)(uintptr_t)&smem[cur][lo];
+          unsigned ahi = (unsigned)(uintptr_t)&smem[cur][hi];
+          if (i == 0) { a0 = alo; a1 = ahi; }
+          if (i == 1) { a2 = alo; a3 = ahi; }
+          if (i == 2) { a4 = alo; a5 = ahi; }
+          if (i == 3) { a6 = alo; a7 = ahi; }
         }
+        u32x2_ t0, t1, t2, t3, t4, t5, t6, t7;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %8\n\t"
+            "ds_read_b64_tr_b16 %1, %9\n\t"
+            "ds_read_b64_tr_b16 %2, %10\n\t"
+            "ds_read_b64_tr_b16 %3, %11\n\t"
+            "ds_read_b64_tr_b16 %4, %12\n\t"
+            "ds_read_b64_tr_b16 %5, %13\n\t"
+            "ds_read_b64_tr_b16 %6, %14\n\t"
+            "ds_read_b64_tr_b16 %7, %15\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(t0), "=&v"(t1), "=&v"(t2), "=&v"(t3), "=&v"(t4), "=&v"(t5),
+              "=&v"(t6), "=&v"(t7)
+            : "v"(a0), "v"(a1), "v"(a2), "v"(a3), "v"(a4), "v"(a5), "v"(a6),
+              "v"(a7)
+            : "memory");
+        Frag f;
+        f.h[0] = t0; f.h[1] = t1; af[0] = f.f;
+        f.h[0] = t2; f.h[1] = t3; af[1] = f.f;
+        f.h[0] = t4; f.h[1] = t5; af[2] = f.f;
+        f.h[0] = t6; f.h[1] = t7; af[3] = f.f;
       }
+      {
+        unsigned b0, b1, b2, b3, b4, b5, b6, b7;
 #pragma unroll
-      for (int j = 0; j < NJ; ++j) {
-        uint32_t n = wc + j * 16 + (lane & 15);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          uint32_t kr = kbase + e;
-          bfr[j][e] = smem[cur][B_OFF + kr * BN + (n ^ (((kr >> 3) & 1) << 4))];
+        for (int j = 0; j < NJ; ++j) {
+          uint32_t lo = B_OFF + (kg)*BN + ((wc + j * 16 + cq) ^ (((kg >> 3) & 1) << 4));
+          uint32_t hi = B_OFF + (kg + 4) * BN +
+                        ((wc + j * 16 + cq) ^ ((((kg + 4) >> 3) & 1) << 4));
+          unsigned alo = (unsigned)(uintptr_t)&smem[cur][lo];
+          unsigned ahi = (unsigned)(uintptr_t)&smem[cur][hi];
+          if (j == 0) { b0 = alo; b1 = ahi; }
+          if (j == 1) { b2 = alo; b3 = ahi; }
+          if (NJ > 2 && j == 2) { b4 = alo; b5 = ahi; }
+          if (NJ > 2 && j == 3) { b6 = alo; b7 = ahi; }
+        }
+        u32x2_ t0, t1, t2, t3;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %4\n\t"
+            "ds_read_b64_tr_b16 %1, %5\n\t"
+            "ds_read_b64_tr_b16 %2, %6\n\t"
+            "ds_read_b64_tr_b16 %3, %7\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(t0), "=&v"(t1), "=&v"(t2), "=&v"(t3)
+            : "v"(b0), "v"(b1), "v"(b2), "v"(b3)
+            : "memory");
+        Frag f;
+        f.h[0] = t0; f.h[1] = t1; bfr[0] = f.f;
+        f.h[0] = t2; f.h[1] = t3; bfr[1] = f.f;
+        if (NJ > 2) {
+          u32x2_ t4, t5, t6, t7;
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %4\n\t"
+              "ds_read_b64_tr_b16 %1, %5\n\t"
+              "ds_read_b64_tr_b16 %2, %6\n\t"
+              "ds_read_b64_tr_b16 %3, %7\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(t4), "=&v"(t5), "=&v"(t6), "=&v"(t7)
+              : "v"(b4), "v"(b5), "v"(b6), "v"(b7)
+              : "memory");
+          f.h[0] = t4; f.h[1] = t5; bfr[2] = f.f;
+          f.h[0] = t6; f.h[1] = t7; bfr[3] = f.f;
         }
       }
       __builtin_amdgcn_s_setprio(1);
