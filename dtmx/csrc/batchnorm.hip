@@ -21,23 +21,24 @@ static hipStream_t bn_stream() { return at::hip::getCurrentHIPStream().stream();
 // a [gridDim.y][C] slab. No atomics: same-address fp32 atomicAdd serializes
 // across blocks (measured ~0.2us PER BLOCK -> stats time linear in grid).
 __device__ __forceinline__ void block_col_reduce(float* red, float (&v)[8],
-                                                 uint32_t cv, uint32_t cvecs,
-                                                 float* slab_row) {
+                                                 uint32_t cv, uint32_t cpb,
+                                                 uint32_t cvecs, float* slab_row) {
   const uint32_t t = threadIdx.x;
   __syncthreads();
 #pragma unroll
   for (int e = 0; e < 8; ++e) red[t * 8 + e] = v[e];
   __syncthreads();
-  for (uint32_t off = 128; off >= cvecs; off >>= 1) {
+  for (uint32_t off = 128; off >= cpb; off >>= 1) {
     if (t < off) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
     }
     __syncthreads();
   }
-  if (t < cvecs) {
+  if (t < cpb && blockIdx.x * cpb + t < cvecs) {
 #pragma unroll
-    for (int e = 0; e < 8; ++e) slab_row[t * 8 + e] = red[t * 8 + e];
+    for (int e = 0; e < 8; ++e)
+      slab_row[(blockIdx.x * cpb + t) * 8 + e] = red[t * 8 + e];
   }
 }
 
@@ -48,12 +49,14 @@ __device__ __forceinline__ void block_col_reduce(float* red, float (&v)[8],
 // strided variant below.
 __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ psum,
                                 float* __restrict__ psumsq, uint32_t rows,
-                                uint32_t cvecs, uint32_t rows_per_block) {
+                                uint32_t cvecs, uint32_t cpb,
+                                uint32_t rows_per_block) {
   const uint32_t t = threadIdx.x;
-  const uint32_t cv = t % cvecs;
-  const uint32_t rstep = blockDim.x / cvecs;
-  const uint32_t r0 = blockIdx.y * rows_per_block + t / cvecs;
-  const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
+  const uint32_t cv = blockIdx.x * cpb + t % cpb;
+  const bool cv_ok = cv < cvecs;
+  const uint32_t rstep = blockDim.x / cpb;
+  const uint32_t r0 = blockIdx.y * rows_per_block + t / cpb;
+  const uint32_t r1 = cv_ok ? min((blockIdx.y + 1) * rows_per_block, rows) : 0;
   const uint32_t C = cvecs * 8;
   float s[8] = {}, ss[8] = {};
   // 4x unrolled so four 16-B loads are in flight per wave (a single-buffer
@@ -80,8 +83,8 @@ __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict_
   // intra-block tree reduction, then a per-block slab row (see
   // block_col_reduce).
   __shared__ float red[256 * 8];
-  block_col_reduce(red, s, cv, cvecs, psum + (size_t)blockIdx.y * C);
-  block_col_reduce(red, ss, cv, cvecs, psumsq + (size_t)blockIdx.y * C);
+  block_col_reduce(red, s, cv, cpb, cvecs, psum + (size_t)blockIdx.y * C);
+  block_col_reduce(red, ss, cv, cpb, cvecs, psumsq + (size_t)blockIdx.y * C);
 }
 
 // device-side body shared by the fused reduce+finalize kernels: reduces the
@@ -89,10 +92,10 @@ __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict_
 // threads t < ncv after the final barrier).
 __device__ __forceinline__ void slab_reduce2_body(const float* a, const float* b,
                                                   uint32_t C, uint32_t nslabs,
+                                                  uint32_t ncv,
                                                   float (&outa)[8], float (&outb)[8],
                                                   uint32_t& cv_out, uint32_t& ncv_out) {
   const uint32_t cvecs = C / 8;
-  const uint32_t ncv = min(cvecs, 8u);
   const uint32_t lanes = blockDim.x / ncv;
   const uint32_t t = threadIdx.x;
   const uint32_t cv = blockIdx.x * ncv + t % ncv;
@@ -150,11 +153,11 @@ __global__ void bn_reduce_finalize_kernel(
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float* __restrict__ save_mean, float* __restrict__ save_invstd,
     float* __restrict__ scale, float* __restrict__ shift, uint32_t C,
-    uint32_t nslabs, uint32_t count, float momentum, float eps) {
+    uint32_t nslabs, uint32_t ncv, uint32_t count, float momentum, float eps) {
   float fsum[8], fsumsq[8];
-  uint32_t cv, ncv;
-  slab_reduce2_body(psum, psumsq, C, nslabs, fsum, fsumsq, cv, ncv);
-  if (threadIdx.x < ncv && cv < C / 8) {
+  uint32_t cv, ncv_;
+  slab_reduce2_body(psum, psumsq, C, nslabs, ncv, fsum, fsumsq, cv, ncv_);
+  if (threadIdx.x < ncv_ && cv < C / 8) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       uint32_t c = cv * 8 + e;
@@ -178,11 +181,11 @@ __global__ void bn_bwd_reduce_finalize_kernel(
     const float* __restrict__ pdb, const float* __restrict__ pdg,
     __bf16* __restrict__ dgamma, __bf16* __restrict__ dbeta,
     float* __restrict__ tdb, float* __restrict__ tdg, uint32_t C,
-    uint32_t nslabs) {
+    uint32_t nslabs, uint32_t ncv) {
   float db[8], dg[8];
-  uint32_t cv, ncv;
-  slab_reduce2_body(pdb, pdg, C, nslabs, db, dg, cv, ncv);
-  if (threadIdx.x < ncv && cv < C / 8) {
+  uint32_t cv, ncv_;
+  slab_reduce2_body(pdb, pdg, C, nslabs, ncv, db, dg, cv, ncv_);
+  if (threadIdx.x < ncv_ && cv < C / 8) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       uint32_t c = cv * 8 + e;
@@ -335,19 +338,20 @@ __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
                                     const float* __restrict__ save_mean,
                                     const float* __restrict__ save_invstd,
                                     float* __restrict__ pdb, float* __restrict__ pdg,
-                                    uint32_t rows, uint32_t cvecs,
+                                    uint32_t rows, uint32_t cvecs, uint32_t cpb,
                                     uint32_t rows_per_block, int relu) {
   const uint32_t t = threadIdx.x;
-  const uint32_t cv = t % cvecs;
-  const uint32_t rstep = blockDim.x / cvecs;
-  const uint32_t r0 = blockIdx.y * rows_per_block + t / cvecs;
-  const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
+  const uint32_t cv = blockIdx.x * cpb + t % cpb;
+  const bool cv_ok = cv < cvecs;
+  const uint32_t rstep = blockDim.x / cpb;
+  const uint32_t r0 = blockIdx.y * rows_per_block + t / cpb;
+  const uint32_t r1 = cv_ok ? min((blockIdx.y + 1) * rows_per_block, rows) : 0;
   const uint32_t C = cvecs * 8;
   float mean[8], invstd[8];
 #pragma unroll
   for (int e = 0; e < 8; ++e) {
-    mean[e] = save_mean[cv * 8 + e];
-    invstd[e] = save_invstd[cv * 8 + e];
+    mean[e] = cv_ok ? save_mean[cv * 8 + e] : 0.f;
+    invstd[e] = cv_ok ? save_invstd[cv * 8 + e] : 0.f;
   }
   float db[8] = {}, dg[8] = {};
   auto accum = [&](bf16x8 xv, bf16x8 gv, bf16x8 yv) {
@@ -376,8 +380,8 @@ __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
           relu ? *(const bf16x8*)(y + o0) : zed);
   }
   __shared__ float red[256 * 8];
-  block_col_reduce(red, db, cv, cvecs, pdb + (size_t)blockIdx.y * C);
-  block_col_reduce(red, dg, cv, cvecs, pdg + (size_t)blockIdx.y * C);
+  block_col_reduce(red, db, cv, cpb, cvecs, pdb + (size_t)blockIdx.y * C);
+  block_col_reduce(red, dg, cv, cpb, cvecs, pdg + (size_t)blockIdx.y * C);
 }
 
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ pdb,
@@ -442,22 +446,24 @@ __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ x,
 
 // ============================================================== host side ==
 
-// vectorized stats geometry: cvecs must divide 256 (C = 8*cvecs); channel
-// counts in the model zoo are powers of two in [64, 2048].
-static bool stats_vec_ok(uint32_t C) {
-  uint32_t cv = C / 8;
-  return C % 8 == 0 && cv <= 256 && 256 % cv == 0;
+// stats geometry: cpb = channel-vectors per block, the largest power-of-2
+// divisor of cvecs (<=256) — exact thread utilization for any C % 8 == 0
+// (e.g. inception's 80/768/1280-channel BNs: cvecs 10 -> cpb 2, grid.x 5).
+static uint32_t bn_cpb(uint32_t cvecs) {
+  uint32_t cpb = cvecs & (~cvecs + 1);  // lowest set bit = largest pow2 divisor
+  while (cpb < cvecs && cpb * 2 <= 256 && cvecs % (cpb * 2) == 0) cpb *= 2;
+  return std::min<uint32_t>(cpb, 256);
 }
 
-static void bn_grid(uint32_t rows, uint32_t cvecs, dim3& grid,
+static void bn_grid(uint32_t rows, uint32_t cvecs, uint32_t cpb, dim3& grid,
                     uint32_t& rows_per_block) {
-  uint32_t rstep = 256 / cvecs;
-  // >=16 strip iterations per thread so the per-block reduce+atomic tail
-  // amortizes; cap at 2048 blocks (8/CU) for latency hiding + BW saturation.
+  uint32_t rstep = 256 / cpb;
+  // >=16 strip iterations per thread so the per-block reduce tail amortizes;
+  // cap at 2048 blocks (8/CU) for latency hiding + BW saturation.
   uint32_t rb = std::min<uint32_t>(2048, std::max<uint32_t>(1, rows / (rstep * 16)));
   rows_per_block = (rows + rb - 1) / rb;
   rb = (rows + rows_per_block - 1) / rows_per_block;
-  grid = dim3(1, rb);
+  grid = dim3((cvecs + cpb - 1) / cpb, rb);
 }
 
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamma,
@@ -467,12 +473,13 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
                                      const c10::optional<at::Tensor>& residual) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
+  TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8, got ", C);
   uint32_t rows = N * H * W, cvecs = C / 8;
+  uint32_t cpb = bn_cpb(cvecs);
   auto opt_f = x.options().dtype(at::kFloat);
   dim3 grid;
   uint32_t rpb;
-  bn_grid(rows, cvecs, grid, rpb);
+  bn_grid(rows, cvecs, cpb, grid, rpb);
   uint32_t nslabs = grid.y;
   auto psum = at::empty({(long)nslabs, (long)C}, opt_f);
   auto psumsq = at::empty({(long)nslabs, (long)C}, opt_f);
@@ -482,14 +489,15 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
   auto s = bn_stream();
   bn_stats_kernel<<<grid, 256, 0, s>>>((const __bf16*)x.data_ptr(),
                                        psum.data_ptr<float>(),
-                                       psumsq.data_ptr<float>(), rows, cvecs, rpb);
-  uint32_t ncv = std::min(cvecs, 8u);
+                                       psumsq.data_ptr<float>(), rows, cvecs,
+                                       cpb, rpb);
+  uint32_t ncv = std::min(cpb, 8u);
   bn_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
       psum.data_ptr<float>(), psumsq.data_ptr<float>(),
       (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(), C, nslabs, rows,
+      scale.data_ptr<float>(), shift.data_ptr<float>(), C, nslabs, ncv, rows,
       momentum, eps);
   uint32_t total8 = rows * cvecs;
   FastDiv dcv;
@@ -536,12 +544,13 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
                                const at::Tensor& save_invstd, bool fuse_relu,
                                const at::Tensor& y, bool want_dres) {
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  TORCH_CHECK(stats_vec_ok(C), "bn: C must be 8*cvecs with cvecs | 256, got ", C);
+  TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8, got ", C);
   uint32_t rows = N * H * W, cvecs = C / 8;
+  uint32_t cpb = bn_cpb(cvecs);
   auto opt_f = x.options().dtype(at::kFloat);
   dim3 grid;
   uint32_t rpb;
-  bn_grid(rows, cvecs, grid, rpb);
+  bn_grid(rows, cvecs, cpb, grid, rpb);
   uint32_t nslabs = grid.y;
   auto pdb = at::empty({(long)nslabs, (long)C}, opt_f);
   auto pdg = at::empty({(long)nslabs, (long)C}, opt_f);
@@ -556,12 +565,12 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
       (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
       (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
       save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
-      pdg.data_ptr<float>(), rows, cvecs, rpb, fuse_relu ? 1 : 0);
-  uint32_t ncv = std::min(cvecs, 8u);
+      pdg.data_ptr<float>(), rows, cvecs, cpb, rpb, fuse_relu ? 1 : 0);
+  uint32_t ncv = std::min(cpb, 8u);
   bn_bwd_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
       pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
       (__bf16*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
-      C, nslabs);
+      C, nslabs, ncv);
   uint32_t total8 = rows * cvecs;
   FastDiv dcv;
   dcv.init(cvecs);

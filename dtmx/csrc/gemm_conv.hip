@@ -727,6 +727,8 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
                       const c10::optional<at::Tensor>& bias) {
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(w);
+  TORCH_CHECK(x.size(1) == w.size(1), "linear: in_features mismatch (x ",
+              x.size(1), " vs w ", w.size(1), ")");
   auto xc = pad_cols8(x);
   auto wc = pad_cols8(w);
   uint32_t M = xc.size(0), K = xc.size(1), N = wc.size(0);

@@ -102,6 +102,77 @@ __global__ void maxpool_bwd_kernel(const __bf16* __restrict__ dy,
   }
 }
 
+// scalar fallbacks for C % 8 != 0 (zoo models like lenet's 20 channels)
+__global__ void maxpool_fwd_scalar_kernel(const __bf16* __restrict__ x,
+                                          __bf16* __restrict__ y,
+                                          uint8_t* __restrict__ idx, uint32_t N,
+                                          uint32_t C, uint32_t H, uint32_t W,
+                                          uint32_t P, uint32_t Q, uint32_t K,
+                                          int u, int pad) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t total = (size_t)N * P * Q * C;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    uint32_t c = i % C;
+    size_t m = i / C;
+    uint32_t q = m % Q;
+    uint32_t p = (m / Q) % P;
+    uint32_t n = m / ((size_t)P * Q);
+    float best = -3.4e38f;
+    uint8_t besti = 0;
+    for (uint32_t kh = 0; kh < K; ++kh) {
+      int ih = (int)(p * u) - pad + (int)kh;
+      if ((uint32_t)ih >= H) continue;
+      for (uint32_t kw = 0; kw < K; ++kw) {
+        int iw = (int)(q * u) - pad + (int)kw;
+        if ((uint32_t)iw >= W) continue;
+        float v = (float)x[(((size_t)n * H + ih) * W + iw) * C + c];
+        if (v > best) { best = v; besti = kh * K + kw; }
+      }
+    }
+    y[i] = (__bf16)best;
+    idx[i] = besti;
+  }
+}
+
+__global__ void maxpool_bwd_scalar_kernel(const __bf16* __restrict__ dy,
+                                          const uint8_t* __restrict__ idx,
+                                          __bf16* __restrict__ dx, uint32_t N,
+                                          uint32_t C, uint32_t H, uint32_t W,
+                                          uint32_t P, uint32_t Q, uint32_t K,
+                                          int u, int pad) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t total = (size_t)N * H * W * C;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    uint32_t c = i % C;
+    size_t m = i / C;
+    uint32_t w = m % W;
+    uint32_t h = (m / W) % H;
+    uint32_t n = m / ((size_t)H * W);
+    int plo = ((int)h + pad - (int)K + u) / u;
+    if (plo < 0) plo = 0;
+    int phi = ((int)h + pad) / u;
+    if (phi >= (int)P) phi = P - 1;
+    int qlo = ((int)w + pad - (int)K + u) / u;
+    if (qlo < 0) qlo = 0;
+    int qhi = ((int)w + pad) / u;
+    if (qhi >= (int)Q) qhi = Q - 1;
+    float acc = 0.f;
+    for (int p = plo; p <= phi; ++p) {
+      uint32_t kh = (uint32_t)((int)h + pad - p * u);
+      if (kh >= K) continue;
+      for (int q = qlo; q <= qhi; ++q) {
+        uint32_t kw = (uint32_t)((int)w + pad - q * u);
+        if (kw >= K) continue;
+        size_t o = (((size_t)n * P + p) * Q + q) * C + c;
+        if (idx[o] == kh * K + kw) acc += (float)dy[o];
+      }
+    }
+    dx[i] = (__bf16)acc;
+  }
+}
+
 // ------------------------------------------------------- global average pool
 
 __global__ void gap_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
@@ -191,14 +262,20 @@ std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, long kernel, long strid
                      at::MemoryFormat::ChannelsLast);
   auto idx = at::empty({(long)N, (long)C, (long)P, (long)Q},
                        x.options().dtype(at::kByte), at::MemoryFormat::ChannelsLast);
-  TORCH_CHECK(C % 8 == 0, "maxpool: C must be a multiple of 8");
-  FastDiv dCv, dQ_, dPQ;
-  dCv.init(C / 8); dQ_.init(Q); dPQ.init(P * Q);
-  size_t total = (size_t)N * P * Q * (C / 8);
-  maxpool_fwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
-      (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
-      dCv, dQ_, dPQ);
+  if (C % 8 == 0) {
+    FastDiv dCv, dQ_, dPQ;
+    dCv.init(C / 8); dQ_.init(Q); dPQ.init(P * Q);
+    size_t total = (size_t)N * P * Q * (C / 8);
+    maxpool_fwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+        (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
+        (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
+        dCv, dQ_, dPQ);
+  } else {
+    size_t total = (size_t)N * P * Q * C;
+    maxpool_fwd_scalar_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+        (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
+        (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+  }
   return {y, idx};
 }
 
@@ -208,13 +285,20 @@ at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx, long H,
   auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
   auto dx = at::empty({(long)N, (long)C, (long)H, (long)W}, dy.options(),
                       at::MemoryFormat::ChannelsLast);
-  FastDiv dCv, dW2, dHW;
-  dCv.init(C / 8); dW2.init(W); dHW.init(H * W);
-  size_t total = (size_t)N * H * W * (C / 8);
-  maxpool_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-      (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
-      (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
-      dCv, dW2, dHW);
+  if (C % 8 == 0) {
+    FastDiv dCv, dW2, dHW;
+    dCv.init(C / 8); dW2.init(W); dHW.init(H * W);
+    size_t total = (size_t)N * H * W * (C / 8);
+    maxpool_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+        (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
+        (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
+        dCv, dW2, dHW);
+  } else {
+    size_t total = (size_t)N * H * W * C;
+    maxpool_bwd_scalar_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+        (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
+        (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+  }
   return dx;
 }
 

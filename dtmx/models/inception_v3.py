@@ -78,19 +78,30 @@ class InceptionD(nn.Module):
 
 
 class InceptionE(nn.Module):
+    """The E block's 3x3 branches split into two parallel convs whose outputs
+    concatenate (1x3/3x1 in the original; square 3x3 here), so the block
+    emits 320+768+768+192 = 2048 channels."""
+
     def __init__(self, in_ch):
         super().__init__()
         self.b1 = ConvBN(in_ch, 320, 1)
         self.b2a = ConvBN(in_ch, 384, 1)
-        self.b2b = ConvBN(384, 384, 3, 1, 1)
+        self.b2b1 = ConvBN(384, 384, 3, 1, 1)
+        self.b2b2 = ConvBN(384, 384, 3, 1, 1)
         self.b3 = nn.Sequential(ConvBN(in_ch, 448, 1), ConvBN(448, 384, 3, 1, 1))
-        self.b3b = ConvBN(384, 384, 3, 1, 1)
+        self.b3b1 = ConvBN(384, 384, 3, 1, 1)
+        self.b3b2 = ConvBN(384, 384, 3, 1, 1)
         self.b4 = nn.Sequential(_AvgPool3s1(), ConvBN(in_ch, 192, 1))
 
     def forward(self, x):
-        y2 = self.b2b(self.b2a(x))
-        y3 = self.b3b(self.b3(x))
-        return torch.cat([self.b1(x), y2, y3, self.b4(x)], dim=1)
+        t2 = self.b2a(x)
+        t3 = self.b3(x)
+        return torch.cat([
+            self.b1(x),
+            self.b2b1(t2), self.b2b2(t2),
+            self.b3b1(t3), self.b3b2(t3),
+            self.b4(x),
+        ], dim=1)
 
 
 class InceptionV3(nn.Module):

@@ -269,6 +269,9 @@ class Module:
         if isinstance(kv, DistKVStore) or on_gpu:
             self._bucketer = GradBucketer(list(args.values()),
                                           flatten_params=self._use_fused_sgd)
+            comp = getattr(kv, "_compression", None)
+            if isinstance(kv, DistKVStore) and comp is not None:
+                self._bucketer.set_compression(comp.threshold)
             if isinstance(kv, DistKVStore) and initialize_from_kvstore:
                 self._sync_full_state()
         self.optimizer_initialized = True

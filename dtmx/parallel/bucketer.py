@@ -137,8 +137,11 @@ class GradBucketer:
         bi = self._param_bucket[id(param)]
         self._pending[bi] -= 1
         if self._pending[bi] == 0 and self._enabled:
-            work = dist.all_reduce(self.flat[bi], op=dist.ReduceOp.SUM, async_op=True)
-            self._works.append((bi, work))
+            if getattr(self, "_gc_residual", None) is not None and self.flat[bi].is_cuda:
+                self._compressed_reduce(bi)
+            else:
+                work = dist.all_reduce(self.flat[bi], op=dist.ReduceOp.SUM, async_op=True)
+                self._works.append((bi, work))
 
     def finish(self):
         """Join outstanding collectives; call before the optimizer step."""
@@ -147,6 +150,38 @@ class GradBucketer:
             if self.average:
                 self.flat[bi].div_(dist.get_world_size())
         self._works.clear()
+
+    # -- 2-bit compressed reduction (reference gradient compression over
+    # collectives: quantize w/ error feedback -> all-gather the packed codes
+    # -> local dequantize-sum; the server dequantized before merging
+    # (kvstore_dist_server.h:606), so the numerics match; wire bytes drop
+    # ~8x (2 bits vs bf16), per-GPU traffic ~2x lower at world=8) --
+    def set_compression(self, threshold: float):
+        self._gc_threshold = float(threshold)
+        self._gc_residual = [
+            torch.zeros_like(b, dtype=torch.float32) for b in self.flat
+        ]
+
+    def _compressed_reduce(self, bi: int):
+        from ..ops.hip import require_ext
+
+        ext = require_ext()
+        world = dist.get_world_size()
+        flat = self.flat[bi]
+        packed = ext.quantize_2bit(flat.view(-1), self._gc_residual[bi].view(-1),
+                                   self._gc_threshold)
+        gathered = [torch.empty_like(packed) for _ in range(world)]
+        dist.all_gather(gathered, packed)
+        out = ext.dequantize_2bit(gathered[0], flat.numel(), self._gc_threshold)
+        for g in gathered[1:]:
+            out += ext.dequantize_2bit(g, flat.numel(), self._gc_threshold)
+        flat.view(-1).copy_(out)
+
+    def _on_grad_ready_compressed(self, param):
+        bi = self._param_bucket[id(param)]
+        self._pending[bi] -= 1
+        if self._pending[bi] == 0 and self._enabled:
+            self._compressed_reduce(bi)
 
     def rebuild_after_membership_change(self):
         """Communicator changed (elastic re-form): nothing to re-shard in the
