@@ -267,8 +267,10 @@ class Module:
                 if st is None:
                     self._updater.states[i] = optimizer.create_state_multi_precision(i, p.data)
         if isinstance(kv, DistKVStore) or on_gpu:
+            async_mode = isinstance(kv, DistKVStore) and "async" in kv.type
             self._bucketer = GradBucketer(list(args.values()),
-                                          flatten_params=self._use_fused_sgd)
+                                          flatten_params=self._use_fused_sgd,
+                                          async_mode=async_mode)
             comp = getattr(kv, "_compression", None)
             if isinstance(kv, DistKVStore) and comp is not None:
                 self._bucketer.set_compression(comp.threshold)
@@ -361,10 +363,15 @@ class Module:
                     opt.clip_gradient or 0.0, hyper=hyper,
                 )
             else:
+                grads = (
+                    self._bucketer.reduced_views()
+                    if self._bucketer.async_mode
+                    else [p.grad for p in args.values()]
+                )
                 for i, (name, p) in enumerate(args.items()):
-                    if name in self.fixed_param_names or p.grad is None:
+                    if name in self.fixed_param_names or grads[i] is None:
                         continue
-                    self._updater(i, p.grad, p.data)
+                    self._updater(i, grads[i], p.data)
         elif isinstance(self._kvstore, LocalKVStore) or self._kvstore is None:
             for i, (name, p) in enumerate(args.items()):
                 if name in self.fixed_param_names or p.grad is None:

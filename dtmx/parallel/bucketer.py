@@ -34,10 +34,17 @@ class GradBucketer:
 
     def __init__(self, params: Sequence[torch.nn.Parameter],
                  bucket_mb: Optional[float] = None, average: bool = False,
-                 flatten_params: bool = False):
+                 flatten_params: bool = False, async_mode: bool = False):
+        """async_mode (kvstore 'dist_async'): the collective analog of the
+        reference's asynchronous PS (kvstore_dist 'dist_async' — workers never
+        barrier on each other's pushes). Gradients are double-buffered: step
+        t's all-reduce completes in the background while step t+1 computes,
+        and the optimizer consumes the one-step-delayed reduced gradients —
+        full comm/compute overlap, staleness bounded at 1 step."""
         self.params = [p for p in params if p.requires_grad]
         self.average = average
         self.flatten_params = flatten_params
+        self.async_mode = async_mode
         self.flat_w: List[torch.Tensor] = []
         self.flat_master: List[torch.Tensor] = []
         self.flat_mom: List[torch.Tensor] = []
@@ -94,6 +101,20 @@ class GradBucketer:
             p.register_post_accumulate_grad_hook(self._on_grad_ready) for p in self.params
         ]
         self._enabled = dist.is_initialized() and dist.get_world_size() > 1
+        if self.async_mode:
+            # shadow grad set: the accumulation/allreduce target alternates
+            self.flat_shadow = [torch.zeros_like(b) for b in self.flat]
+            self._views_main = [p.grad for p in self.params]
+            # shadow views in self.params order
+            vs = {}
+            for bi, bucket in enumerate(self.buckets):
+                off = 0
+                for p in bucket:
+                    vs[id(p)] = self._shaped_view(self.flat_shadow[bi], off, p)
+                    off += p.numel()
+            self._views_shadow = [vs[id(p)] for p in self.params]
+            self._prev_works: List[Tuple[int, dist.Work]] = []
+            self._cur_is_shadow = False
 
     @staticmethod
     def _shaped_view(buf: torch.Tensor, off: int, p: torch.Tensor) -> torch.Tensor:
@@ -115,12 +136,13 @@ class GradBucketer:
         from ..ops.hip import require_ext
 
         ext = require_ext()
+        grads = self.reduced_flats() if self.async_mode else self.flat
         for bi in range(len(self.flat)):
             if hyper is not None:
-                ext.sgd_mom_mp_dev(self.flat_w[bi], self.flat[bi],
+                ext.sgd_mom_mp_dev(self.flat_w[bi], grads[bi],
                                    self.flat_master[bi], self.flat_mom[bi], hyper)
             else:
-                ext.sgd_mom_mp(self.flat_w[bi], self.flat[bi], self.flat_master[bi],
+                ext.sgd_mom_mp(self.flat_w[bi], grads[bi], self.flat_master[bi],
                                self.flat_mom[bi], lr, momentum, wd, rescale, clip)
 
     def state_tensors(self) -> List[torch.Tensor]:
@@ -128,23 +150,54 @@ class GradBucketer:
 
     # -- per-iteration lifecycle -------------------------------------------
     def zero_grad(self):
+        if self.async_mode:
+            # swap the accumulation target; the other set's all-reduce keeps
+            # flying while this step computes
+            self._cur_is_shadow = not self._cur_is_shadow
+            views = self._views_shadow if self._cur_is_shadow else self._views_main
+            for p, v in zip(self.params, views):
+                p.grad = v
+            for buf in self._cur_flats():
+                buf.zero_()
+            self._pending = [len(b) for b in self.buckets]
+            return
         for buf in self.flat:
             buf.zero_()
         self._pending = [len(b) for b in self.buckets]
         self._works.clear()
 
+    def _cur_flats(self):
+        return self.flat_shadow if getattr(self, "_cur_is_shadow", False) else self.flat
+
+    def reduced_flats(self):
+        """async mode: the one-step-delayed, fully reduced gradient set."""
+        return self.flat if self._cur_is_shadow else self.flat_shadow
+
+    def reduced_views(self):
+        return self._views_main if self._cur_is_shadow else self._views_shadow
+
     def _on_grad_ready(self, param: torch.nn.Parameter):
         bi = self._param_bucket[id(param)]
         self._pending[bi] -= 1
         if self._pending[bi] == 0 and self._enabled:
-            if getattr(self, "_gc_residual", None) is not None and self.flat[bi].is_cuda:
+            flat = self._cur_flats()[bi] if self.async_mode else self.flat[bi]
+            if getattr(self, "_gc_residual", None) is not None and flat.is_cuda:
                 self._compressed_reduce(bi)
             else:
-                work = dist.all_reduce(self.flat[bi], op=dist.ReduceOp.SUM, async_op=True)
+                work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
                 self._works.append((bi, work))
 
     def finish(self):
-        """Join outstanding collectives; call before the optimizer step."""
+        """Join outstanding collectives; call before the optimizer step.
+        async mode: joins the PREVIOUS step's collectives (usually already
+        complete) and rotates the work queues — the optimizer then reads
+        reduced_flats()/reduced_views()."""
+        if self.async_mode:
+            for bi, work in self._prev_works:
+                work.wait()
+            self._prev_works = self._works
+            self._works = []
+            return
         for bi, work in self._works:
             work.wait()
             if self.average:
