@@ -91,6 +91,26 @@ class KVStore:
     def set_gradient_compression(self, compression_params):
         raise NotImplementedError(f"{self.type} kvstore does not support compression")
 
+    def row_sparse_pull(self, key, out=None, priority: int = 0, row_ids=None):
+        """Pull selected rows of a 2-D value (reference kvstore.py:314 /
+        kvstore_dist.h PullRowSparse_): `row_ids` selects rows of the stored
+        tensor; rows not selected are zero in `out`."""
+        if row_ids is None:
+            return self.pull(key, out=out, priority=priority)
+        stored = self.pull(key)[0] if out is None else None
+        targets = _as_list(out) if out is not None else [torch.zeros(0)]
+        results = []
+        for o, rids in zip(
+            targets, _as_list(row_ids) if isinstance(row_ids, (list, tuple)) else [row_ids]
+        ):
+            src = stored if stored is not None else self.pull(key)[0]
+            rids = rids.to(torch.long)
+            dst = o if out is not None else torch.zeros_like(src)
+            dst.zero_()
+            dst[rids] = src[rids].to(dst.dtype)
+            results.append(dst)
+        return None if out is not None else results
+
     @property
     def type(self) -> str:
         raise NotImplementedError

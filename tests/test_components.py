@@ -184,3 +184,21 @@ def test_kvstore_optimizer_states_roundtrip(tmp_path):
     assert set(s1) == set(s2)
     for k in s1:
         assert torch.allclose(s1[k], s2[k])
+
+
+# --------------------------------------------------- test_utils + row_sparse
+
+def test_check_numeric_gradient_harness():
+    from dtmx.test_utils import check_numeric_gradient
+
+    w = torch.randn(3, 3)
+    check_numeric_gradient(lambda t: (t * t).sum(), [w])
+
+
+def test_row_sparse_pull():
+    kv = dtmx.kvstore.create("local")
+    kv.init("emb", torch.arange(20, dtype=torch.float32).reshape(5, 4))
+    out = torch.zeros(5, 4)
+    kv.row_sparse_pull("emb", out=out, row_ids=torch.tensor([1, 3]))
+    assert out[1, 0] == 4 and out[3, 3] == 15
+    assert out[0].sum() == 0 and out[2].sum() == 0
