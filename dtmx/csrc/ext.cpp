@@ -41,6 +41,8 @@ void sgd_mom_f32(at::Tensor, const at::Tensor&, at::Tensor, double, double,
 // compress.hip
 at::Tensor quantize_2bit(const at::Tensor&, at::Tensor, double);
 at::Tensor dequantize_2bit(const at::Tensor&, long, double);
+// recordio.cpp
+void register_recordio(py::module_& m);
 }  // namespace dtmx
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -67,4 +69,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_mom_f32", &dtmx::sgd_mom_f32);
   m.def("quantize_2bit", &dtmx::quantize_2bit);
   m.def("dequantize_2bit", &dtmx::dequantize_2bit);
+  dtmx::register_recordio(m);
 }

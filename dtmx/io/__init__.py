@@ -346,3 +346,50 @@ class PrefetchingIter(DataIter):
         if batch is None:
             raise StopIteration
         return batch
+
+
+class ImageRecordIter(DataIter):
+    """RecordIO-backed image iterator on the native C++ pipeline
+    (reference src/io/iter_image_recordio_2.cc; dtmx/csrc/recordio.cpp):
+    threaded record parsing + batch assembly + bounded prefetch queue,
+    sharded by (part_index, num_parts). Records carry raw uint8 HWC payloads
+    packed by tools/im2rec.py (no JPEG stage in this environment)."""
+
+    def __init__(self, path_imgrec: str, data_shape, batch_size: int,
+                 shuffle: bool = False, part_index: int = 0, num_parts: int = 1,
+                 preprocess_threads: int = 4, prefetch_buffer: int = 4,
+                 seed: int = 0, label_name: str = "softmax_label", **kwargs):
+        super().__init__(batch_size)
+        from dtmx.ops.hip import require_ext
+
+        ext = require_ext()
+        self._reader = ext.RecordIOReader(path_imgrec)
+        # data_shape is CHW (reference convention); records store HWC raw
+        c, h, w = data_shape
+        self._chw = (c, h, w)
+        self._loader = ext.RecordBatchLoader(
+            self._reader, batch_size, [h, w, c], part_index, num_parts,
+            shuffle, preprocess_threads, prefetch_buffer, seed,
+        )
+        self.label_name = label_name
+
+    @property
+    def provide_data(self):
+        return [DataDesc("data", (self.batch_size,) + self._chw, torch.float32)]
+
+    @property
+    def provide_label(self):
+        return [DataDesc(self.label_name, (self.batch_size,), torch.float32)]
+
+    def reset(self):
+        self._loader.reset()
+
+    def next(self):
+        out = self._loader.next()
+        if not out:
+            raise StopIteration
+        data, label = out
+        data = data.permute(0, 3, 1, 2)  # HWC records -> logical NCHW (NHWC memory)
+        return DataBatch(data=[data], label=[label], pad=0,
+                         provide_data=self.provide_data,
+                         provide_label=self.provide_label)
