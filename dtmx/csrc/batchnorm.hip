@@ -1,5 +1,5 @@
 // NHWC BatchNorm fwd/bwd for gfx950 (reference src/operator/nn/batch_norm.cu:
-// 208-360 — redesigned for NHWC/bf16: channel reductions are bf16x8-vectorized
+// 208-360 — redesigned for NHWC/bf16: channel reductions are V8-vectorized
 // column sums with fp32 atomically-merged partials; apply passes are
 // vectorized elementwise with optional fused ReLU; all hot-loop indexing is
 // 32-bit with FastDiv (64-bit div/mod on the elementwise path measured ~5x).
@@ -47,10 +47,12 @@ __device__ __forceinline__ void block_col_reduce(float* red, float (&v)[8],
 // r = r0 + t/cvecs, stepping by 256/cvecs. Requires cvecs <= 256 divisor of
 // 256 (C is a power-of-two multiple of 8 in practice); general C uses the
 // strided variant below.
-__global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ psum,
+template <typename elem_t>
+__global__ void bn_stats_kernel(const elem_t* __restrict__ x, float* __restrict__ psum,
                                 float* __restrict__ psumsq, uint32_t rows,
                                 uint32_t cvecs, uint32_t cpb,
                                 uint32_t rows_per_block) {
+  using V8 = typename E8<elem_t>::v8;
   const uint32_t t = threadIdx.x;
   const uint32_t cv = blockIdx.x * cpb + t % cpb;
   const bool cv_ok = cv < cvecs;
@@ -62,7 +64,7 @@ __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict_
   // 4x unrolled so four 16-B loads are in flight per wave (a single-buffer
   // loop compiles to load -> vmcnt(0) -> use and runs HBM-latency-bound:
   // measured ~20x off the bandwidth roofline).
-  auto accum = [&](bf16x8 v) {
+  auto accum = [&](V8 v) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float f = (float)v[e];
@@ -72,14 +74,14 @@ __global__ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict_
   };
   uint32_t r = r0;
   for (; r + 3 * rstep < r1; r += 4 * rstep) {
-    bf16x8 v0 = *(const bf16x8*)(x + (size_t)r * C + cv * 8);
-    bf16x8 v1 = *(const bf16x8*)(x + (size_t)(r + rstep) * C + cv * 8);
-    bf16x8 v2 = *(const bf16x8*)(x + (size_t)(r + 2 * rstep) * C + cv * 8);
-    bf16x8 v3 = *(const bf16x8*)(x + (size_t)(r + 3 * rstep) * C + cv * 8);
+    V8 v0 = *(const V8*)(x + (size_t)r * C + cv * 8);
+    V8 v1 = *(const V8*)(x + (size_t)(r + rstep) * C + cv * 8);
+    V8 v2 = *(const V8*)(x + (size_t)(r + 2 * rstep) * C + cv * 8);
+    V8 v3 = *(const V8*)(x + (size_t)(r + 3 * rstep) * C + cv * 8);
     accum(v0); accum(v1); accum(v2); accum(v3);
   }
   for (; r < r1; r += rstep)
-    accum(*(const bf16x8*)(x + (size_t)r * C + cv * 8));
+    accum(*(const V8*)(x + (size_t)r * C + cv * 8));
   // intra-block tree reduction, then a per-block slab row (see
   // block_col_reduce).
   __shared__ float red[256 * 8];
@@ -147,13 +149,15 @@ __device__ __forceinline__ void slab_reduce2_body(const float* a, const float* b
 }
 
 // fused: slab reduce + forward finalize (mean/invstd, running stats, scale/shift)
+template <typename elem_t>
 __global__ void bn_reduce_finalize_kernel(
     const float* __restrict__ psum, const float* __restrict__ psumsq,
-    const __bf16* __restrict__ gamma, const __bf16* __restrict__ beta,
+    const elem_t* __restrict__ gamma, const elem_t* __restrict__ beta,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float* __restrict__ save_mean, float* __restrict__ save_invstd,
     float* __restrict__ scale, float* __restrict__ shift, uint32_t C,
     uint32_t nslabs, uint32_t ncv, uint32_t count, float momentum, float eps) {
+  using V8 = typename E8<elem_t>::v8;
   float fsum[8], fsumsq[8];
   uint32_t cv, ncv_;
   slab_reduce2_body(psum, psumsq, C, nslabs, ncv, fsum, fsumsq, cv, ncv_);
@@ -177,11 +181,13 @@ __global__ void bn_reduce_finalize_kernel(
 }
 
 // fused: slab reduce + backward finalize (dgamma/dbeta + totals for dx)
+template <typename elem_t>
 __global__ void bn_bwd_reduce_finalize_kernel(
     const float* __restrict__ pdb, const float* __restrict__ pdg,
-    __bf16* __restrict__ dgamma, __bf16* __restrict__ dbeta,
+    elem_t* __restrict__ dgamma, elem_t* __restrict__ dbeta,
     float* __restrict__ tdb, float* __restrict__ tdg, uint32_t C,
     uint32_t nslabs, uint32_t ncv) {
+  using V8 = typename E8<elem_t>::v8;
   float db[8], dg[8];
   uint32_t cv, ncv_;
   slab_reduce2_body(pdb, pdg, C, nslabs, ncv, db, dg, cv, ncv_);
@@ -189,8 +195,8 @@ __global__ void bn_bwd_reduce_finalize_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       uint32_t c = cv * 8 + e;
-      dgamma[c] = (__bf16)dg[e];
-      dbeta[c] = (__bf16)db[e];
+      dgamma[c] = (elem_t)dg[e];
+      dbeta[c] = (elem_t)db[e];
       tdb[c] = db[e];
       tdg[c] = dg[e];
     }
@@ -258,10 +264,11 @@ __global__ void slab_reduce2_kernel(const float* __restrict__ a,
 }
 
 // ---- finalize: mean/invstd, running stats, fused scale/shift -------------
+template <typename elem_t>
 __global__ void bn_finalize_kernel(const float* __restrict__ psum,
                                    const float* __restrict__ psumsq,
-                                   const __bf16* __restrict__ gamma,
-                                   const __bf16* __restrict__ beta,
+                                   const elem_t* __restrict__ gamma,
+                                   const elem_t* __restrict__ beta,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
                                    float* __restrict__ save_mean,
@@ -270,6 +277,7 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,
                                    float* __restrict__ shift, uint32_t C,
                                    uint32_t nslabs, uint32_t count,
                                    float momentum, float eps) {
+  using V8 = typename E8<elem_t>::v8;
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float fsum = 0.f, fsumsq = 0.f;
@@ -290,13 +298,15 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,
   shift[c] = (float)beta[c] - mean * g * invstd;
 }
 
-__global__ void bn_infer_prep_kernel(const __bf16* __restrict__ gamma,
-                                     const __bf16* __restrict__ beta,
+template <typename elem_t>
+__global__ void bn_infer_prep_kernel(const elem_t* __restrict__ gamma,
+                                     const elem_t* __restrict__ beta,
                                      const float* __restrict__ running_mean,
                                      const float* __restrict__ running_var,
                                      float* __restrict__ scale,
                                      float* __restrict__ shift, uint32_t C,
                                      float eps) {
+  using V8 = typename E8<elem_t>::v8;
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float invstd = rsqrtf(running_var[c] + eps);
@@ -305,41 +315,45 @@ __global__ void bn_infer_prep_kernel(const __bf16* __restrict__ gamma,
   shift[c] = (float)beta[c] - running_mean[c] * g * invstd;
 }
 
-// ---- apply: y = x*scale + shift (+relu), bf16x8, 32-bit indexing ---------
-__global__ void bn_apply_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+// ---- apply: y = x*scale + shift (+relu), V8, 32-bit indexing ---------
+template <typename elem_t>
+__global__ void bn_apply_kernel(const elem_t* __restrict__ x, elem_t* __restrict__ y,
                                 const float* __restrict__ scale,
                                 const float* __restrict__ shift,
-                                const __bf16* __restrict__ residual,  // nullable
+                                const elem_t* __restrict__ residual,  // nullable
                                 uint32_t total8, FastDiv dcv, int relu) {
+  using V8 = typename E8<elem_t>::v8;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t stride = gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
-    bf16x8 v = *(const bf16x8*)(x + (size_t)i * 8);
-    bf16x8 res;
-    if (residual) res = *(const bf16x8*)(residual + (size_t)i * 8);
+    V8 v = *(const V8*)(x + (size_t)i * 8);
+    V8 res;
+    if (residual) res = *(const V8*)(residual + (size_t)i * 8);
     uint32_t q = dcv.div(i);
     uint32_t c0 = dcv.mod(i, q) * 8;
-    bf16x8 o;
+    V8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float r = (float)v[e] * scale[c0 + e] + shift[c0 + e];
       if (residual) r += (float)res[e];
       if (relu) r = fmaxf(r, 0.f);
-      o[e] = (__bf16)r;
+      o[e] = (elem_t)r;
     }
-    *(bf16x8*)(y + (size_t)i * 8) = o;
+    *(V8*)(y + (size_t)i * 8) = o;
   }
 }
 
 // ---- backward stats: per-channel sum(dy), sum(dy*xhat), vectorized -------
-__global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
-                                    const __bf16* __restrict__ dy,
-                                    const __bf16* __restrict__ y,  // relu mask
+template <typename elem_t>
+__global__ void bn_bwd_stats_kernel(const elem_t* __restrict__ x,
+                                    const elem_t* __restrict__ dy,
+                                    const elem_t* __restrict__ y,  // relu mask
                                     const float* __restrict__ save_mean,
                                     const float* __restrict__ save_invstd,
                                     float* __restrict__ pdb, float* __restrict__ pdg,
                                     uint32_t rows, uint32_t cvecs, uint32_t cpb,
                                     uint32_t rows_per_block, int relu) {
+  using V8 = typename E8<elem_t>::v8;
   const uint32_t t = threadIdx.x;
   const uint32_t cv = blockIdx.x * cpb + t % cpb;
   const bool cv_ok = cv < cvecs;
@@ -354,7 +368,7 @@ __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
     invstd[e] = cv_ok ? save_invstd[cv * 8 + e] : 0.f;
   }
   float db[8] = {}, dg[8] = {};
-  auto accum = [&](bf16x8 xv, bf16x8 gv, bf16x8 yv) {
+  auto accum = [&](V8 xv, V8 gv, V8 yv) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float g = (float)gv[e];
@@ -363,34 +377,36 @@ __global__ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
       dg[e] += g * ((float)xv[e] - mean[e]) * invstd[e];
     }
   };
-  bf16x8 zed = {};
+  V8 zed = {};
   uint32_t r = r0;
   for (; r + rstep < r1; r += 2 * rstep) {  // 2x unroll: 4-6 loads in flight
     size_t o0 = (size_t)r * C + cv * 8, o1 = (size_t)(r + rstep) * C + cv * 8;
-    bf16x8 x0 = *(const bf16x8*)(x + o0), x1 = *(const bf16x8*)(x + o1);
-    bf16x8 g0 = *(const bf16x8*)(dy + o0), g1 = *(const bf16x8*)(dy + o1);
-    bf16x8 y0 = relu ? *(const bf16x8*)(y + o0) : zed;
-    bf16x8 y1 = relu ? *(const bf16x8*)(y + o1) : zed;
+    V8 x0 = *(const V8*)(x + o0), x1 = *(const V8*)(x + o1);
+    V8 g0 = *(const V8*)(dy + o0), g1 = *(const V8*)(dy + o1);
+    V8 y0 = relu ? *(const V8*)(y + o0) : zed;
+    V8 y1 = relu ? *(const V8*)(y + o1) : zed;
     accum(x0, g0, y0);
     accum(x1, g1, y1);
   }
   for (; r < r1; r += rstep) {
     size_t o0 = (size_t)r * C + cv * 8;
-    accum(*(const bf16x8*)(x + o0), *(const bf16x8*)(dy + o0),
-          relu ? *(const bf16x8*)(y + o0) : zed);
+    accum(*(const V8*)(x + o0), *(const V8*)(dy + o0),
+          relu ? *(const V8*)(y + o0) : zed);
   }
   __shared__ float red[256 * 8];
   block_col_reduce(red, db, cv, cpb, cvecs, pdb + (size_t)blockIdx.y * C);
   block_col_reduce(red, dg, cv, cpb, cvecs, pdg + (size_t)blockIdx.y * C);
 }
 
+template <typename elem_t>
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ pdb,
                                        const float* __restrict__ pdg,
-                                       __bf16* __restrict__ dgamma,
-                                       __bf16* __restrict__ dbeta,
+                                       elem_t* __restrict__ dgamma,
+                                       elem_t* __restrict__ dbeta,
                                        float* __restrict__ tdb,
                                        float* __restrict__ tdg, uint32_t C,
                                        uint32_t nslabs) {
+  using V8 = typename E8<elem_t>::v8;
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float db = 0.f, dg = 0.f;
@@ -398,49 +414,51 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ pdb,
     db += pdb[(size_t)b * C + c];
     dg += pdg[(size_t)b * C + c];
   }
-  dgamma[c] = (__bf16)dg;
-  dbeta[c] = (__bf16)db;
+  dgamma[c] = (elem_t)dg;
+  dbeta[c] = (elem_t)db;
   tdb[c] = db;
   tdg[c] = dg;
 }
 
 // dx = gamma*invstd * (dy - dbeta/M - xhat * dgamma/M)
-__global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ x,
-                                 const __bf16* __restrict__ dy,
-                                 const __bf16* __restrict__ y,
+template <typename elem_t>
+__global__ void bn_bwd_dx_kernel(const elem_t* __restrict__ x,
+                                 const elem_t* __restrict__ dy,
+                                 const elem_t* __restrict__ y,
                                  const float* __restrict__ save_mean,
                                  const float* __restrict__ save_invstd,
-                                 const __bf16* __restrict__ gamma,
+                                 const elem_t* __restrict__ gamma,
                                  const float* __restrict__ pdb,
                                  const float* __restrict__ pdg,
-                                 __bf16* __restrict__ dx,
-                                 __bf16* __restrict__ dres,  // nullable
+                                 elem_t* __restrict__ dx,
+                                 elem_t* __restrict__ dres,  // nullable
                                  uint32_t total8, FastDiv dcv, float inv_count,
                                  int relu) {
+  using V8 = typename E8<elem_t>::v8;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t stride = gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
     size_t off = (size_t)i * 8;
-    bf16x8 xv = *(const bf16x8*)(x + off);
-    bf16x8 gv = *(const bf16x8*)(dy + off);
-    bf16x8 yv;
-    if (relu) yv = *(const bf16x8*)(y + off);
+    V8 xv = *(const V8*)(x + off);
+    V8 gv = *(const V8*)(dy + off);
+    V8 yv;
+    if (relu) yv = *(const V8*)(y + off);
     uint32_t q = dcv.div(i);
     uint32_t c0 = dcv.mod(i, q) * 8;
-    bf16x8 o, om;
+    V8 o, om;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       uint32_t c = c0 + e;
       float g = (float)gv[e];
       if (relu && (float)yv[e] <= 0.f) g = 0.f;
-      if (dres) om[e] = (__bf16)g;
+      if (dres) om[e] = (elem_t)g;
       float invstd = save_invstd[c];
       float xh = ((float)xv[e] - save_mean[c]) * invstd;
-      o[e] = (__bf16)((float)gamma[c] * invstd *
+      o[e] = (elem_t)((float)gamma[c] * invstd *
                       (g - pdb[c] * inv_count - xh * pdg[c] * inv_count));
     }
-    *(bf16x8*)(dx + off) = o;
-    if (dres) *(bf16x8*)(dres + off) = om;
+    *(V8*)(dx + off) = o;
+    if (dres) *(V8*)(dres + off) = om;
   }
 }
 
@@ -487,27 +505,29 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
   auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
   auto y = at::empty_like(x);
   auto s = bn_stream();
-  bn_stats_kernel<<<grid, 256, 0, s>>>((const __bf16*)x.data_ptr(),
-                                       psum.data_ptr<float>(),
-                                       psumsq.data_ptr<float>(), rows, cvecs,
-                                       cpb, rpb);
-  uint32_t ncv = std::min(cpb, 8u);
-  bn_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
-      psum.data_ptr<float>(), psumsq.data_ptr<float>(),
-      (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
-      running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
-      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(), C, nslabs, ncv, rows,
-      momentum, eps);
-  uint32_t total8 = rows * cvecs;
-  FastDiv dcv;
-  dcv.init(cvecs);
-  uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-  bn_apply_kernel<<<blocks, 256, 0, s>>>(
-      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(),
-      residual.has_value() ? (const __bf16*)residual->data_ptr() : nullptr,
-      total8, dcv, fuse_relu ? 1 : 0);
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_fwd", {
+    bn_stats_kernel<<<grid, 256, 0, s>>>((const elem_t*)x.data_ptr(),
+                                         psum.data_ptr<float>(),
+                                         psumsq.data_ptr<float>(), rows, cvecs,
+                                         cpb, rpb);
+    uint32_t ncv = std::min(cpb, 8u);
+    bn_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+        psum.data_ptr<float>(), psumsq.data_ptr<float>(),
+        (const elem_t*)gamma.data_ptr(), (const elem_t*)beta.data_ptr(),
+        running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
+        scale.data_ptr<float>(), shift.data_ptr<float>(), C, nslabs, ncv, rows,
+        momentum, eps);
+    uint32_t total8 = rows * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_apply_kernel<<<blocks, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
+        scale.data_ptr<float>(), shift.data_ptr<float>(),
+        residual.has_value() ? (const elem_t*)residual->data_ptr() : nullptr,
+        total8, dcv, fuse_relu ? 1 : 0);
+  });
   return {y, save_mean, save_invstd};
 }
 
@@ -523,19 +543,21 @@ at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
   auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
   auto y = at::empty_like(x);
   auto s = bn_stream();
-  bn_infer_prep_kernel<<<(C + 255) / 256, 256, 0, s>>>(
-      (const __bf16*)gamma.data_ptr(), (const __bf16*)beta.data_ptr(),
-      running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(), C, eps);
-  uint32_t total8 = rows * cvecs;
-  FastDiv dcv;
-  dcv.init(cvecs);
-  uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-  bn_apply_kernel<<<blocks, 256, 0, s>>>(
-      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
-      scale.data_ptr<float>(), shift.data_ptr<float>(),
-      residual.has_value() ? (const __bf16*)residual->data_ptr() : nullptr,
-      total8, dcv, fuse_relu ? 1 : 0);
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_infer", {
+    bn_infer_prep_kernel<<<(C + 255) / 256, 256, 0, s>>>(
+        (const elem_t*)gamma.data_ptr(), (const elem_t*)beta.data_ptr(),
+        running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+        scale.data_ptr<float>(), shift.data_ptr<float>(), C, eps);
+    uint32_t total8 = rows * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_apply_kernel<<<blocks, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
+        scale.data_ptr<float>(), shift.data_ptr<float>(),
+        residual.has_value() ? (const elem_t*)residual->data_ptr() : nullptr,
+        total8, dcv, fuse_relu ? 1 : 0);
+  });
   return y;
 }
 
@@ -561,27 +583,29 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   at::Tensor dres;
   if (want_dres) dres = at::empty_like(x);
   auto s = bn_stream();
-  bn_bwd_stats_kernel<<<grid, 256, 0, s>>>(
-      (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
-      (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
-      save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
-      pdg.data_ptr<float>(), rows, cvecs, cpb, rpb, fuse_relu ? 1 : 0);
-  uint32_t ncv = std::min(cpb, 8u);
-  bn_bwd_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
-      pdb.data_ptr<float>(), pdg.data_ptr<float>(), (__bf16*)dgamma.data_ptr(),
-      (__bf16*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
-      C, nslabs, ncv);
-  uint32_t total8 = rows * cvecs;
-  FastDiv dcv;
-  dcv.init(cvecs);
-  uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-  bn_bwd_dx_kernel<<<blocks, 256, 0, s>>>(
-      (const __bf16*)x.data_ptr(), (const __bf16*)dy.data_ptr(),
-      (const __bf16*)y.data_ptr(), save_mean.data_ptr<float>(),
-      save_invstd.data_ptr<float>(), (const __bf16*)gamma.data_ptr(),
-      tdb.data_ptr<float>(), tdg.data_ptr<float>(), (__bf16*)dx.data_ptr(),
-      want_dres ? (__bf16*)dres.data_ptr() : nullptr, total8, dcv, 1.f / rows,
-      fuse_relu ? 1 : 0);
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_bwd", {
+    bn_bwd_stats_kernel<<<grid, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
+        (const elem_t*)y.data_ptr(), save_mean.data_ptr<float>(),
+        save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
+        pdg.data_ptr<float>(), rows, cvecs, cpb, rpb, fuse_relu ? 1 : 0);
+    uint32_t ncv = std::min(cpb, 8u);
+    bn_bwd_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+        pdb.data_ptr<float>(), pdg.data_ptr<float>(), (elem_t*)dgamma.data_ptr(),
+        (elem_t*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
+        C, nslabs, ncv);
+    uint32_t total8 = rows * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_bwd_dx_kernel<<<blocks, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
+        (const elem_t*)y.data_ptr(), save_mean.data_ptr<float>(),
+        save_invstd.data_ptr<float>(), (const elem_t*)gamma.data_ptr(),
+        tdb.data_ptr<float>(), tdg.data_ptr<float>(), (elem_t*)dx.data_ptr(),
+        want_dres ? (elem_t*)dres.data_ptr() : nullptr, total8, dcv, 1.f / rows,
+        fuse_relu ? 1 : 0);
+  });
   if (want_dres) return {dx, dgamma, dbeta, dres};
   return {dx, dgamma, dbeta};
 }

@@ -12,7 +12,8 @@ namespace dtmx {
 
 static hipStream_t cc_stream() { return at::hip::getCurrentHIPStream().stream(); }
 
-__global__ void quantize_2bit_kernel(const __bf16* __restrict__ grad,
+template <typename elem_t>
+__global__ void quantize_2bit_kernel(const elem_t* __restrict__ grad,
                                      float* __restrict__ residual,
                                      uint32_t* __restrict__ out, size_t words,
                                      size_t n, float threshold) {
@@ -42,8 +43,9 @@ __global__ void quantize_2bit_kernel(const __bf16* __restrict__ grad,
   }
 }
 
+template <typename elem_t>
 __global__ void dequantize_2bit_kernel(const uint32_t* __restrict__ in,
-                                       __bf16* __restrict__ out, size_t words,
+                                       elem_t* __restrict__ out, size_t words,
                                        size_t n, float threshold) {
   size_t w = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -55,22 +57,23 @@ __global__ void dequantize_2bit_kernel(const uint32_t* __restrict__ in,
       size_t i = base + e;
       if (i >= n) break;
       uint32_t code = (packed >> (2 * e)) & 0x3;
-      out[i] = (__bf16)(code == 1 ? threshold : code == 2 ? -threshold : 0.f);
+      out[i] = (elem_t)(code == 1 ? threshold : code == 2 ? -threshold : 0.f);
     }
   }
 }
 
 at::Tensor quantize_2bit(const at::Tensor& grad, at::Tensor residual,
                          double threshold) {
-  TORCH_CHECK(grad.is_cuda() && grad.scalar_type() == at::kBFloat16,
-              "quantize_2bit: bf16 CUDA tensor required");
+  TORCH_CHECK(grad.is_cuda(), "quantize_2bit: CUDA tensor required");
   size_t n = grad.numel();
   size_t words = (n + 15) / 16;
   auto out = at::empty({(long)words}, grad.options().dtype(at::kInt));
   uint32_t blocks = std::min<size_t>((words + 255) / 256, 2048);
-  quantize_2bit_kernel<<<blocks, 256, 0, cc_stream()>>>(
-      (const __bf16*)grad.data_ptr(), residual.data_ptr<float>(),
-      (uint32_t*)out.data_ptr(), words, n, threshold);
+  DTMX_DISPATCH_16(grad.scalar_type(), "quantize_2bit", {
+    quantize_2bit_kernel<<<blocks, 256, 0, cc_stream()>>>(
+        (const elem_t*)grad.data_ptr(), residual.data_ptr<float>(),
+        (uint32_t*)out.data_ptr(), words, n, threshold);
+  });
   return out;
 }
 
@@ -78,9 +81,11 @@ at::Tensor dequantize_2bit(const at::Tensor& packed, long numel, double threshol
   size_t words = packed.numel();
   auto out = at::empty({numel}, packed.options().dtype(at::kBFloat16));
   uint32_t blocks = std::min<size_t>((words + 255) / 256, 2048);
-  dequantize_2bit_kernel<<<blocks, 256, 0, cc_stream()>>>(
-      (const uint32_t*)packed.data_ptr(), (__bf16*)out.data_ptr(), words, numel,
-      threshold);
+  DTMX_DISPATCH_16(out.scalar_type(), "dequantize_2bit", {
+    dequantize_2bit_kernel<<<blocks, 256, 0, cc_stream()>>>(
+        (const uint32_t*)packed.data_ptr(), (elem_t*)out.data_ptr(), words,
+        numel, threshold);
+  });
   return out;
 }
 

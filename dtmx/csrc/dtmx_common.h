@@ -11,9 +11,43 @@
 #define DTMX_XCDS 8
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(2))) float f32x2;
 typedef __attribute__((ext_vector_type(4))) short s16x4;
+
+// 16-bit element traits: bf16 and fp16 share every kernel (the reference's
+// fp16 multi-precision path); the MFMA intrinsic is the only divergence.
+template <typename E>
+struct E8;
+template <>
+struct E8<__bf16> {
+  using v8 = bf16x8;
+  static __device__ __forceinline__ f32x4 mfma(v8 a, v8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <>
+struct E8<_Float16> {
+  using v8 = f16x8;
+  static __device__ __forceinline__ f32x4 mfma(v8 a, v8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
+
+// host-side dispatch over the two 16-bit dtypes
+#define DTMX_DISPATCH_16(TYPE, NAME, ...)                         \
+  do {                                                            \
+    if ((TYPE) == at::kBFloat16) {                                \
+      using elem_t = __bf16;                                      \
+      __VA_ARGS__                                                 \
+    } else if ((TYPE) == at::kHalf) {                             \
+      using elem_t = _Float16;                                    \
+      __VA_ARGS__                                                 \
+    } else {                                                      \
+      TORCH_CHECK(false, NAME, ": dtype must be bfloat16/float16"); \
+    }                                                             \
+  } while (0)
 
 // 16-byte async global->LDS copy (one lane's 16 B; LDS dest is
 // wave-uniform base + lane*16 — cdna_hip_programming.md §5).

@@ -30,13 +30,15 @@ namespace dtmx {
 // block's rows are fixed across the whole K loop, so the (n,p,q)/(n,h,w)
 // FastDiv decode does not belong in the per-piece staging path (measured
 // gather overhead vs same-shape dense GEMM ~30-40%).
+template <typename elem_t>
 struct DenseP {
-  const __bf16* base;
-  const __bf16* zero;
+  using elem = elem_t;
+  const elem_t* base;
+  const elem_t* zero;
   uint32_t M, K;  // rows, k extent (elements, multiple of 8)
   uint32_t ld;    // row stride in elements
   struct Row {
-    const __bf16* p;  // nullptr = out-of-range row
+    const elem_t* p;  // nullptr = out-of-range row
   };
   __device__ __forceinline__ Row row(uint32_t m) const {
     return {m < M ? base + (size_t)m * ld : nullptr};
@@ -49,15 +51,17 @@ struct DenseP {
 };
 
 // conv forward A: row m = (n,p,q) output pixel, k = (r,s,c), x is NHWC.
+template <typename elem_t>
 struct ConvFwdA {
-  const __bf16* x;
-  const __bf16* zero;
+  using elem = elem_t;
+  const elem_t* x;
+  const elem_t* zero;
   uint32_t M, Ktot;              // M = N*P*Q, Ktot = R*S*C
   uint32_t C, H, W, Q, S;
   int u, v, ph, pw;              // stride, padding
   FastDiv dQ, dPQ, dC, dS;
   struct Row {
-    const __bf16* pixel0;  // &x[n][p*u-ph][q*v-pw][0] (may point out of range)
+    const elem_t* pixel0;  // &x[n][p*u-ph][q*v-pw][0] (may point out of range)
     int ih0, iw0;          // p*u-ph, q*v-pw
     uint32_t valid;
   };
@@ -85,15 +89,17 @@ struct ConvFwdA {
 
 // conv dgrad A: row m = (n,h,w) input pixel, k = (r,s,ko), dy is NHWC.
 // dx[n,h,w,c] = sum_{r,s,ko} dy[n,(h+ph-r)/u,(w+pw-s)/v,ko] * w[ko,r,s,c]
+template <typename elem_t>
 struct ConvDgradA {
-  const __bf16* dy;
-  const __bf16* zero;
+  using elem = elem_t;
+  const elem_t* dy;
+  const elem_t* zero;
   uint32_t M, Ktot;              // M = N*H*W, Ktot = R*S*Kout
   uint32_t Ko, H, W, P, Q, S;
   int u, v, ph, pw;
   FastDiv dW_, dHW, dKo, dS;
   struct Row {
-    const __bf16* base_n;  // &dy[n][0][0][0]
+    const elem_t* base_n;  // &dy[n][0][0][0]
     int hp, wp;            // h+ph, w+pw
     uint32_t valid;
   };
@@ -126,25 +132,28 @@ struct ConvDgradA {
 
 // ---------------------------------------------------------------- epilogues
 
+template <typename elem_t>
 struct EpiBF16 {
+  using elem = elem_t;
+  using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
-  __bf16* c;
+  elem_t* c;
   const float* bias;  // nullable
   uint32_t M, N;
   int relu;
   // coalesced row-chunk store used by the kernel's LDS-staged epilogue
   __device__ __forceinline__ void store_chunk(uint32_t m, uint32_t n0,
-                                              bf16x8 v) const {
+                                              V8 v) const {
     if (m >= M || n0 >= N) return;  // N % 8 == 0: chunks never straddle
     if (bias || relu) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float f = (float)v[e] + (bias ? bias[n0 + e] : 0.f);
         if (relu) f = fmaxf(f, 0.f);
-        v[e] = (__bf16)f;
+        v[e] = (elem_t)f;
       }
     }
-    *(bf16x8*)(c + (size_t)m * N + n0) = v;
+    *(V8*)(c + (size_t)m * N + n0) = v;
   }
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,
@@ -162,7 +171,7 @@ struct EpiBF16 {
           if (m >= M) continue;
           float v = acc[i][j][r] + b;
           if (relu) v = fmaxf(v, 0.f);
-          c[(size_t)m * N + n] = (__bf16)v;
+          c[(size_t)m * N + n] = (elem_t)v;
         }
       }
   }
@@ -170,31 +179,37 @@ struct EpiBF16 {
 
 // 1x1 stride-u dgrad: rows m = (n,p,q) of the dense dy@W^T GEMM scatter to
 // input pixels (n, p*u, q*v); everything else in dx stays zero.
+template <typename elem_t>
 struct EpiBF16Scatter {
+  using elem = elem_t;
+  using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
-  __bf16* dx;
+  elem_t* dx;
   uint32_t M, N;  // M = NPQ, N = C
   uint32_t H, W, Q;
   int u, v;
   FastDiv dQ, dPQ;
   __device__ __forceinline__ void store_chunk(uint32_t m, uint32_t n0,
-                                              bf16x8 val) const {
+                                              V8 val) const {
     if (m >= M || n0 >= N) return;
     uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
     uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
     size_t off = (((size_t)n * H + p * u) * W + q * v) * N + n0;
-    *(bf16x8*)(dx + off) = val;
+    *(V8*)(dx + off) = val;
   }
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t,
                                         uint32_t, uint32_t) const {}
 };
 
-struct EpiAtomicF32 {  // split-K partial accumulation (conv wgrad)
+template <typename elem_t>
+struct EpiAtomicF32 {
+  using elem = elem_t;
+  using V8 = typename E8<elem_t>::v8;  // split-K partial accumulation (conv wgrad)
   static constexpr bool kLdsStage = false;
   float* c;
   uint32_t M, N;
-  __device__ __forceinline__ void store_chunk(uint32_t, uint32_t, bf16x8) const {}
+  __device__ __forceinline__ void store_chunk(uint32_t, uint32_t, V8) const {}
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,
                                         uint32_t n0, uint32_t lane) const {
@@ -220,6 +235,8 @@ template <int NJ, class PA, class PB, class EPI>
 __launch_bounds__(256, 2) __global__
 void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
                     uint32_t tiles_n, uint32_t kt_per_slice) {
+  using elem_t = typename PA::elem;
+  using V8 = typename E8<elem_t>::v8;
   // NJ = 16-col fragments per wave: BN tile = NJ*32 (128 for square work,
   // 64 for narrow-N layers like resnet's K_out=64 convs where a 128 tile
   // wastes half the MFMA work).
@@ -229,7 +246,7 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   const uint32_t kt0 = blockIdx.y * kt_per_slice;
   const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
   if (kt0 >= ktiles_total) return;
-  __shared__ __bf16 smem[2][(128 + BN) * 64];  // A then B images per buffer
+  __shared__ elem_t smem[2][(128 + BN) * 64];  // A then B images per buffer
   constexpr uint32_t B_OFF = 128 * 64;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
@@ -293,26 +310,25 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[NJ];
+      V8 af[4], bfr[NJ];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         uint32_t row = wr + i * 16 + (lane & 15);
         uint32_t colb = (kk * 64 + ((lane >> 4) << 4)) ^ ((row & 7) << 4);
-        af[i] = *(const bf16x8*)((const char*)&smem[cur][0] + row * 128 + colb);
+        af[i] = *(const V8*)((const char*)&smem[cur][0] + row * 128 + colb);
       }
 #pragma unroll
       for (int j = 0; j < NJ; ++j) {
         uint32_t row = wc + j * 16 + (lane & 15);
         uint32_t colb = (kk * 64 + ((lane >> 4) << 4)) ^ ((row & 7) << 4);
-        bfr[j] = *(const bf16x8*)((const char*)&smem[cur][B_OFF] + row * 128 + colb);
+        bfr[j] = *(const V8*)((const char*)&smem[cur][B_OFF] + row * 128 + colb);
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < NJ; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j],
-                                                              acc[i][j], 0, 0, 0);
+          acc[i][j] = E8<elem_t>::mfma(af[i], bfr[j], acc[i][j]);
       __builtin_amdgcn_s_setprio(0);
     }
     __builtin_amdgcn_s_barrier();        // every wave done reading smem[cur]
@@ -323,7 +339,7 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     // the direct fragment store is 16-64 half-coalesced 2-B stores per lane
     // (store-issue-bound; cdna_hip_programming.md T21 diagnosis).
     __syncthreads();  // nothing in flight; reuse smem[0] as [128][BN] bf16
-    __bf16* ct = &smem[0][0];
+    elem_t* ct = &smem[0][0];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -332,14 +348,14 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           uint32_t row = wr + i * 16 + ((lane >> 4) << 2) + r;
-          ct[row * BN + col] = (__bf16)acc[i][j][r];
+          ct[row * BN + col] = (elem_t)acc[i][j][r];
         }
       }
     __syncthreads();
     constexpr uint32_t CHUNKS = 128 * BN / 8;
     for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
       uint32_t row = idx / (BN / 8), nc = idx % (BN / 8);
-      epi.store_chunk(bm + row, bn + nc * 8, *(const bf16x8*)(ct + row * BN + nc * 8));
+      epi.store_chunk(bm + row, bn + nc * 8, *(const V8*)(ct + row * BN + nc * 8));
     }
   } else {
     epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
@@ -353,9 +369,11 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
 // LDS images [64 kd][BM|BN]; fragments gathered by scalar u16 column reads
 // (8 per fragment), hidden under the MFMA stream.
 
-struct WgradDyA {  // A: [kd = npq][m = ko] = dy NHWC as-is
-  const __bf16* dy;
-  const __bf16* zero;
+template <typename elem_t>
+struct WgradDyA {
+  using elem = elem_t;  // A: [kd = npq][m = ko] = dy NHWC as-is
+  const elem_t* dy;
+  const elem_t* zero;
   uint32_t Kd, Mdim;
   __device__ __forceinline__ const void* addr(uint32_t kd, uint32_t c8) const {
     if (kd >= Kd || c8 * 8 >= Mdim) return zero;
@@ -363,9 +381,11 @@ struct WgradDyA {  // A: [kd = npq][m = ko] = dy NHWC as-is
   }
 };
 
-struct WgradXcolB {  // B: [kd = npq][n = (r,s,c)] gathered from x NHWC
-  const __bf16* x;
-  const __bf16* zero;
+template <typename elem_t>
+struct WgradXcolB {
+  using elem = elem_t;  // B: [kd = npq][n = (r,s,c)] gathered from x NHWC
+  const elem_t* x;
+  const elem_t* zero;
   uint32_t Kd, Ndim;  // Kd = NPQ, Ndim = R*S*C
   uint32_t C, H, W, Q, S;
   int u, v, ph, pw;
@@ -388,13 +408,15 @@ template <int NJ, class PA, class PB, class EPI>
 __launch_bounds__(256, 2) __global__
 void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
                     uint32_t tiles_n, uint32_t kt_per_slice) {
+  using elem_t = typename PA::elem;
+  using V8 = typename E8<elem_t>::v8;
   constexpr uint32_t BN = NJ * 32;
   const uint32_t kt0 = blockIdx.y * kt_per_slice;
   const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
   if (kt0 >= ktiles_total) return;
   // 16-B alignment: ds_read_b64_tr_b16 at a misaligned address silently
   // returns the 8-aligned address's data (G17)
-  __shared__ __attribute__((aligned(16))) __bf16 smem[2][64 * (128 + BN)];
+  __shared__ __attribute__((aligned(16))) elem_t smem[2][64 * (128 + BN)];
   constexpr uint32_t B_OFF = 64 * 128;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
@@ -449,7 +471,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[NJ];
+      V8 af[4], bfr[NJ];
       // hardware transpose reads: each 16-lane group reads a [4 kd][16 col]
       // block of the contraction-major image and receives it column-wise —
       // 2 ds_read_b64_tr_b16 per fragment instead of 8 scalar u16 reads
@@ -460,7 +482,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
       const uint32_t kg = kk * 32 + ((lane >> 4) << 3) + kq;
       typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_;
       union Frag {
-        bf16x8 f;
+        V8 f;
         u32x2_ h[2];
       };
       {
@@ -546,8 +568,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < NJ; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j],
-                                                              acc[i][j], 0, 0, 0);
+          acc[i][j] = E8<elem_t>::mfma(af[i], bfr[j], acc[i][j]);
       __builtin_amdgcn_s_setprio(0);
     }
     __builtin_amdgcn_s_barrier();
@@ -561,21 +582,25 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
 // (linear dgrad/wgrad operands, conv-wgrad dy^T and gathered im2col^T).
 // C % 8 == 0 required; rows beyond M (or gather-invalid) write zeros.
 
+template <typename elem_t>
 struct IdentityRows {
-  const __bf16* base;
+  using elem = elem_t;
+  const elem_t* base;
   uint32_t M, ld;
-  __device__ __forceinline__ const __bf16* row(uint32_t m) const {
+  __device__ __forceinline__ const elem_t* row(uint32_t m) const {
     return m < M ? base + (size_t)m * ld : nullptr;
   }
 };
 
 // rows of the im2col matrix for a FIXED (r,s): row m = (n,p,q) -> x pixel
+template <typename elem_t>
 struct Im2colRows {
-  const __bf16* x;
+  using elem = elem_t;
+  const elem_t* x;
   uint32_t M, C, H, W, Q;
   int u, v, ph, pw, r, s;
   FastDiv dQ, dPQ;
-  __device__ __forceinline__ const __bf16* row(uint32_t m) const {
+  __device__ __forceinline__ const elem_t* row(uint32_t m) const {
     if (m >= M) return nullptr;
     uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
     uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
@@ -587,37 +612,40 @@ struct Im2colRows {
 };
 
 template <class ROWS>
-__global__ void transpose_rowgather_kernel(ROWS rows, __bf16* out, uint32_t C,
-                                           uint32_t out_ld, uint32_t row_off,
-                                           uint32_t tiles_m) {
-  __shared__ __bf16 tile[64][72];  // +8 pad (16 B) against bank conflicts
+__global__ void transpose_rowgather_kernel(ROWS rows, typename ROWS::elem* out,
+                                           uint32_t C, uint32_t out_ld,
+                                           uint32_t row_off, uint32_t tiles_m) {
+  using elem_t = typename ROWS::elem;
+  using V8 = typename E8<elem_t>::v8;
+  __shared__ elem_t tile[64][72];  // +8 pad (16 B) against bank conflicts
   const uint32_t t = threadIdx.x;
   const uint32_t bm = (blockIdx.x % tiles_m) * 64;   // input row tile
   const uint32_t bc = (blockIdx.x / tiles_m) * 64;   // input col tile
 #pragma unroll
   for (int it = 0; it < 2; ++it) {
     uint32_t rl = it * 32 + (t >> 3), cl = (t & 7) * 8;
-    const __bf16* p = rows.row(bm + rl);
-    bf16x8 v = {};
-    if (p && bc + cl < C) v = *(const bf16x8*)(p + bc + cl);
-    *(bf16x8*)&tile[rl][cl] = v;
+    const elem_t* p = rows.row(bm + rl);
+    V8 v = {};
+    if (p && bc + cl < C) v = *(const V8*)(p + bc + cl);
+    *(V8*)&tile[rl][cl] = v;
   }
   __syncthreads();
 #pragma unroll
   for (int it = 0; it < 2; ++it) {
     uint32_t cl = it * 32 + (t >> 3), ml = (t & 7) * 8;
     if (bc + cl >= C || bm + ml >= out_ld) continue;
-    bf16x8 v;
+    V8 v;
 #pragma unroll
     for (int e = 0; e < 8; ++e) v[e] = tile[ml + e][cl];
-    *(bf16x8*)&out[(size_t)(row_off + bc + cl) * out_ld + bm + ml] = v;
+    *(V8*)&out[(size_t)(row_off + bc + cl) * out_ld + bm + ml] = v;
   }
 }
 
 // ------------------------------------------------------- im2col (small-C)
 // Explicit im2col for C % 8 != 0 (the 3-channel stem): out[m][k] with
 // k = (r*S+s)*C + c, zero-padded to Kpad (reference nn/im2col.cuh analog).
-__global__ void im2col_kernel(const __bf16* x, __bf16* out, uint32_t M,
+template <typename elem_t>
+__global__ void im2col_kernel(const elem_t* x, elem_t* out, uint32_t M,
                               uint32_t Kpad, uint32_t Ktot, uint32_t C,
                               uint32_t H, uint32_t W, uint32_t Q, uint32_t S,
                               int u, int v, int ph, int pw, FastDiv dQ,
@@ -626,7 +654,7 @@ __global__ void im2col_kernel(const __bf16* x, __bf16* out, uint32_t M,
   size_t total = (size_t)M * Kpad;
   for (; idx < total; idx += (size_t)gridDim.x * blockDim.x) {
     uint32_t m = idx / Kpad, k = idx % Kpad;
-    __bf16 val = (__bf16)0.f;
+    elem_t val = (elem_t)0.f;
     if (k < Ktot) {
       uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
       uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
@@ -641,25 +669,29 @@ __global__ void im2col_kernel(const __bf16* x, __bf16* out, uint32_t M,
   }
 }
 
+template <typename elem_t>
 __global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
-                                     __bf16* __restrict__ out, size_t total8) {
+                                     elem_t* __restrict__ out, size_t total8) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
-    bf16x8 o;
+    V8 o;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) o[e] = (__bf16)in[i * 8 + e];
-    *(bf16x8*)(out + i * 8) = o;
+    for (int e = 0; e < 8; ++e) o[e] = (elem_t)in[i * 8 + e];
+    *(V8*)(out + i * 8) = o;
   }
 }
 
 // ============================================================== host side ==
 
-static const __bf16* zero_page(const at::Tensor& like) {
+template <typename elem_t>
+static const elem_t* zero_page(const at::Tensor& like) {
+  // a zeroed 16-bit pattern reads as 0.0 in bf16 and fp16 alike
   static at::Tensor z;
   if (!z.defined() || z.device() != like.device())
     z = at::zeros({64}, like.options().dtype(at::kBFloat16));
-  return (const __bf16*)z.data_ptr();
+  return (const elem_t*)z.data_ptr();
 }
 
 static inline uint32_t ceil_div(uint32_t a, uint32_t b) { return (a + b - 1) / b; }
@@ -668,8 +700,10 @@ static hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
-#define CHECK_BF16_CUDA(t) \
-  TORCH_CHECK((t).is_cuda() && (t).scalar_type() == at::kBFloat16, #t " must be a CUDA bf16 tensor")
+#define CHECK_BF16_CUDA(t)                                                   \
+  TORCH_CHECK((t).is_cuda() && ((t).scalar_type() == at::kBFloat16 ||        \
+                                (t).scalar_type() == at::kHalf),             \
+              #t " must be a CUDA bf16/fp16 tensor")
 
 // small grids underfill 256 CUs: callers pass want_splitk=true to let the
 // launcher split K into an fp32 atomic buffer (the caller casts back).
@@ -725,27 +759,32 @@ static at::Tensor pad_cols8(const at::Tensor& t) {
 
 at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
                       const c10::optional<at::Tensor>& bias) {
-  CHECK_BF16_CUDA(x);
-  CHECK_BF16_CUDA(w);
-  TORCH_CHECK(x.size(1) == w.size(1), "linear: in_features mismatch (x ",
-              x.size(1), " vs w ", w.size(1), ")");
-  auto xc = pad_cols8(x);
-  auto wc = pad_cols8(w);
-  uint32_t M = xc.size(0), K = xc.size(1), N = wc.size(0);
-  auto y = at::empty({(long)M, (long)N}, x.options());
-  at::Tensor bias_f;
-  const float* bp = nullptr;
-  if (bias.has_value()) {
-    bias_f = bias->to(at::kFloat).contiguous();
-    bp = bias_f.data_ptr<float>();
-  }
-  DenseP pa{(const __bf16*)xc.data_ptr(), zero_page(x), M, K, K};
-  DenseP pb{(const __bf16*)wc.data_ptr(), zero_page(x), N, K, K};
-  EpiBF16 epi{(__bf16*)y.data_ptr(), bp, M, N, 0};
-  launch_gemm(pa, pb, epi, M, N, K);
-  return y;
+  DTMX_DISPATCH_16(x.scalar_type(), "linear_fwd", {
+    CHECK_BF16_CUDA(x);
+    CHECK_BF16_CUDA(w);
+    TORCH_CHECK(x.size(1) == w.size(1), "linear: in_features mismatch (x ",
+                x.size(1), " vs w ", w.size(1), ")");
+    auto xc = pad_cols8(x);
+    auto wc = pad_cols8(w);
+    uint32_t M = xc.size(0), K = xc.size(1), N = wc.size(0);
+    auto y = at::empty({(long)M, (long)N}, x.options());
+    at::Tensor bias_f;
+    const float* bp = nullptr;
+    if (bias.has_value()) {
+      bias_f = bias->to(at::kFloat).contiguous();
+      bp = bias_f.data_ptr<float>();
+    }
+    DenseP<elem_t> pa{(const elem_t*)xc.data_ptr(), zero_page<elem_t>(x), M, K, K};
+    DenseP<elem_t> pb{(const elem_t*)wc.data_ptr(), zero_page<elem_t>(x), N, K, K};
+    EpiBF16<elem_t> epi{(elem_t*)y.data_ptr(), bp, M, N, 0};
+    launch_gemm(pa, pb, epi, M, N, K);
+    return y;
+
+  });
+  return at::Tensor();
 }
 
+template <typename elem_t>
 static at::Tensor transpose2d(const at::Tensor& in, uint32_t out_ld_pad = 0) {
   // [M][C] -> [C][Mpad64?]: out_ld defaults to M rounded to 8
   CHECK_BF16_CUDA(in);
@@ -755,39 +794,47 @@ static at::Tensor transpose2d(const at::Tensor& in, uint32_t out_ld_pad = 0) {
   TORCH_CHECK(C % 8 == 0, "transpose2d: cols must be a multiple of 8");
   auto out = at::empty({(long)C, (long)out_ld}, in.options());
   uint32_t tiles_m = ceil_div(out_ld, 64), tiles_c = ceil_div(C, 64);
-  IdentityRows rows{(const __bf16*)inc.data_ptr(), M, C};
-  transpose_rowgather_kernel<IdentityRows>
+  IdentityRows<elem_t> rows{(const elem_t*)inc.data_ptr(), M, C};
+  transpose_rowgather_kernel<IdentityRows<elem_t>>
       <<<tiles_m * tiles_c, 256, 0, cur_stream()>>>(
-          rows, (__bf16*)out.data_ptr(), C, out_ld, 0, tiles_m);
+          rows, (elem_t*)out.data_ptr(), C, out_ld, 0, tiles_m);
   return out;
 }
 
 at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
-  CHECK_BF16_CUDA(dy);
-  uint32_t M = dy.size(0), N = dy.size(1), K = w.size(1);
-  uint32_t Npad = ((N + 7) / 8) * 8;
-  auto dyc = pad_cols8(dy);                      // [M][Npad]
-  auto wt = transpose2d(pad_cols8(w), Npad);     // [Kpad8][Npad] (rows>N zero)
-  auto dx = at::empty({(long)M, (long)K}, dy.options());
-  DenseP pa{(const __bf16*)dyc.data_ptr(), zero_page(dy), M, Npad, Npad};
-  DenseP pb{(const __bf16*)wt.data_ptr(), zero_page(dy), K, Npad, Npad};
-  EpiBF16 epi{(__bf16*)dx.data_ptr(), nullptr, M, K, 0};
-  launch_gemm(pa, pb, epi, M, K, Npad);
-  return dx;
+  DTMX_DISPATCH_16(dy.scalar_type(), "linear_dgrad", {
+    CHECK_BF16_CUDA(dy);
+    uint32_t M = dy.size(0), N = dy.size(1), K = w.size(1);
+    uint32_t Npad = ((N + 7) / 8) * 8;
+    auto dyc = pad_cols8(dy);                      // [M][Npad]
+    auto wt = transpose2d<elem_t>(pad_cols8(w), Npad);     // [Kpad8][Npad] (rows>N zero)
+    auto dx = at::empty({(long)M, (long)K}, dy.options());
+    DenseP<elem_t> pa{(const elem_t*)dyc.data_ptr(), zero_page<elem_t>(dy), M, Npad, Npad};
+    DenseP<elem_t> pb{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy), K, Npad, Npad};
+    EpiBF16<elem_t> epi{(elem_t*)dx.data_ptr(), nullptr, M, K, 0};
+    launch_gemm(pa, pb, epi, M, K, Npad);
+    return dx;
+
+  });
+  return at::Tensor();
 }
 
 at::Tensor linear_wgrad(const at::Tensor& dy, const at::Tensor& x) {
-  CHECK_BF16_CUDA(dy);
-  uint32_t M = dy.size(0), N = dy.size(1), K = x.size(1);
-  uint32_t Mpad = ((M + 7) / 8) * 8;
-  auto dyt = transpose2d(pad_cols8(dy), Mpad);  // [Npad][Mpad]
-  auto xt = transpose2d(pad_cols8(x), Mpad);    // [Kpad][Mpad]
-  auto dw = at::empty({(long)N, (long)K}, dy.options());
-  DenseP pa{(const __bf16*)dyt.data_ptr(), zero_page(dy), N, Mpad, Mpad};
-  DenseP pb{(const __bf16*)xt.data_ptr(), zero_page(dy), K, Mpad, Mpad};
-  EpiBF16 epi{(__bf16*)dw.data_ptr(), nullptr, N, K, 0};
-  launch_gemm(pa, pb, epi, N, K, Mpad);
-  return dw;
+  DTMX_DISPATCH_16(dy.scalar_type(), "linear_wgrad", {
+    CHECK_BF16_CUDA(dy);
+    uint32_t M = dy.size(0), N = dy.size(1), K = x.size(1);
+    uint32_t Mpad = ((M + 7) / 8) * 8;
+    auto dyt = transpose2d<elem_t>(pad_cols8(dy), Mpad);  // [Npad][Mpad]
+    auto xt = transpose2d<elem_t>(pad_cols8(x), Mpad);    // [Kpad][Mpad]
+    auto dw = at::empty({(long)N, (long)K}, dy.options());
+    DenseP<elem_t> pa{(const elem_t*)dyt.data_ptr(), zero_page<elem_t>(dy), N, Mpad, Mpad};
+    DenseP<elem_t> pb{(const elem_t*)xt.data_ptr(), zero_page<elem_t>(dy), K, Mpad, Mpad};
+    EpiBF16<elem_t> epi{(elem_t*)dw.data_ptr(), nullptr, N, K, 0};
+    launch_gemm(pa, pb, epi, N, K, Mpad);
+    return dw;
+
+  });
+  return at::Tensor();
 }
 
 // ------------------------------------------------------------- conv fwd
@@ -803,6 +850,7 @@ static void conv_out_dims(uint32_t H, uint32_t W, uint32_t R, uint32_t S,
 template <class PA, class PB>
 static bool smallgrid_splitk(const PA& pa, const PB& pb, at::Tensor& out_bf16,
                              uint32_t M, uint32_t N, uint32_t K) {
+  using elem_t = typename PA::elem;
   uint32_t tiles = ceil_div(M, 128) * ceil_div(N, N <= 64 ? 64 : 128);
   uint32_t ktiles = ceil_div(K, 64);
   // only for severely underfilled grids: the fp32-atomic + cast overhead and
@@ -811,123 +859,131 @@ static bool smallgrid_splitk(const PA& pa, const PB& pb, at::Tensor& out_bf16,
   if (tiles >= 160 || ktiles < 8) return false;
   uint32_t splitk = std::min(ktiles / 4, std::max<uint32_t>(2, 1024 / tiles));
   auto acc = at::zeros({(long)M, (long)N}, out_bf16.options().dtype(at::kFloat));
-  EpiAtomicF32 epi{acc.data_ptr<float>(), M, N};
+  EpiAtomicF32<elem_t> epi{acc.data_ptr<float>(), M, N};
   launch_gemm(pa, pb, epi, M, N, K, splitk);
   size_t t8 = (size_t)M * N / 8;  // N % 8 == 0 everywhere this is used
   cast_f32_bf16_kernel<<<std::min<size_t>((t8 + 255) / 256, 2048), 256, 0,
                          cur_stream()>>>(acc.data_ptr<float>(),
-                                         (__bf16*)out_bf16.data_ptr(), t8);
+                                         (elem_t*)out_bf16.data_ptr(), t8);
   return true;
 }
 
 at::Tensor conv_fwd(const at::Tensor& x, const at::Tensor& w, long stride,
                     long pad) {
-  CHECK_BF16_CUDA(x);
-  CHECK_BF16_CUDA(w);
-  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x must be NHWC");
-  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast), "w must be KRSC");
-  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
-  uint32_t Ko = w.size(0), R = w.size(2), S = w.size(3);
-  uint32_t P, Q;
-  conv_out_dims(H, W_, R, S, stride, pad, P, Q);
-  auto y = at::empty({(long)N, (long)Ko, (long)P, (long)Q},
-                     x.options(), at::MemoryFormat::ChannelsLast);
-  uint32_t M = N * P * Q, Ktot = R * S * C;
-  EpiBF16 epi{(__bf16*)y.data_ptr(), nullptr, M, Ko, 0};
-  if (R == 1 && S == 1 && stride == 1 && pad == 0 && C % 8 == 0) {
-    // 1x1/s1: the im2col matrix IS x — pure dense GEMM, no gather decode
-    DenseP pa{(const __bf16*)x.data_ptr(), zero_page(x), M, C, C};
-    DenseP pb{(const __bf16*)w.data_ptr(), zero_page(x), Ko, C, C};
-    if (!smallgrid_splitk(pa, pb, y, M, Ko, C))
-      launch_gemm(pa, pb, epi, M, Ko, C);
-  } else if (C % 8 == 0) {
-    DenseP pb{(const __bf16*)w.data_ptr(), zero_page(x), Ko, Ktot, Ktot};
-    ConvFwdA pa;
-    pa.x = (const __bf16*)x.data_ptr();
-    pa.zero = zero_page(x);
-    pa.M = M; pa.Ktot = Ktot; pa.C = C; pa.H = H; pa.W = W_; pa.Q = Q; pa.S = S;
-    pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
-    pa.dQ.init(Q); pa.dPQ.init(P * Q); pa.dC.init(C); pa.dS.init(S);
-    if (!smallgrid_splitk(pa, pb, y, M, Ko, Ktot))
-      launch_gemm(pa, pb, epi, M, Ko, Ktot);
-  } else {
-    // small-C path (3-channel stem): materialized im2col, then dense GEMM.
-    uint32_t Kpad = ((Ktot + 63) / 64) * 64;
-    auto col = at::empty({(long)M, (long)Kpad}, x.options());
-    FastDiv dQ, dPQ, dC, dS;
-    dQ.init(Q); dPQ.init(P * Q); dC.init(C); dS.init(S);
-    size_t total = (size_t)M * Kpad;
-    uint32_t blocks = std::min<size_t>((total + 255) / 256, 16384);
-    im2col_kernel<<<blocks, 256, 0, cur_stream()>>>(
-        (const __bf16*)x.data_ptr(), (__bf16*)col.data_ptr(), M, Kpad, Ktot, C,
-        H, W_, Q, S, stride, stride, pad, pad, dQ, dPQ, dC, dS);
-    // zero-pad the weight rows to Kpad so the k tail multiplies 0*0, not 0*NaN
-    auto wpad = at::constant_pad_nd(
-        w.permute({0, 2, 3, 1}).reshape({(long)Ko, (long)Ktot}),
-        {0, (long)(Kpad - Ktot)}, 0.0).contiguous();
-    DenseP pa{(const __bf16*)col.data_ptr(), zero_page(x), M, Kpad, Kpad};
-    DenseP pb{(const __bf16*)wpad.data_ptr(), zero_page(x), Ko, Kpad, Kpad};
-    launch_gemm(pa, pb, epi, M, Ko, Kpad);
-  }
-  return y;
+  DTMX_DISPATCH_16(x.scalar_type(), "conv_fwd", {
+    CHECK_BF16_CUDA(x);
+    CHECK_BF16_CUDA(w);
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x must be NHWC");
+    TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast), "w must be KRSC");
+    uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    uint32_t Ko = w.size(0), R = w.size(2), S = w.size(3);
+    uint32_t P, Q;
+    conv_out_dims(H, W_, R, S, stride, pad, P, Q);
+    auto y = at::empty({(long)N, (long)Ko, (long)P, (long)Q},
+                       x.options(), at::MemoryFormat::ChannelsLast);
+    uint32_t M = N * P * Q, Ktot = R * S * C;
+    EpiBF16<elem_t> epi{(elem_t*)y.data_ptr(), nullptr, M, Ko, 0};
+    if (R == 1 && S == 1 && stride == 1 && pad == 0 && C % 8 == 0) {
+      // 1x1/s1: the im2col matrix IS x — pure dense GEMM, no gather decode
+      DenseP<elem_t> pa{(const elem_t*)x.data_ptr(), zero_page<elem_t>(x), M, C, C};
+      DenseP<elem_t> pb{(const elem_t*)w.data_ptr(), zero_page<elem_t>(x), Ko, C, C};
+      if (!smallgrid_splitk(pa, pb, y, M, Ko, C))
+        launch_gemm(pa, pb, epi, M, Ko, C);
+    } else if (C % 8 == 0) {
+      DenseP<elem_t> pb{(const elem_t*)w.data_ptr(), zero_page<elem_t>(x), Ko, Ktot, Ktot};
+      ConvFwdA<elem_t> pa;
+      pa.x = (const elem_t*)x.data_ptr();
+      pa.zero = zero_page<elem_t>(x);
+      pa.M = M; pa.Ktot = Ktot; pa.C = C; pa.H = H; pa.W = W_; pa.Q = Q; pa.S = S;
+      pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
+      pa.dQ.init(Q); pa.dPQ.init(P * Q); pa.dC.init(C); pa.dS.init(S);
+      if (!smallgrid_splitk(pa, pb, y, M, Ko, Ktot))
+        launch_gemm(pa, pb, epi, M, Ko, Ktot);
+    } else {
+      // small-C path (3-channel stem): materialized im2col, then dense GEMM.
+      uint32_t Kpad = ((Ktot + 63) / 64) * 64;
+      auto col = at::empty({(long)M, (long)Kpad}, x.options());
+      FastDiv dQ, dPQ, dC, dS;
+      dQ.init(Q); dPQ.init(P * Q); dC.init(C); dS.init(S);
+      size_t total = (size_t)M * Kpad;
+      uint32_t blocks = std::min<size_t>((total + 255) / 256, 16384);
+      im2col_kernel<<<blocks, 256, 0, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)col.data_ptr(), M, Kpad, Ktot, C,
+          H, W_, Q, S, stride, stride, pad, pad, dQ, dPQ, dC, dS);
+      // zero-pad the weight rows to Kpad so the k tail multiplies 0*0, not 0*NaN
+      auto wpad = at::constant_pad_nd(
+          w.permute({0, 2, 3, 1}).reshape({(long)Ko, (long)Ktot}),
+          {0, (long)(Kpad - Ktot)}, 0.0).contiguous();
+      DenseP<elem_t> pa{(const elem_t*)col.data_ptr(), zero_page<elem_t>(x), M, Kpad, Kpad};
+      DenseP<elem_t> pb{(const elem_t*)wpad.data_ptr(), zero_page<elem_t>(x), Ko, Kpad, Kpad};
+      launch_gemm(pa, pb, epi, M, Ko, Kpad);
+    }
+    return y;
+
+  });
+  return at::Tensor();
 }
 
 // ------------------------------------------------------------- conv dgrad
 
 at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
                       long pad, long H, long W_) {
-  CHECK_BF16_CUDA(dy);
-  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy must be NHWC");
-  uint32_t N = dy.size(0), Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
-  uint32_t C = w.size(1), R = w.size(2), S = w.size(3);
-  // pad the out-channel (k-piece) dim to a multiple of 8 when needed
-  at::Tensor dyk = dy.permute({0, 2, 3, 1});        // NHWC view (contiguous)
-  at::Tensor wtk = w.permute({1, 2, 3, 0});          // (C,R,S,K)
-  if (Ko % 8) {
-    long padk = 8 - (Ko % 8);
-    dyk = at::constant_pad_nd(dyk, {0, padk}, 0.0);
-    wtk = at::constant_pad_nd(wtk, {0, padk}, 0.0);
-    Ko += padk;
-  }
-  auto dyc = dyk.contiguous();
-  auto wt = wtk.contiguous();  // W^T in (C,R,S,Ko) dense layout
-  auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
-                      at::MemoryFormat::ChannelsLast);
-  uint32_t M = N * H * W_, Ktot = R * S * Ko;
-  if (R == 1 && S == 1 && pad == 0 && C % 8 == 0) {
-    // 1x1 dgrad: dense dy @ W^T. Stride 1 writes rows directly; stride u>1
-    // scatters row (n,p,q) to pixel (n, p*u, q*v) of a zeroed dx — the
-    // gathered formulation wastes 1-1/u^2 of its blocks on all-zero rows
-    // (measured 107 TF vs ~300 dense).
-    uint32_t Mn = N * P * Q;
-    DenseP pad_{(const __bf16*)dyc.data_ptr(), zero_page(dy), Mn, Ko, Ko};
-    DenseP pbd{(const __bf16*)wt.data_ptr(), zero_page(dy), C, Ko, Ko};
-    if (stride == 1) {
-      EpiBF16 epid{(__bf16*)dx.data_ptr(), nullptr, Mn, C, 0};
-      if (!smallgrid_splitk(pad_, pbd, dx, Mn, C, Ko))
-        launch_gemm(pad_, pbd, epid, Mn, C, Ko);
-    } else {
-      dx.zero_();
-      EpiBF16Scatter epis;
-      epis.dx = (__bf16*)dx.data_ptr();
-      epis.M = Mn; epis.N = C; epis.H = H; epis.W = W_; epis.Q = Q;
-      epis.u = stride; epis.v = stride;
-      epis.dQ.init(Q); epis.dPQ.init(P * Q);
-      launch_gemm(pad_, pbd, epis, Mn, C, Ko);
+  DTMX_DISPATCH_16(dy.scalar_type(), "conv_dgrad", {
+    CHECK_BF16_CUDA(dy);
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy must be NHWC");
+    uint32_t N = dy.size(0), Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
+    uint32_t C = w.size(1), R = w.size(2), S = w.size(3);
+    // pad the out-channel (k-piece) dim to a multiple of 8 when needed
+    at::Tensor dyk = dy.permute({0, 2, 3, 1});        // NHWC view (contiguous)
+    at::Tensor wtk = w.permute({1, 2, 3, 0});          // (C,R,S,K)
+    if (Ko % 8) {
+      long padk = 8 - (Ko % 8);
+      dyk = at::constant_pad_nd(dyk, {0, padk}, 0.0);
+      wtk = at::constant_pad_nd(wtk, {0, padk}, 0.0);
+      Ko += padk;
     }
+    auto dyc = dyk.contiguous();
+    auto wt = wtk.contiguous();  // W^T in (C,R,S,Ko) dense layout
+    auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
+                        at::MemoryFormat::ChannelsLast);
+    uint32_t M = N * H * W_, Ktot = R * S * Ko;
+    if (R == 1 && S == 1 && pad == 0 && C % 8 == 0) {
+      // 1x1 dgrad: dense dy @ W^T. Stride 1 writes rows directly; stride u>1
+      // scatters row (n,p,q) to pixel (n, p*u, q*v) of a zeroed dx — the
+      // gathered formulation wastes 1-1/u^2 of its blocks on all-zero rows
+      // (measured 107 TF vs ~300 dense).
+      uint32_t Mn = N * P * Q;
+      DenseP<elem_t> pad_{(const elem_t*)dyc.data_ptr(), zero_page<elem_t>(dy), Mn, Ko, Ko};
+      DenseP<elem_t> pbd{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy), C, Ko, Ko};
+      if (stride == 1) {
+        EpiBF16<elem_t> epid{(elem_t*)dx.data_ptr(), nullptr, Mn, C, 0};
+        if (!smallgrid_splitk(pad_, pbd, dx, Mn, C, Ko))
+          launch_gemm(pad_, pbd, epid, Mn, C, Ko);
+      } else {
+        dx.zero_();
+        EpiBF16Scatter<elem_t> epis;
+        epis.dx = (elem_t*)dx.data_ptr();
+        epis.M = Mn; epis.N = C; epis.H = H; epis.W = W_; epis.Q = Q;
+        epis.u = stride; epis.v = stride;
+        epis.dQ.init(Q); epis.dPQ.init(P * Q);
+        launch_gemm(pad_, pbd, epis, Mn, C, Ko);
+      }
+      return dx;
+    }
+    ConvDgradA<elem_t> pa;
+    pa.dy = (const elem_t*)dyc.data_ptr();
+    pa.zero = zero_page<elem_t>(dy);
+    pa.M = M; pa.Ktot = Ktot; pa.Ko = Ko; pa.H = H; pa.W = W_; pa.P = P; pa.Q = Q;
+    pa.S = S; pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
+    pa.dW_.init(W_); pa.dHW.init(H * W_); pa.dKo.init(Ko); pa.dS.init(S);
+    DenseP<elem_t> pb{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy), C, Ktot, Ktot};
+    EpiBF16<elem_t> epi{(elem_t*)dx.data_ptr(), nullptr, M, C, 0};
+    if (!smallgrid_splitk(pa, pb, dx, M, C, Ktot))
+      launch_gemm(pa, pb, epi, M, C, Ktot);
     return dx;
-  }
-  ConvDgradA pa;
-  pa.dy = (const __bf16*)dyc.data_ptr();
-  pa.zero = zero_page(dy);
-  pa.M = M; pa.Ktot = Ktot; pa.Ko = Ko; pa.H = H; pa.W = W_; pa.P = P; pa.Q = Q;
-  pa.S = S; pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
-  pa.dW_.init(W_); pa.dHW.init(H * W_); pa.dKo.init(Ko); pa.dS.init(S);
-  DenseP pb{(const __bf16*)wt.data_ptr(), zero_page(dy), C, Ktot, Ktot};
-  EpiBF16 epi{(__bf16*)dx.data_ptr(), nullptr, M, C, 0};
-  if (!smallgrid_splitk(pa, pb, dx, M, C, Ktot))
-    launch_gemm(pa, pb, epi, M, C, Ktot);
-  return dx;
+
+  });
+  return at::Tensor();
 }
 
 // ------------------------------------------------------------- conv wgrad
@@ -936,92 +992,96 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
 
 at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
                       long stride, long pad) {
-  CHECK_BF16_CUDA(x);
-  CHECK_BF16_CUDA(dy);
-  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
-  uint32_t Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
-  uint32_t M = N * P * Q;               // the contraction length
-  uint32_t Mpad = ((M + 7) / 8) * 8;
-  uint32_t RSC = R * S * C;
+  DTMX_DISPATCH_16(x.scalar_type(), "conv_wgrad", {
+    CHECK_BF16_CUDA(x);
+    CHECK_BF16_CUDA(dy);
+    uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    uint32_t Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
+    uint32_t M = N * P * Q;               // the contraction length
+    uint32_t Mpad = ((M + 7) / 8) * 8;
+    uint32_t RSC = R * S * C;
 
-  if (C % 8 == 0 && Ko % 8 == 0) {
-    // transpose-free: dy and the im2col view of x are read NHWC-native by
-    // the contraction-major (NT) kernel
-    uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, RSC <= 64 ? 64 : 128);
-    uint32_t ktiles = ceil_div(M, 64);
-    uint32_t splitk = std::max<uint32_t>(
-        1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
+    if (C % 8 == 0 && Ko % 8 == 0) {
+      // transpose-free: dy and the im2col view of x are read NHWC-native by
+      // the contraction-major (NT) kernel
+      uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, RSC <= 64 ? 64 : 128);
+      uint32_t ktiles = ceil_div(M, 64);
+      uint32_t splitk = std::max<uint32_t>(
+          1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
+      auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
+      WgradDyA<elem_t> pa;
+      pa.dy = (const elem_t*)dy.data_ptr();
+      pa.zero = zero_page<elem_t>(x);
+      pa.Kd = M;
+      pa.Mdim = Ko;
+      WgradXcolB<elem_t> pb;
+      pb.x = (const elem_t*)x.data_ptr();
+      pb.zero = zero_page<elem_t>(x);
+      pb.Kd = M; pb.Ndim = RSC; pb.C = C; pb.H = H; pb.W = W_; pb.Q = Q; pb.S = S;
+      pb.u = stride; pb.v = stride; pb.ph = pad; pb.pw = pad;
+      pb.dQ.init(Q); pb.dPQ.init(P * Q); pb.dC.init(C); pb.dS.init(S);
+      EpiAtomicF32<elem_t> epi{dw32.data_ptr<float>(), Ko, RSC};
+      launch_gemm_nt(pa, pb, epi, Ko, RSC, M, splitk);
+      auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
+                    .to(at::kBFloat16)
+                    .permute({0, 3, 1, 2});
+      return dw.contiguous(at::MemoryFormat::ChannelsLast);
+    }
+
+    // dy^T : [NPQ][Ko] -> [Ko(+pad)][Mpad]
+    auto dyt = transpose2d<elem_t>(
+        pad_cols8(dy.permute({0, 2, 3, 1}).reshape({(long)M, (long)Ko})), Mpad);
+
+    // gathered im2col^T: [RSC][Mpad] (one transpose-gather launch per (r,s))
+    at::Tensor xt;
+    if (C % 8 == 0) {
+      xt = at::empty({(long)RSC, (long)Mpad}, x.options());
+      uint32_t tiles_m = ceil_div(Mpad, 64), tiles_c = ceil_div(C, 64);
+      for (uint32_t r = 0; r < (uint32_t)R; ++r)
+        for (uint32_t s = 0; s < (uint32_t)S; ++s) {
+          Im2colRows<elem_t> rows;
+          rows.x = (const elem_t*)x.data_ptr();
+          rows.M = M; rows.C = C; rows.H = H; rows.W = W_; rows.Q = Q;
+          rows.u = stride; rows.v = stride; rows.ph = pad; rows.pw = pad;
+          rows.r = r; rows.s = s;
+          rows.dQ.init(Q); rows.dPQ.init(P * Q);
+          transpose_rowgather_kernel<Im2colRows<elem_t>>
+              <<<tiles_m * tiles_c, 256, 0, cur_stream()>>>(
+                  rows, (elem_t*)xt.data_ptr(), C, Mpad, (r * S + s) * C, tiles_m);
+        }
+    } else {
+      // stem: materialize im2col then transpose
+      uint32_t Kpad = ((RSC + 7) / 8) * 8;
+      auto col = at::empty({(long)M, (long)Kpad}, x.options());
+      FastDiv dQ, dPQ, dC, dS;
+      dQ.init(Q); dPQ.init(P * Q); dC.init(C); dS.init(S);
+      size_t total = (size_t)M * Kpad;
+      uint32_t blocks = std::min<size_t>((total + 255) / 256, 16384);
+      im2col_kernel<<<blocks, 256, 0, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)col.data_ptr(), M, Kpad, RSC, C,
+          H, W_, Q, S, stride, stride, pad, pad, dQ, dPQ, dC, dS);
+      xt = transpose2d<elem_t>(col, Mpad).narrow(0, 0, RSC).contiguous();
+    }
+
+    // choose split-K so the grid fills the chip (~1024 blocks; 2 blocks/CU on
+    // 256 CUs plus headroom for tail effects)
+    uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, 128);
+    uint32_t ktiles = ceil_div(Mpad, 64);
+    uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
+
     auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
-    WgradDyA pa;
-    pa.dy = (const __bf16*)dy.data_ptr();
-    pa.zero = zero_page(x);
-    pa.Kd = M;
-    pa.Mdim = Ko;
-    WgradXcolB pb;
-    pb.x = (const __bf16*)x.data_ptr();
-    pb.zero = zero_page(x);
-    pb.Kd = M; pb.Ndim = RSC; pb.C = C; pb.H = H; pb.W = W_; pb.Q = Q; pb.S = S;
-    pb.u = stride; pb.v = stride; pb.ph = pad; pb.pw = pad;
-    pb.dQ.init(Q); pb.dPQ.init(P * Q); pb.dC.init(C); pb.dS.init(S);
-    EpiAtomicF32 epi{dw32.data_ptr<float>(), Ko, RSC};
-    launch_gemm_nt(pa, pb, epi, Ko, RSC, M, splitk);
+    DenseP<elem_t> pa{(const elem_t*)dyt.data_ptr(), zero_page<elem_t>(x), Ko, Mpad, Mpad};
+    DenseP<elem_t> pb{(const elem_t*)xt.data_ptr(), zero_page<elem_t>(x), RSC, Mpad, Mpad};
+    EpiAtomicF32<elem_t> epi{dw32.data_ptr<float>(), Ko, RSC};
+    launch_gemm(pa, pb, epi, Ko, RSC, Mpad, splitk);
+    // (Ko, R, S, C) fp32 -> bf16, viewed back to logical (K,C,R,S) channels_last
     auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
                   .to(at::kBFloat16)
                   .permute({0, 3, 1, 2});
     return dw.contiguous(at::MemoryFormat::ChannelsLast);
-  }
 
-  // dy^T : [NPQ][Ko] -> [Ko(+pad)][Mpad]
-  auto dyt = transpose2d(
-      pad_cols8(dy.permute({0, 2, 3, 1}).reshape({(long)M, (long)Ko})), Mpad);
-
-  // gathered im2col^T: [RSC][Mpad] (one transpose-gather launch per (r,s))
-  at::Tensor xt;
-  if (C % 8 == 0) {
-    xt = at::empty({(long)RSC, (long)Mpad}, x.options());
-    uint32_t tiles_m = ceil_div(Mpad, 64), tiles_c = ceil_div(C, 64);
-    for (uint32_t r = 0; r < (uint32_t)R; ++r)
-      for (uint32_t s = 0; s < (uint32_t)S; ++s) {
-        Im2colRows rows;
-        rows.x = (const __bf16*)x.data_ptr();
-        rows.M = M; rows.C = C; rows.H = H; rows.W = W_; rows.Q = Q;
-        rows.u = stride; rows.v = stride; rows.ph = pad; rows.pw = pad;
-        rows.r = r; rows.s = s;
-        rows.dQ.init(Q); rows.dPQ.init(P * Q);
-        transpose_rowgather_kernel<Im2colRows>
-            <<<tiles_m * tiles_c, 256, 0, cur_stream()>>>(
-                rows, (__bf16*)xt.data_ptr(), C, Mpad, (r * S + s) * C, tiles_m);
-      }
-  } else {
-    // stem: materialize im2col then transpose
-    uint32_t Kpad = ((RSC + 7) / 8) * 8;
-    auto col = at::empty({(long)M, (long)Kpad}, x.options());
-    FastDiv dQ, dPQ, dC, dS;
-    dQ.init(Q); dPQ.init(P * Q); dC.init(C); dS.init(S);
-    size_t total = (size_t)M * Kpad;
-    uint32_t blocks = std::min<size_t>((total + 255) / 256, 16384);
-    im2col_kernel<<<blocks, 256, 0, cur_stream()>>>(
-        (const __bf16*)x.data_ptr(), (__bf16*)col.data_ptr(), M, Kpad, RSC, C,
-        H, W_, Q, S, stride, stride, pad, pad, dQ, dPQ, dC, dS);
-    xt = transpose2d(col, Mpad).narrow(0, 0, RSC).contiguous();
-  }
-
-  // choose split-K so the grid fills the chip (~1024 blocks; 2 blocks/CU on
-  // 256 CUs plus headroom for tail effects)
-  uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, 128);
-  uint32_t ktiles = ceil_div(Mpad, 64);
-  uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
-
-  auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
-  DenseP pa{(const __bf16*)dyt.data_ptr(), zero_page(x), Ko, Mpad, Mpad};
-  DenseP pb{(const __bf16*)xt.data_ptr(), zero_page(x), RSC, Mpad, Mpad};
-  EpiAtomicF32 epi{dw32.data_ptr<float>(), Ko, RSC};
-  launch_gemm(pa, pb, epi, Ko, RSC, Mpad, splitk);
-  // (Ko, R, S, C) fp32 -> bf16, viewed back to logical (K,C,R,S) channels_last
-  auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
-                .to(at::kBFloat16)
-                .permute({0, 3, 1, 2});
-  return dw.contiguous(at::MemoryFormat::ChannelsLast);
+  });
+  return at::Tensor();
 }
 
 }  // namespace dtmx

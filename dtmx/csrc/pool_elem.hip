@@ -11,13 +11,15 @@ static hipStream_t pe_stream() { return at::hip::getCurrentHIPStream().stream();
 
 // ------------------------------------------------------------- max pooling
 
-// one thread = 8 channels (bf16x8 loads/stores; scalar 2-B accesses measured
+// one thread = 8 channels (V8 loads/stores; scalar 2-B accesses measured
 // ~10x off the bandwidth roofline on the stem maxpool)
-__global__ void maxpool_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+template <typename elem_t>
+__global__ void maxpool_fwd_kernel(const elem_t* __restrict__ x, elem_t* __restrict__ y,
                                    uint8_t* __restrict__ idx, uint32_t N, uint32_t C,
                                    uint32_t H, uint32_t W, uint32_t P, uint32_t Q,
                                    uint32_t K, int u, int pad, FastDiv dCv,
                                    FastDiv dQ_, FastDiv dPQ) {
+  using V8 = typename E8<elem_t>::v8;
   const uint32_t cvecs = C / 8;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t total = (uint32_t)(N * P * Q) * cvecs;
@@ -37,7 +39,7 @@ __global__ void maxpool_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restr
       for (uint32_t kw = 0; kw < K; ++kw) {
         int iw = (int)(q * u) - pad + (int)kw;
         if ((uint32_t)iw >= W) continue;
-        bf16x8 v = *(const bf16x8*)(x + (((size_t)n * H + ih) * W + iw) * C + cv * 8);
+        V8 v = *(const V8*)(x + (((size_t)n * H + ih) * W + iw) * C + cv * 8);
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
           float f = (float)v[e];
@@ -45,21 +47,23 @@ __global__ void maxpool_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restr
         }
       }
     }
-    bf16x8 o;
+    V8 o;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) o[e] = (__bf16)best[e];
+    for (int e = 0; e < 8; ++e) o[e] = (elem_t)best[e];
     size_t off = (size_t)m * C + cv * 8;
-    *(bf16x8*)(y + off) = o;
+    *(V8*)(y + off) = o;
     *(u8x8*)(idx + off) = besti;
   }
 }
 
-__global__ void maxpool_bwd_kernel(const __bf16* __restrict__ dy,
+template <typename elem_t>
+__global__ void maxpool_bwd_kernel(const elem_t* __restrict__ dy,
                                    const uint8_t* __restrict__ idx,
-                                   __bf16* __restrict__ dx, uint32_t N, uint32_t C,
+                                   elem_t* __restrict__ dx, uint32_t N, uint32_t C,
                                    uint32_t H, uint32_t W, uint32_t P, uint32_t Q,
                                    uint32_t K, int u, int pad, FastDiv dCv,
                                    FastDiv dW2, FastDiv dHW) {
+  using V8 = typename E8<elem_t>::v8;
   const uint32_t cvecs = C / 8;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t total = (uint32_t)(N * H * W) * cvecs;
@@ -88,27 +92,29 @@ __global__ void maxpool_bwd_kernel(const __bf16* __restrict__ dy,
         if (kw >= K) continue;
         size_t o = (((size_t)n * P + p) * Q + q) * C + cv * 8;
         u8x8 iv = *(const u8x8*)(idx + o);
-        bf16x8 gv = *(const bf16x8*)(dy + o);
+        V8 gv = *(const V8*)(dy + o);
         uint8_t want = (uint8_t)(kh * K + kw);
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           if (iv[e] == want) acc8[e] += (float)gv[e];
       }
     }
-    bf16x8 o8;
+    V8 o8;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) o8[e] = (__bf16)acc8[e];
-    *(bf16x8*)(dx + (size_t)m * C + cv * 8) = o8;
+    for (int e = 0; e < 8; ++e) o8[e] = (elem_t)acc8[e];
+    *(V8*)(dx + (size_t)m * C + cv * 8) = o8;
   }
 }
 
 // scalar fallbacks for C % 8 != 0 (zoo models like lenet's 20 channels)
-__global__ void maxpool_fwd_scalar_kernel(const __bf16* __restrict__ x,
-                                          __bf16* __restrict__ y,
+template <typename elem_t>
+__global__ void maxpool_fwd_scalar_kernel(const elem_t* __restrict__ x,
+                                          elem_t* __restrict__ y,
                                           uint8_t* __restrict__ idx, uint32_t N,
                                           uint32_t C, uint32_t H, uint32_t W,
                                           uint32_t P, uint32_t Q, uint32_t K,
                                           int u, int pad) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t total = (size_t)N * P * Q * C;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -130,17 +136,19 @@ __global__ void maxpool_fwd_scalar_kernel(const __bf16* __restrict__ x,
         if (v > best) { best = v; besti = kh * K + kw; }
       }
     }
-    y[i] = (__bf16)best;
+    y[i] = (elem_t)best;
     idx[i] = besti;
   }
 }
 
-__global__ void maxpool_bwd_scalar_kernel(const __bf16* __restrict__ dy,
+template <typename elem_t>
+__global__ void maxpool_bwd_scalar_kernel(const elem_t* __restrict__ dy,
                                           const uint8_t* __restrict__ idx,
-                                          __bf16* __restrict__ dx, uint32_t N,
+                                          elem_t* __restrict__ dx, uint32_t N,
                                           uint32_t C, uint32_t H, uint32_t W,
                                           uint32_t P, uint32_t Q, uint32_t K,
                                           int u, int pad) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t total = (size_t)N * H * W * C;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -169,25 +177,29 @@ __global__ void maxpool_bwd_scalar_kernel(const __bf16* __restrict__ dy,
         if (idx[o] == kh * K + kw) acc += (float)dy[o];
       }
     }
-    dx[i] = (__bf16)acc;
+    dx[i] = (elem_t)acc;
   }
 }
 
 // ------------------------------------------------------- global average pool
 
-__global__ void gap_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+template <typename elem_t>
+__global__ void gap_fwd_kernel(const elem_t* __restrict__ x, elem_t* __restrict__ y,
                                uint32_t N, uint32_t C, uint32_t HW) {
+  using V8 = typename E8<elem_t>::v8;
   uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
   uint32_t n = blockIdx.y;
   if (c >= C) return;
-  const __bf16* base = x + (size_t)n * HW * C + c;
+  const elem_t* base = x + (size_t)n * HW * C + c;
   float s = 0.f;
   for (uint32_t i = 0; i < HW; ++i) s += (float)base[(size_t)i * C];
-  y[(size_t)n * C + c] = (__bf16)(s / HW);
+  y[(size_t)n * C + c] = (elem_t)(s / HW);
 }
 
-__global__ void gap_bwd_kernel(const __bf16* __restrict__ dy, __bf16* __restrict__ dx,
+template <typename elem_t>
+__global__ void gap_bwd_kernel(const elem_t* __restrict__ dy, elem_t* __restrict__ dx,
                                uint32_t N, uint32_t C, uint32_t HW) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t total = (size_t)N * HW * C;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -195,54 +207,60 @@ __global__ void gap_bwd_kernel(const __bf16* __restrict__ dy, __bf16* __restrict
   for (; i < total; i += stride) {
     uint32_t c = i % C;
     uint32_t n = i / ((size_t)HW * C);
-    dx[i] = (__bf16)((float)dy[(size_t)n * C + c] * inv);
+    dx[i] = (elem_t)((float)dy[(size_t)n * C + c] * inv);
   }
 }
 
 // -------------------------------------------------------------- elementwise
 
-__global__ void relu_fwd_kernel(const __bf16* __restrict__ x, __bf16* __restrict__ y,
+template <typename elem_t>
+__global__ void relu_fwd_kernel(const elem_t* __restrict__ x, elem_t* __restrict__ y,
                                 size_t total8) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
-    bf16x8 v = *(const bf16x8*)(x + i * 8);
-    bf16x8 o;
+    V8 v = *(const V8*)(x + i * 8);
+    V8 o;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) o[e] = (__bf16)fmaxf((float)v[e], 0.f);
-    *(bf16x8*)(y + i * 8) = o;
+    for (int e = 0; e < 8; ++e) o[e] = (elem_t)fmaxf((float)v[e], 0.f);
+    *(V8*)(y + i * 8) = o;
   }
 }
 
-__global__ void relu_bwd_kernel(const __bf16* __restrict__ dy,
-                                const __bf16* __restrict__ y,
-                                __bf16* __restrict__ dx, size_t total8) {
+template <typename elem_t>
+__global__ void relu_bwd_kernel(const elem_t* __restrict__ dy,
+                                const elem_t* __restrict__ y,
+                                elem_t* __restrict__ dx, size_t total8) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
-    bf16x8 g = *(const bf16x8*)(dy + i * 8);
-    bf16x8 v = *(const bf16x8*)(y + i * 8);
-    bf16x8 o;
+    V8 g = *(const V8*)(dy + i * 8);
+    V8 v = *(const V8*)(y + i * 8);
+    V8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      o[e] = (float)v[e] > 0.f ? g[e] : (__bf16)0.f;
-    *(bf16x8*)(dx + i * 8) = o;
+      o[e] = (float)v[e] > 0.f ? g[e] : (elem_t)0.f;
+    *(V8*)(dx + i * 8) = o;
   }
 }
 
-__global__ void add_relu_kernel(const __bf16* __restrict__ a,
-                                const __bf16* __restrict__ b,
-                                __bf16* __restrict__ y, size_t total8) {
+template <typename elem_t>
+__global__ void add_relu_kernel(const elem_t* __restrict__ a,
+                                const elem_t* __restrict__ b,
+                                elem_t* __restrict__ y, size_t total8) {
+  using V8 = typename E8<elem_t>::v8;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < total8; i += stride) {
-    bf16x8 va = *(const bf16x8*)(a + i * 8);
-    bf16x8 vb = *(const bf16x8*)(b + i * 8);
-    bf16x8 o;
+    V8 va = *(const V8*)(a + i * 8);
+    V8 vb = *(const V8*)(b + i * 8);
+    V8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      o[e] = (__bf16)fmaxf((float)va[e] + (float)vb[e], 0.f);
-    *(bf16x8*)(y + i * 8) = o;
+      o[e] = (elem_t)fmaxf((float)va[e] + (float)vb[e], 0.f);
+    *(V8*)(y + i * 8) = o;
   }
 }
 
@@ -262,20 +280,22 @@ std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, long kernel, long strid
                      at::MemoryFormat::ChannelsLast);
   auto idx = at::empty({(long)N, (long)C, (long)P, (long)Q},
                        x.options().dtype(at::kByte), at::MemoryFormat::ChannelsLast);
-  if (C % 8 == 0) {
-    FastDiv dCv, dQ_, dPQ;
-    dCv.init(C / 8); dQ_.init(Q); dPQ.init(P * Q);
-    size_t total = (size_t)N * P * Q * (C / 8);
-    maxpool_fwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-        (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
-        (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
-        dCv, dQ_, dPQ);
-  } else {
-    size_t total = (size_t)N * P * Q * C;
-    maxpool_fwd_scalar_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-        (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(),
-        (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
-  }
+  DTMX_DISPATCH_16(x.scalar_type(), "maxpool", {
+    if (C % 8 == 0) {
+      FastDiv dCv, dQ_, dPQ;
+      dCv.init(C / 8); dQ_.init(Q); dPQ.init(P * Q);
+      size_t total = (size_t)N * P * Q * (C / 8);
+      maxpool_fwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
+          (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
+          dCv, dQ_, dPQ);
+    } else {
+      size_t total = (size_t)N * P * Q * C;
+      maxpool_fwd_scalar_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
+          (uint8_t*)idx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+    }
+  });
   return {y, idx};
 }
 
@@ -285,20 +305,22 @@ at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx, long H,
   auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
   auto dx = at::empty({(long)N, (long)C, (long)H, (long)W}, dy.options(),
                       at::MemoryFormat::ChannelsLast);
-  if (C % 8 == 0) {
-    FastDiv dCv, dW2, dHW;
-    dCv.init(C / 8); dW2.init(W); dHW.init(H * W);
-    size_t total = (size_t)N * H * W * (C / 8);
-    maxpool_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-        (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
-        (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
-        dCv, dW2, dHW);
-  } else {
-    size_t total = (size_t)N * H * W * C;
-    maxpool_bwd_scalar_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-        (const __bf16*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
-        (__bf16*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
-  }
+  DTMX_DISPATCH_16(dy.scalar_type(), "maxpool_bwd", {
+    if (C % 8 == 0) {
+      FastDiv dCv, dW2, dHW;
+      dCv.init(C / 8); dW2.init(W); dHW.init(H * W);
+      size_t total = (size_t)N * H * W * (C / 8);
+      maxpool_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+          (const elem_t*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
+          (elem_t*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad,
+          dCv, dW2, dHW);
+    } else {
+      size_t total = (size_t)N * H * W * C;
+      maxpool_bwd_scalar_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+          (const elem_t*)dyc.data_ptr(), (const uint8_t*)idx.data_ptr(),
+          (elem_t*)dx.data_ptr(), N, C, H, W, P, Q, kernel, stride, pad);
+    }
+  });
   return dx;
 }
 
@@ -307,8 +329,10 @@ at::Tensor global_avgpool_fwd(const at::Tensor& x) {
   uint32_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto y = at::empty({(long)N, (long)C}, x.options());
   dim3 grid((C + 255) / 256, N);
-  gap_fwd_kernel<<<grid, 256, 0, pe_stream()>>>(
-      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(), N, C, HW);
+  DTMX_DISPATCH_16(x.scalar_type(), "gap", {
+    gap_fwd_kernel<<<grid, 256, 0, pe_stream()>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), N, C, HW);
+  });
   return y;
 }
 
@@ -317,21 +341,27 @@ at::Tensor global_avgpool_bwd(const at::Tensor& dy, long H, long W) {
   auto dx = at::empty({(long)N, (long)C, H, W}, dy.options(),
                       at::MemoryFormat::ChannelsLast);
   size_t total = (size_t)N * HW * C;
-  gap_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
-      (const __bf16*)dy.data_ptr(), (__bf16*)dx.data_ptr(), N, C, HW);
+  DTMX_DISPATCH_16(dy.scalar_type(), "gap_bwd", {
+    gap_bwd_kernel<<<ew_blocks(total), 256, 0, pe_stream()>>>(
+        (const elem_t*)dy.data_ptr(), (elem_t*)dx.data_ptr(), N, C, HW);
+  });
   return dx;
 }
 
-#define EW_CHECK(t)                                             \
-  TORCH_CHECK((t).scalar_type() == at::kBFloat16, "bf16 only"); \
+#define EW_CHECK(t)                                                        \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16 ||                        \
+                  (t).scalar_type() == at::kHalf,                          \
+              "16-bit float only");                                        \
   TORCH_CHECK((t).numel() % 8 == 0, "numel must be a multiple of 8")
 
 at::Tensor relu_fwd(const at::Tensor& x) {
   EW_CHECK(x);
   auto y = at::empty_like(x);
   size_t t8 = x.numel() / 8;
-  relu_fwd_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
-      (const __bf16*)x.data_ptr(), (__bf16*)y.data_ptr(), t8);
+  DTMX_DISPATCH_16(x.scalar_type(), "relu", {
+    relu_fwd_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), t8);
+  });
   return y;
 }
 
@@ -339,9 +369,11 @@ at::Tensor relu_bwd(const at::Tensor& dy, const at::Tensor& y) {
   EW_CHECK(dy);
   auto dx = at::empty_like(dy);
   size_t t8 = dy.numel() / 8;
-  relu_bwd_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
-      (const __bf16*)dy.data_ptr(), (const __bf16*)y.data_ptr(),
-      (__bf16*)dx.data_ptr(), t8);
+  DTMX_DISPATCH_16(dy.scalar_type(), "relu_bwd", {
+    relu_bwd_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
+        (const elem_t*)dy.data_ptr(), (const elem_t*)y.data_ptr(),
+        (elem_t*)dx.data_ptr(), t8);
+  });
   return dx;
 }
 
@@ -349,9 +381,11 @@ at::Tensor add_relu_fwd(const at::Tensor& a, const at::Tensor& b) {
   EW_CHECK(a);
   auto y = at::empty_like(a);
   size_t t8 = a.numel() / 8;
-  add_relu_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
-      (const __bf16*)a.data_ptr(), (const __bf16*)b.data_ptr(),
-      (__bf16*)y.data_ptr(), t8);
+  DTMX_DISPATCH_16(a.scalar_type(), "add_relu", {
+    add_relu_kernel<<<ew_blocks(t8), 256, 0, pe_stream()>>>(
+        (const elem_t*)a.data_ptr(), (const elem_t*)b.data_ptr(),
+        (elem_t*)y.data_ptr(), t8);
+  });
   return y;
 }
 

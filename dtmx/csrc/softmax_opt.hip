@@ -12,14 +12,15 @@ namespace dtmx {
 static hipStream_t so_stream() { return at::hip::getCurrentHIPStream().stream(); }
 
 // one block (256 threads) per row: max -> exp-sum -> probs + per-row loss
-__global__ void softmax_ce_fwd_kernel(const __bf16* __restrict__ logits,
+template <typename elem_t>
+__global__ void softmax_ce_fwd_kernel(const elem_t* __restrict__ logits,
                                       const int* __restrict__ label,
-                                      __bf16* __restrict__ probs,
+                                      elem_t* __restrict__ probs,
                                       float* __restrict__ loss, uint32_t V) {
   const uint32_t row = blockIdx.x;
   const uint32_t t = threadIdx.x;
-  const __bf16* in = logits + (size_t)row * V;
-  __bf16* out = probs + (size_t)row * V;
+  const elem_t* in = logits + (size_t)row * V;
+  elem_t* out = probs + (size_t)row * V;
   __shared__ float red[8];
 
   float m = -3.4e38f;
@@ -40,7 +41,7 @@ __global__ void softmax_ce_fwd_kernel(const __bf16* __restrict__ logits,
   const float inv = 1.f / s, logs = __logf(s);
 
   for (uint32_t i = t; i < V; i += 256)
-    out[i] = (__bf16)(__expf((float)in[i] - m) * inv);
+    out[i] = (elem_t)(__expf((float)in[i] - m) * inv);
   if (t == 0) {
     int y = label[row];
     float lp = ((float)in[y] - m) - logs;  // log softmax at the label
@@ -49,10 +50,11 @@ __global__ void softmax_ce_fwd_kernel(const __bf16* __restrict__ logits,
 }
 
 // d_logits = (p - onehot) * dloss   (sum-CE; reference SoftmaxOutput grad)
-__global__ void softmax_ce_bwd_kernel(const __bf16* __restrict__ probs,
+template <typename elem_t>
+__global__ void softmax_ce_bwd_kernel(const elem_t* __restrict__ probs,
                                       const int* __restrict__ label,
                                       const float* __restrict__ dloss,
-                                      __bf16* __restrict__ dlogits, uint32_t V,
+                                      elem_t* __restrict__ dlogits, uint32_t V,
                                       size_t total) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -61,7 +63,7 @@ __global__ void softmax_ce_bwd_kernel(const __bf16* __restrict__ probs,
     uint32_t row = i / V, col = i % V;
     float p = (float)probs[i];
     if ((int)col == label[row]) p -= 1.f;
-    dlogits[i] = (__bf16)(p * g);
+    dlogits[i] = (elem_t)(p * g);
   }
 }
 
@@ -69,7 +71,8 @@ __global__ void softmax_ce_bwd_kernel(const __bf16* __restrict__ probs,
 //   g32  = clip(grad * rescale) + wd * master
 //   mom  = momentum * mom - lr * g32
 //   master += mom;  w = bf16(master)
-__global__ void sgd_mom_mp_kernel(__bf16* __restrict__ w, const __bf16* __restrict__ g,
+template <typename elem_t>
+__global__ void sgd_mom_mp_kernel(elem_t* __restrict__ w, const elem_t* __restrict__ g,
                                   float* __restrict__ master, float* __restrict__ mom,
                                   size_t total, float lr, float momentum, float wd,
                                   float rescale, float clip) {
@@ -83,15 +86,16 @@ __global__ void sgd_mom_mp_kernel(__bf16* __restrict__ w, const __bf16* __restri
     mom[i] = m;
     float nw = master[i] + m;
     master[i] = nw;
-    w[i] = (__bf16)nw;
+    w[i] = (elem_t)nw;
   }
 }
 
 // device-hyperparameter variant (hipGraph-capturable: lr/rescale live in a
 // device buffer updated between replays, not baked into kernel args).
 // hyper = {lr, momentum, wd, rescale, clip}
-__global__ void sgd_mom_mp_dev_kernel(__bf16* __restrict__ w,
-                                      const __bf16* __restrict__ g,
+template <typename elem_t>
+__global__ void sgd_mom_mp_dev_kernel(elem_t* __restrict__ w,
+                                      const elem_t* __restrict__ g,
                                       float* __restrict__ master,
                                       float* __restrict__ mom, size_t total,
                                       const float* __restrict__ hyper) {
@@ -107,7 +111,7 @@ __global__ void sgd_mom_mp_dev_kernel(__bf16* __restrict__ w,
     mom[i] = m;
     float nw = master[i] + m;
     master[i] = nw;
-    w[i] = (__bf16)nw;
+    w[i] = (elem_t)nw;
   }
 }
 
@@ -137,9 +141,11 @@ std::vector<at::Tensor> softmax_ce_fwd(const at::Tensor& logits,
   uint32_t B = lc.size(0), V = lc.size(1);
   auto probs = at::empty_like(lc);
   auto loss = at::zeros({}, lc.options().dtype(at::kFloat));
-  softmax_ce_fwd_kernel<<<B, 256, 0, so_stream()>>>(
-      (const __bf16*)lc.data_ptr(), yc.data_ptr<int>(),
-      (__bf16*)probs.data_ptr(), loss.data_ptr<float>(), V);
+  DTMX_DISPATCH_16(lc.scalar_type(), "softmax_ce", {
+    softmax_ce_fwd_kernel<<<B, 256, 0, so_stream()>>>(
+        (const elem_t*)lc.data_ptr(), yc.data_ptr<int>(),
+        (elem_t*)probs.data_ptr(), loss.data_ptr<float>(), V);
+  });
   return {loss, probs};
 }
 
@@ -151,9 +157,11 @@ at::Tensor softmax_ce_bwd(const at::Tensor& probs, const at::Tensor& label,
   auto dlf = dloss.to(at::kFloat).contiguous();
   size_t total = (size_t)B * V;
   uint32_t blocks = std::min<size_t>((total + 255) / 256, 2048);
-  softmax_ce_bwd_kernel<<<blocks, 256, 0, so_stream()>>>(
-      (const __bf16*)probs.data_ptr(), yc.data_ptr<int>(),
-      dlf.data_ptr<float>(), (__bf16*)dl.data_ptr(), V, total);
+  DTMX_DISPATCH_16(probs.scalar_type(), "softmax_ce_bwd", {
+    softmax_ce_bwd_kernel<<<blocks, 256, 0, so_stream()>>>(
+        (const elem_t*)probs.data_ptr(), yc.data_ptr<int>(),
+        dlf.data_ptr<float>(), (elem_t*)dl.data_ptr(), V, total);
+  });
   return dl;
 }
 
@@ -162,20 +170,24 @@ void sgd_mom_mp(at::Tensor w, const at::Tensor& g, at::Tensor master,
                 double rescale, double clip) {
   size_t total = w.numel();
   uint32_t blocks = std::min<size_t>((total + 255) / 256, 4096);
-  sgd_mom_mp_kernel<<<blocks, 256, 0, so_stream()>>>(
-      (__bf16*)w.data_ptr(), (const __bf16*)g.data_ptr(),
-      master.data_ptr<float>(), mom.data_ptr<float>(), total, lr, momentum, wd,
-      rescale, clip);
+  DTMX_DISPATCH_16(w.scalar_type(), "sgd_mom_mp", {
+    sgd_mom_mp_kernel<<<blocks, 256, 0, so_stream()>>>(
+        (elem_t*)w.data_ptr(), (const elem_t*)g.data_ptr(),
+        master.data_ptr<float>(), mom.data_ptr<float>(), total, lr, momentum, wd,
+        rescale, clip);
+  });
 }
 
 void sgd_mom_mp_dev(at::Tensor w, const at::Tensor& g, at::Tensor master,
                     at::Tensor mom, const at::Tensor& hyper) {
   size_t total = w.numel();
   uint32_t blocks = std::min<size_t>((total + 255) / 256, 4096);
-  sgd_mom_mp_dev_kernel<<<blocks, 256, 0, so_stream()>>>(
-      (__bf16*)w.data_ptr(), (const __bf16*)g.data_ptr(),
-      master.data_ptr<float>(), mom.data_ptr<float>(), total,
-      hyper.data_ptr<float>());
+  DTMX_DISPATCH_16(w.scalar_type(), "sgd_mom_mp_dev", {
+    sgd_mom_mp_dev_kernel<<<blocks, 256, 0, so_stream()>>>(
+        (elem_t*)w.data_ptr(), (const elem_t*)g.data_ptr(),
+        master.data_ptr<float>(), mom.data_ptr<float>(), total,
+        hyper.data_ptr<float>());
+  });
 }
 
 void sgd_mom_f32(at::Tensor w, const at::Tensor& g, at::Tensor mom, double lr,

@@ -252,7 +252,7 @@ class Module:
         self._use_fused_sgd = (
             on_gpu
             and type(optimizer) is _SGD
-            and self._dtype == torch.bfloat16
+            and self._dtype in (torch.bfloat16, torch.float16)
             and not self.fixed_param_names
         )
         if self._use_fused_sgd:
