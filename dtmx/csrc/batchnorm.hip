@@ -488,7 +488,9 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
                                      const at::Tensor& beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
                                      double eps, bool fuse_relu,
-                                     const c10::optional<at::Tensor>& residual) {
+                                     const c10::optional<at::Tensor>& residual,
+                                     const c10::optional<at::Tensor>& pre_psum,
+                                     const c10::optional<at::Tensor>& pre_psumsq) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
   uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8, got ", C);
@@ -499,17 +501,28 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
   uint32_t rpb;
   bn_grid(rows, cvecs, cpb, grid, rpb);
   uint32_t nslabs = grid.y;
-  auto psum = at::empty({(long)nslabs, (long)C}, opt_f);
-  auto psumsq = at::empty({(long)nslabs, (long)C}, opt_f);
+  at::Tensor psum, psumsq;
+  const bool have_pre =
+      pre_psum.has_value() && pre_psum->defined() && pre_psum->numel() > 0;
+  if (have_pre) {
+    // fused path: the producing conv's epilogue already wrote the slabs
+    psum = *pre_psum;
+    psumsq = *pre_psumsq;
+    nslabs = psum.size(0);
+  } else {
+    psum = at::empty({(long)nslabs, (long)C}, opt_f);
+    psumsq = at::empty({(long)nslabs, (long)C}, opt_f);
+  }
   auto save_mean = at::empty({(long)C}, opt_f), save_invstd = at::empty({(long)C}, opt_f);
   auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
   auto y = at::empty_like(x);
   auto s = bn_stream();
   DTMX_DISPATCH_16(x.scalar_type(), "bn_fwd", {
-    bn_stats_kernel<<<grid, 256, 0, s>>>((const elem_t*)x.data_ptr(),
-                                         psum.data_ptr<float>(),
-                                         psumsq.data_ptr<float>(), rows, cvecs,
-                                         cpb, rpb);
+    if (!have_pre)
+      bn_stats_kernel<<<grid, 256, 0, s>>>((const elem_t*)x.data_ptr(),
+                                           psum.data_ptr<float>(),
+                                           psumsq.data_ptr<float>(), rows, cvecs,
+                                           cpb, rpb);
     uint32_t ncv = std::min(cpb, 8u);
     bn_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
         psum.data_ptr<float>(), psumsq.data_ptr<float>(),

@@ -13,7 +13,11 @@ at::Tensor conv_wgrad(const at::Tensor&, const at::Tensor&, long, long, long, lo
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor&, const at::Tensor&,
                                      const at::Tensor&, at::Tensor, at::Tensor,
                                      double, double, bool,
+                                     const c10::optional<at::Tensor>&,
+                                     const c10::optional<at::Tensor>&,
                                      const c10::optional<at::Tensor>&);
+std::vector<at::Tensor> conv_fwd_stats(const at::Tensor&, const at::Tensor&,
+                                       long, long);
 at::Tensor bn_fwd_infer(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                         const at::Tensor&, const at::Tensor&, double, bool,
                         const c10::optional<at::Tensor>&);
@@ -50,6 +54,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_dgrad", &dtmx::linear_dgrad);
   m.def("linear_wgrad", &dtmx::linear_wgrad);
   m.def("conv_fwd", &dtmx::conv_fwd);
+  m.def("conv_fwd_stats", &dtmx::conv_fwd_stats);
   m.def("conv_dgrad", &dtmx::conv_dgrad);
   m.def("conv_wgrad", &dtmx::conv_wgrad);
   m.def("bn_fwd_train", &dtmx::bn_fwd_train);

@@ -141,6 +141,12 @@ struct EpiBF16 {
   const float* bias;  // nullable
   uint32_t M, N;
   int relu;
+  // optional fused BN statistics: per-block column sums/sumsq of the output
+  // tile into [tiles_m][N] slabs (zeroed by the launcher) — the following
+  // BatchNorm then skips its own stats read of the whole tensor. Only set
+  // when bias == nullptr && !relu (stats must equal the written values).
+  float* bn_psum = nullptr;
+  float* bn_psumsq = nullptr;
   // coalesced row-chunk store used by the kernel's LDS-staged epilogue
   __device__ __forceinline__ void store_chunk(uint32_t m, uint32_t n0,
                                               V8 v) const {
@@ -175,6 +181,30 @@ struct EpiBF16 {
         }
       }
   }
+
+  // column sums/sumsq of the block's LDS-staged tile; distinct (slab_row, n)
+  // targets per block, 256/BN stripes share one via atomicAdd.
+  __device__ __forceinline__ void bn_stats(const elem_t* ct, uint32_t bm,
+                                           uint32_t bn, uint32_t BN,
+                                           uint32_t t) const {
+    if (!bn_psum) return;
+    const uint32_t stripes = 256 / BN;
+    const uint32_t col = t % BN;
+    const uint32_t n = bn + col;
+    if (n >= N) return;
+    const uint32_t rows_per = 128 / stripes;
+    const uint32_t r0 = (t / BN) * rows_per;
+    float sum = 0.f, sumsq = 0.f;
+    for (uint32_t r = 0; r < rows_per; ++r) {
+      if (bm + r0 + r >= M) break;
+      float v = (float)ct[(r0 + r) * BN + col];
+      sum += v;
+      sumsq += v * v;
+    }
+    const size_t slab = (size_t)(bm >> 7) * N + n;
+    atomicAdd(&bn_psum[slab], sum);
+    atomicAdd(&bn_psumsq[slab], sumsq);
+  }
 };
 
 // 1x1 stride-u dgrad: rows m = (n,p,q) of the dense dy@W^T GEMM scatter to
@@ -200,6 +230,8 @@ struct EpiBF16Scatter {
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t,
                                         uint32_t, uint32_t) const {}
+  __device__ __forceinline__ void bn_stats(const elem_t*, uint32_t, uint32_t,
+                                           uint32_t, uint32_t) const {}
 };
 
 template <typename elem_t>
@@ -357,6 +389,7 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
       uint32_t row = idx / (BN / 8), nc = idx % (BN / 8);
       epi.store_chunk(bm + row, bn + nc * 8, *(const V8*)(ct + row * BN + nc * 8));
     }
+    if constexpr (EPI::kLdsStage) epi.bn_stats(ct, bm, bn, BN, t);
   } else {
     epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
   }
@@ -922,6 +955,56 @@ at::Tensor conv_fwd(const at::Tensor& x, const at::Tensor& w, long stride,
 
   });
   return at::Tensor();
+}
+
+// conv forward with fused BN statistics: returns (y, psum, psumsq) where the
+// slabs are [tiles_m][Ko] per-block column sums/sumsq of y (epilogue fusion —
+// the following BatchNorm skips its whole-tensor stats read). Returns empty
+// slabs when the shape routed through a non-fusable path; the caller falls
+// back to the standalone stats kernel.
+std::vector<at::Tensor> conv_fwd_stats(const at::Tensor& x, const at::Tensor& w,
+                                       long stride, long pad) {
+  DTMX_DISPATCH_16(x.scalar_type(), "conv_fwd_stats", {
+    CHECK_BF16_CUDA(x);
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x must be NHWC");
+    uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    uint32_t Ko = w.size(0), R = w.size(2), S = w.size(3);
+    uint32_t P, Q;
+    conv_out_dims(H, W_, R, S, stride, pad, P, Q);
+    uint32_t M = N * P * Q, Ktot = R * S * C;
+    uint32_t tiles_m = ceil_div(M, 128);
+    uint32_t tiles = tiles_m * ceil_div(Ko, Ko <= 64 ? 64 : 128);
+    uint32_t ktiles = ceil_div(Ktot, 64);
+    bool smallgrid = tiles < 160 && ktiles >= 8;  // mirrors smallgrid_splitk
+    if (C % 8 != 0 || smallgrid) {
+      auto y = conv_fwd(x, w, stride, pad);
+      auto empty = at::empty({0}, x.options().dtype(at::kFloat));
+      return {y, empty, empty};
+    }
+    auto y = at::empty({(long)N, (long)Ko, (long)P, (long)Q}, x.options(),
+                       at::MemoryFormat::ChannelsLast);
+    auto opt_f = x.options().dtype(at::kFloat);
+    auto psum = at::zeros({(long)tiles_m, (long)Ko}, opt_f);
+    auto psumsq = at::zeros({(long)tiles_m, (long)Ko}, opt_f);
+    EpiBF16<elem_t> epi{(elem_t*)y.data_ptr(), nullptr, M, Ko, 0,
+                        psum.data_ptr<float>(), psumsq.data_ptr<float>()};
+    if (R == 1 && S == 1 && stride == 1 && pad == 0) {
+      DenseP<elem_t> pa{(const elem_t*)x.data_ptr(), zero_page<elem_t>(x), M, C, C};
+      DenseP<elem_t> pb{(const elem_t*)w.data_ptr(), zero_page<elem_t>(x), Ko, C, C};
+      launch_gemm(pa, pb, epi, M, Ko, C);
+    } else {
+      DenseP<elem_t> pb{(const elem_t*)w.data_ptr(), zero_page<elem_t>(x), Ko, Ktot, Ktot};
+      ConvFwdA<elem_t> pa;
+      pa.x = (const elem_t*)x.data_ptr();
+      pa.zero = zero_page<elem_t>(x);
+      pa.M = M; pa.Ktot = Ktot; pa.C = C; pa.H = H; pa.W = W_; pa.Q = Q; pa.S = S;
+      pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
+      pa.dQ.init(Q); pa.dPQ.init(P * Q); pa.dC.init(C); pa.dS.init(S);
+      launch_gemm(pa, pb, epi, M, Ko, Ktot);
+    }
+    return {y, psum, psumsq};
+  });
+  return {};
 }
 
 // ------------------------------------------------------------- conv dgrad

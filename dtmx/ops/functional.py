@@ -32,15 +32,22 @@ def _use_hip(*tensors) -> bool:
 
 class _ConvNHWC(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, stride, padding):
+    def forward(ctx, x, w, stride, padding, want_stats):
         ext = require_ext()
         ctx.save_for_backward(x, w)
         ctx.stride = stride
         ctx.padding = padding
-        return ext.conv_fwd(x, w, stride, padding)
+        if want_stats:
+            y, ps, pss = ext.conv_fwd_stats(x, w, stride, padding)
+            ctx.mark_non_differentiable(ps, pss)
+            return y, ps, pss
+        y = ext.conv_fwd(x, w, stride, padding)
+        z = y.new_zeros(0)
+        ctx.mark_non_differentiable(z)
+        return y, z, z
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, _dps=None, _dpss=None):
         ext = require_ext()
         x, w = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
@@ -49,10 +56,13 @@ class _ConvNHWC(torch.autograd.Function):
             dx = ext.conv_dgrad(dy, w, ctx.stride, ctx.padding, x.shape[2], x.shape[3])
         if ctx.needs_input_grad[1]:
             dw = ext.conv_wgrad(x, dy, w.shape[2], w.shape[3], ctx.stride, ctx.padding)
-        return dx, dw, None, None
+        return dx, dw, None, None, None
 
 
-def conv2d(x, w, stride: int = 1, padding: int = 0):
+def conv2d(x, w, stride: int = 1, padding: int = 0, want_stats: bool = False):
+    """want_stats: also return the fused per-block BN statistic slabs
+    (psum, psumsq) computed in the conv epilogue — the following BatchNorm
+    then skips its own whole-tensor stats pass."""
     if _use_hip(x):
         C = x.shape[1]
         if C % 8 != 0:
@@ -67,7 +77,10 @@ def conv2d(x, w, stride: int = 1, padding: int = 0):
             w = F.pad(w, (0, 0, 0, 0, 0, padc)).contiguous(
                 memory_format=torch.channels_last
             )
-        return _ConvNHWC.apply(x, w, stride, padding)
+        y, ps, pss = _ConvNHWC.apply(x, w, stride, padding, want_stats)
+        if want_stats:
+            return y, ps, pss
+        return y
     return F.conv2d(x, w, None, stride, padding)
 
 
@@ -76,12 +89,12 @@ def conv2d(x, w, stride: int = 1, padding: int = 0):
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum,
-                eps, fuse_relu, residual):
+                eps, fuse_relu, residual, pre_psum=None, pre_psumsq=None):
         ext = require_ext()
         if training:
             y, save_mean, save_invstd = ext.bn_fwd_train(
                 x, gamma, beta, running_mean, running_var, momentum, eps,
-                fuse_relu, residual
+                fuse_relu, residual, pre_psum, pre_psumsq
             )
             ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
             ctx.fuse_relu = fuse_relu
@@ -99,12 +112,13 @@ class _BatchNormNHWC(torch.autograd.Function):
         out = ext.bn_bwd(x, dy, gamma, save_mean, save_invstd,
                          ctx.fuse_relu, y, ctx.has_residual)
         dres = out[3] if ctx.has_residual else None
-        return (out[0], out[1], out[2], None, None, None, None, None, None, dres)
+        return (out[0], out[1], out[2], None, None, None, None, None, None, dres,
+                None, None)
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
                momentum: float = 0.9, eps: float = 1e-5, fuse_relu: bool = False,
-               residual=None):
+               residual=None, pre_stats=None):
     """NHWC batch norm. `momentum` follows mxnet semantics:
     moving = moving*momentum + batch*(1-momentum) (reference batch_norm-inl.h),
     i.e. torch's momentum is (1 - mxnet momentum).
@@ -112,8 +126,10 @@ def batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
     `residual`: fused y = [relu](bn(x) + residual) — the resnet block tail in
     one pass (bn_apply's extra read beats a separate add_relu round trip)."""
     if _use_hip(x):
+        ps, pss = pre_stats if pre_stats is not None else (None, None)
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
-                                    training, momentum, eps, fuse_relu, residual)
+                                    training, momentum, eps, fuse_relu, residual,
+                                    ps, pss)
     y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
                      1.0 - momentum, eps)
     if residual is not None:

@@ -30,6 +30,20 @@ class Conv2dNHWC(nn.Module):
         self.bias = nn.Parameter(torch.zeros(out_channels)) if bias else None
 
     def forward(self, x):
+        # bias-free training convs can emit fused BN-stat slabs for the
+        # (conventionally following) BatchNorm; carried as a tensor attribute
+        # so unrelated consumers simply ignore them. Off by default: the
+        # epilogue-side cost slightly outweighs the saved BN stats pass on
+        # resnet-50 (measured -3.5%); enable with DTMX_FUSE_BN_STATS=1 for
+        # stats-bound models.
+        import os as _os
+        if (self.bias is None and x.is_cuda and self.training
+                and _os.environ.get("DTMX_FUSE_BN_STATS", "0") == "1"):
+            y, ps, pss = DF.conv2d(x, self.weight, self.stride, self.padding,
+                                   want_stats=True)
+            if ps.numel():
+                y._dtmx_bn_stats = (ps, pss)
+            return y
         y = DF.conv2d(x, self.weight, self.stride, self.padding)
         if self.bias is not None:
             y = y + self.bias.reshape(1, -1, 1, 1).to(y.dtype)
@@ -64,6 +78,7 @@ class BatchNorm2dNHWC(nn.Module):
         return self
 
     def forward(self, x, residual=None):
+        pre_stats = getattr(x, "_dtmx_bn_stats", None) if self.training else None
         return DF.batch_norm(
             x,
             self.weight,
@@ -75,6 +90,7 @@ class BatchNorm2dNHWC(nn.Module):
             self.eps,
             self.fuse_relu,
             residual,
+            pre_stats,
         )
 
 

@@ -132,7 +132,7 @@ def test_bn_fwd_train_and_bwd():
     rm = torch.zeros(C, dtype=torch.float32, device=DEV)
     rv = torch.ones(C, dtype=torch.float32, device=DEV)
     y, sm, si = ext.bn_fwd_train(nhwc(x), gamma.to(DEV), beta.to(DEV), rm, rv,
-                                 0.9, 1e-5, False, None)
+                                 0.9, 1e-5, False, None, None, None)
     xf = x.float().requires_grad_(True)
     ref = F.batch_norm(xf, None, None, gamma.float(), beta.float(), True, 0.1, 1e-5)
     assert_close(y.contiguous(), ref, name="bn_fwd")
@@ -160,7 +160,7 @@ def test_bn_fused_relu():
     rm = torch.zeros(C, dtype=torch.float32, device=DEV)
     rv = torch.ones(C, dtype=torch.float32, device=DEV)
     y, sm, si = ext.bn_fwd_train(nhwc(x), gamma.to(DEV), beta.to(DEV), rm, rv,
-                                 0.9, 1e-5, True, None)
+                                 0.9, 1e-5, True, None, None, None)
     ref = F.relu(F.batch_norm(x.float(), None, None, gamma.float(), beta.float(),
                               True, 0.1, 1e-5))
     assert_close(y.contiguous(), ref, name="bn_relu")
@@ -326,3 +326,42 @@ def test_resnet18_training_step_runs():
     assert all(np.isfinite(losses)), losses
     # same fixed batch replayed: loss must drop
     assert losses[-1] < losses[0], losses
+
+
+def test_conv_fwd_fused_bn_stats():
+    """conv_fwd_stats slab sums must equal per-channel sums of y."""
+    from dtmx.ops.hip import require_ext
+    ext = require_ext()
+    N, C, H, K, R = 32, 64, 28, 128, 3
+    x = mk((N, C, H, H), seed=30)
+    w = mk((K, C, R, R), scale=0.1, seed=31)
+    y, ps, pss = ext.conv_fwd_stats(nhwc(x), nhwc(w), 1, 1)
+    assert ps.numel() > 0
+    yf = y.contiguous().float()
+    ref_sum = yf.sum(dim=(0, 2, 3)).cpu()
+    ref_sumsq = (yf * yf).sum(dim=(0, 2, 3)).cpu()
+    assert_close(ps.sum(0), ref_sum, rtol=0.01, name="fused psum")
+    assert_close(pss.sum(0), ref_sumsq, rtol=0.01, name="fused psumsq")
+
+
+def test_bn_with_fused_stats_matches_plain():
+    import dtmx
+    from dtmx.ops import functional as DF
+    from dtmx.ops.hip import require_ext
+    ext = require_ext()
+    N, C, H, K, R = 32, 64, 28, 64, 3
+    x = mk((N, C, H, H), seed=32)
+    w = mk((K, C, R, R), scale=0.1, seed=33)
+    y, ps, pss = ext.conv_fwd_stats(nhwc(x), nhwc(w), 1, 1)
+    gamma = torch.ones(C).bfloat16().to(DEV)
+    beta = torch.zeros(C).bfloat16().to(DEV)
+    rm1 = torch.zeros(C, dtype=torch.float32, device=DEV)
+    rv1 = torch.ones(C, dtype=torch.float32, device=DEV)
+    out_fused, sm1, si1 = ext.bn_fwd_train(y, gamma, beta, rm1, rv1, 0.9, 1e-5,
+                                           False, None, ps, pss)
+    rm2 = torch.zeros(C, dtype=torch.float32, device=DEV)
+    rv2 = torch.ones(C, dtype=torch.float32, device=DEV)
+    out_plain, sm2, si2 = ext.bn_fwd_train(y, gamma, beta, rm2, rv2, 0.9, 1e-5,
+                                           False, None, None, None)
+    assert_close(out_fused, out_plain, rtol=0.01, name="bn fused-vs-plain")
+    assert_close(sm1, sm2, rtol=0.01, name="save_mean")
