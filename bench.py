@@ -116,11 +116,18 @@ def main():
     n_gpus = world if dist_mode else 1
     imgs = B * n_gpus * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
-    baseline = 363.69  # 1xV100 fp32 bs128; weak scaling vs 1 GPU
+    # reference training baselines, 1xV100 fp32 (BASELINE.md / docs/faq/perf.md)
+    baselines = {"resnet": 363.69, "inception-v3": 253.68, "alexnet": 2919.02}
+    baseline = baselines.get(args.network)
+    uses_layers = args.network in ("resnet", "resnet-v1", "vgg")
+    model_name = (f"{args.network}-{args.num_layers}" if uses_layers
+                  else args.network)
 
     if rank == 0:
         print(json.dumps({
-            "metric": "images/sec ResNet-50 ImageNet-shape",
+            "metric": (f"images/sec ResNet-{args.num_layers} ImageNet-shape"
+                       if args.network.startswith("resnet")
+                       else f"images/sec {model_name} ImageNet-shape"),
             "value": round(imgs, 2),
             "unit": "images/sec",
             "n_gpus": n_gpus,
@@ -129,11 +136,11 @@ def main():
             "ms_per_step": round(ms_per_step, 3),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": round(imgs / baseline, 3),
+            "vs_baseline": round(imgs / baseline, 3) if baseline else None,
             "dtype": str(dtype).replace("torch.", ""),
             "data": "synthetic",
             "config": {
-                "model": f"{args.network}-{args.num_layers}",
+                "model": model_name,
                 "global_batch": B * n_gpus,
                 "seq_len": None,
                 "image_shape": args.image_shape,

@@ -42,6 +42,14 @@ void sgd_mom_mp_dev(at::Tensor, const at::Tensor&, at::Tensor, at::Tensor,
                     const at::Tensor&);
 void sgd_mom_f32(at::Tensor, const at::Tensor&, at::Tensor, double, double,
                  double, double, double);
+// dropout_ln.hip
+std::vector<at::Tensor> dropout_fwd(const at::Tensor&, double, int64_t);
+at::Tensor dropout_bwd(const at::Tensor&, const at::Tensor&, double);
+std::vector<at::Tensor> layer_norm_fwd(const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, double);
+std::vector<at::Tensor> layer_norm_bwd(const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&);
 // compress.hip
 at::Tensor quantize_2bit(const at::Tensor&, at::Tensor, double);
 at::Tensor dequantize_2bit(const at::Tensor&, long, double);
@@ -72,6 +80,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_mom_mp", &dtmx::sgd_mom_mp);
   m.def("sgd_mom_mp_dev", &dtmx::sgd_mom_mp_dev);
   m.def("sgd_mom_f32", &dtmx::sgd_mom_f32);
+  m.def("dropout_fwd", &dtmx::dropout_fwd);
+  m.def("dropout_bwd", &dtmx::dropout_bwd);
+  m.def("layer_norm_fwd", &dtmx::layer_norm_fwd);
+  m.def("layer_norm_bwd", &dtmx::layer_norm_bwd);
   m.def("quantize_2bit", &dtmx::quantize_2bit);
   m.def("dequantize_2bit", &dtmx::dequantize_2bit);
   dtmx::register_recordio(m);

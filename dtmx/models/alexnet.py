@@ -4,7 +4,7 @@ from __future__ import annotations
 
 import torch.nn as nn
 
-from ..ops.layers import Conv2dNHWC, LinearBF16, MaxPool2dNHWC, ReLU
+from ..ops.layers import Conv2dNHWC, Dropout, LinearBF16, MaxPool2dNHWC, ReLU
 
 
 class AlexNet(nn.Module):
@@ -19,8 +19,8 @@ class AlexNet(nn.Module):
             Conv2dNHWC(256, 256, 3, 1, 1, bias=True), ReLU(), MaxPool2dNHWC(3, 2),
         )
         self.classifier = nn.Sequential(
-            LinearBF16(256 * 6 * 6, 4096), ReLU(), nn.Dropout(0.5),
-            LinearBF16(4096, 4096), ReLU(), nn.Dropout(0.5),
+            LinearBF16(256 * 6 * 6, 4096), ReLU(), Dropout(0.5),
+            LinearBF16(4096, 4096), ReLU(), Dropout(0.5),
             LinearBF16(4096, num_classes),
         )
 

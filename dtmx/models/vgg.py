@@ -4,7 +4,7 @@ from __future__ import annotations
 
 import torch.nn as nn
 
-from ..ops.layers import Conv2dNHWC, LinearBF16, MaxPool2dNHWC, ReLU
+from ..ops.layers import Conv2dNHWC, Dropout, LinearBF16, MaxPool2dNHWC, ReLU
 
 _CFG = {
     11: [64, "M", 128, "M", 256, 256, "M", 512, 512, "M", 512, 512, "M"],
@@ -29,8 +29,8 @@ class VGG(nn.Module):
                 in_ch = v
         self.features = nn.Sequential(*layers)
         self.classifier = nn.Sequential(
-            LinearBF16(512 * 7 * 7, 4096), ReLU(), nn.Dropout(0.5),
-            LinearBF16(4096, 4096), ReLU(), nn.Dropout(0.5),
+            LinearBF16(512 * 7 * 7, 4096), ReLU(), Dropout(0.5),
+            LinearBF16(4096, 4096), ReLU(), Dropout(0.5),
             LinearBF16(4096, num_classes),
         )
 

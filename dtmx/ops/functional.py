@@ -296,3 +296,61 @@ def softmax_cross_entropy_sum(logits, label):
     if _use_hip(logits):
         return _SoftmaxCE.apply(logits, label.to(torch.int32))
     return F.cross_entropy(logits.float(), label.long(), reduction="sum")
+
+
+# ------------------------------------------------------- dropout / layernorm
+
+class _Dropout(torch.autograd.Function):
+    """Mask-gen + scale dropout (reference src/operator/nn/dropout.cu) on the
+    stateless splitmix64 counter RNG — no cuRAND state arrays."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = require_ext()
+        y, mask = ext.dropout_fwd(x, p, seed)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        (mask,) = ctx.saved_tensors
+        return ext.dropout_bwd(dy, mask, ctx.p).view_as(dy), None, None
+
+
+def dropout(x, p: float, training: bool, seed: int | None = None):
+    if not training or p <= 0.0:
+        return x
+    if _use_hip(x):
+        if seed is None:
+            seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+        return _Dropout.apply(x, float(p), seed)
+    return F.dropout(x, p, training)
+
+
+class _LayerNorm(torch.autograd.Function):
+    """Row LayerNorm (reference src/operator/nn/layer_norm.cu): one block per
+    row, fp32 accumulation; dgamma/dbeta via fp32 atomics."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        ext = require_ext()
+        y, mean, rstd = ext.layer_norm_fwd(x, gamma, beta, eps)
+        ctx.save_for_backward(x, gamma, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        x, gamma, mean, rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = ext.layer_norm_bwd(x, dy, gamma, mean, rstd)
+        return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
+
+
+def layer_norm(x, gamma, beta, eps: float = 1e-5):
+    """Normalize over the last dim of a 2D input."""
+    if _use_hip(x) and x.dim() == 2:
+        return _LayerNorm.apply(x, gamma, beta, eps)
+    return F.layer_norm(x.float(), (x.shape[-1],), gamma.float(), beta.float(),
+                        eps).to(x.dtype)

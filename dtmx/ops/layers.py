@@ -132,3 +132,37 @@ class LinearBF16(nn.Module):
 
     def forward(self, x):
         return DF.linear(x, self.weight, self.bias)
+
+
+class Dropout(nn.Module):
+    """Train-time mask+scale dropout on the HIP counter-RNG kernel
+    (reference nn/dropout.cu semantics; torch fallback on CPU)."""
+
+    def __init__(self, p: float = 0.5):
+        super().__init__()
+        self.p = p
+
+    def forward(self, x):
+        return DF.dropout(x, self.p, self.training)
+
+    def extra_repr(self):
+        return f"p={self.p}"
+
+
+class LayerNorm(nn.Module):
+    """Last-dim LayerNorm with fp32 gamma/beta (reference nn/layer_norm.cu)."""
+
+    def __init__(self, dim: int, eps: float = 1e-5):
+        super().__init__()
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(dim, dtype=torch.float32))
+        self.bias = nn.Parameter(torch.zeros(dim, dtype=torch.float32))
+
+    def _apply(self, fn, recurse=True):  # keep affine params fp32 under .to(bf16)
+        dev = fn(torch.empty(0))
+        self.weight.data = self.weight.data.to(device=dev.device)
+        self.bias.data = self.bias.data.to(device=dev.device)
+        return self
+
+    def forward(self, x):
+        return DF.layer_norm(x, self.weight, self.bias, self.eps)

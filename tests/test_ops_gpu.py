@@ -365,3 +365,50 @@ def test_bn_with_fused_stats_matches_plain():
                                            False, None, None, None)
     assert_close(out_fused, out_plain, rtol=0.01, name="bn fused-vs-plain")
     assert_close(sm1, sm2, rtol=0.01, name="save_mean")
+
+
+# ---------------------------------------------------- dropout / layer norm
+
+def test_dropout_fwd_bwd():
+    from dtmx.ops import functional as DF
+    torch.manual_seed(3)
+    x = mk((64, 4096), seed=11).to(DEV).requires_grad_(True)
+    p = 0.5
+    y = DF.dropout(x, p, training=True, seed=1234)
+    keep = (y != 0) | (x == 0)
+    frac = keep.float().mean().item()
+    assert abs(frac - (1 - p)) < 0.02  # RNG keep-rate
+    # kept elements are scaled by 1/(1-p)
+    m = y != 0
+    assert_close(y[m], x.detach()[m] * 2.0, name="dropout scale")
+    # backward applies the SAME mask and scale
+    dy = mk((64, 4096), seed=12).to(DEV)
+    y.backward(dy)
+    assert_close(x.grad[m], dy[m].float() * 2.0, name="dropout bwd kept")
+    assert x.grad[~m].abs().max().item() == 0.0
+    # determinism: same seed -> same mask
+    y2 = DF.dropout(x.detach(), p, training=True, seed=1234)
+    assert torch.equal((y2 != 0), (y != 0))
+    # eval mode: identity
+    assert DF.dropout(x.detach(), p, training=False) is x.detach() or torch.equal(
+        DF.dropout(x.detach(), p, training=False), x.detach())
+
+
+def test_layer_norm_fwd_bwd():
+    from dtmx.ops import functional as DF
+    B, D = 96, 1000
+    x = mk((B, D), seed=21).to(DEV).requires_grad_(True)
+    gamma = (torch.randn(D) * 0.2 + 1.0).to(DEV).requires_grad_(True)
+    beta = (torch.randn(D) * 0.1).to(DEV).requires_grad_(True)
+    y = DF.layer_norm(x, gamma, beta)
+    xf = x.detach().float().requires_grad_(True)
+    gf = gamma.detach().float().requires_grad_(True)
+    bf_ = beta.detach().float().requires_grad_(True)
+    yref = F.layer_norm(xf, (D,), gf, bf_, 1e-5)
+    assert_close(y, yref, name="ln fwd")
+    dy = mk((B, D), seed=22).to(DEV)
+    y.backward(dy)
+    yref.backward(dy.float())
+    assert_close(x.grad, xf.grad, name="ln dx")
+    assert_close(gamma.grad, gf.grad, rtol=0.03, name="ln dgamma")
+    assert_close(beta.grad, bf_.grad, rtol=0.03, name="ln dbeta")
