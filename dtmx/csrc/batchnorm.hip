@@ -623,4 +623,130 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   return {dx, dgamma, dbeta};
 }
 
+// ---- SyncBatchNorm entries (reference contrib/sync_batch_norm.cu) --------
+// Cross-rank BN = local per-channel sums -> RCCL all-reduce (Python side) ->
+// finalize/apply with the GLOBAL count. These entries split bn_fwd_train /
+// bn_bwd at exactly that seam, reusing the same kernels.
+
+std::vector<at::Tensor> bn_local_sums(const at::Tensor& x) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8, got ", C);
+  uint32_t rows = N * H * W, cvecs = C / 8;
+  uint32_t cpb = bn_cpb(cvecs);
+  auto opt_f = x.options().dtype(at::kFloat);
+  dim3 grid;
+  uint32_t rpb;
+  bn_grid(rows, cvecs, cpb, grid, rpb);
+  uint32_t nslabs = grid.y;
+  auto psum = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto psumsq = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto sum = at::empty({(long)C}, opt_f), sumsq = at::empty({(long)C}, opt_f);
+  auto s = bn_stream();
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_local_sums", {
+    bn_stats_kernel<<<grid, 256, 0, s>>>((const elem_t*)x.data_ptr(),
+                                         psum.data_ptr<float>(),
+                                         psumsq.data_ptr<float>(), rows, cvecs,
+                                         cpb, rpb);
+  });
+  uint32_t ncv = std::min(cvecs, 8u);
+  slab_reduce2_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+      psum.data_ptr<float>(), psumsq.data_ptr<float>(), sum.data_ptr<float>(),
+      sumsq.data_ptr<float>(), C, nslabs);
+  return {sum, sumsq};
+}
+
+std::vector<at::Tensor> bn_fwd_presummed(
+    const at::Tensor& x, const at::Tensor& gamma, const at::Tensor& beta,
+    at::Tensor running_mean, at::Tensor running_var, double momentum, double eps,
+    bool fuse_relu, const c10::optional<at::Tensor>& residual,
+    const at::Tensor& sum, const at::Tensor& sumsq, long count) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "bn: x must be NHWC");
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  uint32_t rows = N * H * W, cvecs = C / 8;
+  auto opt_f = x.options().dtype(at::kFloat);
+  auto save_mean = at::empty({(long)C}, opt_f), save_invstd = at::empty({(long)C}, opt_f);
+  auto scale = at::empty({(long)C}, opt_f), shift = at::empty({(long)C}, opt_f);
+  auto y = at::empty_like(x);
+  auto s = bn_stream();
+  (void)rows;
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_fwd_presummed", {
+    bn_finalize_kernel<<<(C + 255) / 256, 256, 0, s>>>(
+        sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+        (const elem_t*)gamma.data_ptr(), (const elem_t*)beta.data_ptr(),
+        running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
+        scale.data_ptr<float>(), shift.data_ptr<float>(), C, /*nslabs=*/1,
+        (uint32_t)count, momentum, eps);
+    uint32_t total8 = N * H * W * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_apply_kernel<<<blocks, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
+        scale.data_ptr<float>(), shift.data_ptr<float>(),
+        residual.has_value() ? (const elem_t*)residual->data_ptr() : nullptr,
+        total8, dcv, fuse_relu ? 1 : 0);
+  });
+  return {y, save_mean, save_invstd};
+}
+
+std::vector<at::Tensor> bn_bwd_sums(const at::Tensor& x, const at::Tensor& dy,
+                                    const at::Tensor& y,
+                                    const at::Tensor& save_mean,
+                                    const at::Tensor& save_invstd, bool fuse_relu) {
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  uint32_t rows = N * H * W, cvecs = C / 8;
+  uint32_t cpb = bn_cpb(cvecs);
+  auto opt_f = x.options().dtype(at::kFloat);
+  dim3 grid;
+  uint32_t rpb;
+  bn_grid(rows, cvecs, cpb, grid, rpb);
+  uint32_t nslabs = grid.y;
+  auto pdb = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto pdg = at::empty({(long)nslabs, (long)C}, opt_f);
+  auto tdb = at::empty({(long)C}, opt_f), tdg = at::empty({(long)C}, opt_f);
+  auto s = bn_stream();
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_bwd_sums", {
+    bn_bwd_stats_kernel<<<grid, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
+        (const elem_t*)y.data_ptr(), save_mean.data_ptr<float>(),
+        save_invstd.data_ptr<float>(), pdb.data_ptr<float>(),
+        pdg.data_ptr<float>(), rows, cvecs, cpb, rpb, fuse_relu ? 1 : 0);
+  });
+  uint32_t ncv = std::min(cvecs, 8u);
+  slab_reduce2_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+      pdb.data_ptr<float>(), pdg.data_ptr<float>(), tdb.data_ptr<float>(),
+      tdg.data_ptr<float>(), C, nslabs);
+  return {tdb, tdg};
+}
+
+std::vector<at::Tensor> bn_bwd_dx_presummed(
+    const at::Tensor& x, const at::Tensor& dy, const at::Tensor& y,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& gamma, const at::Tensor& tdb, const at::Tensor& tdg,
+    long count, bool fuse_relu, bool want_dres) {
+  uint32_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  uint32_t cvecs = C / 8;
+  auto dx = at::empty_like(x);
+  at::Tensor dres;
+  if (want_dres) dres = at::empty_like(x);
+  auto s = bn_stream();
+  DTMX_DISPATCH_16(x.scalar_type(), "bn_bwd_dx_presummed", {
+    uint32_t total8 = N * H * W * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_bwd_dx_kernel<<<blocks, 256, 0, s>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
+        (const elem_t*)y.data_ptr(), save_mean.data_ptr<float>(),
+        save_invstd.data_ptr<float>(), (const elem_t*)gamma.data_ptr(),
+        tdb.data_ptr<float>(), tdg.data_ptr<float>(), (elem_t*)dx.data_ptr(),
+        want_dres ? (elem_t*)dres.data_ptr() : nullptr, total8, dcv,
+        1.f / (float)count, fuse_relu ? 1 : 0);
+  });
+  if (want_dres) return {dx, dres};
+  return {dx};
+}
+
 }  // namespace dtmx
