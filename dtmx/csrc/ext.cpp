@@ -18,6 +18,21 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor&, const at::Tensor&,
                                      const c10::optional<at::Tensor>&);
 std::vector<at::Tensor> conv_fwd_stats(const at::Tensor&, const at::Tensor&,
                                        long, long);
+std::vector<at::Tensor> bn_local_sums(const at::Tensor&);
+std::vector<at::Tensor> bn_fwd_presummed(const at::Tensor&, const at::Tensor&,
+                                         const at::Tensor&, at::Tensor,
+                                         at::Tensor, double, double, bool,
+                                         const c10::optional<at::Tensor>&,
+                                         const at::Tensor&, const at::Tensor&,
+                                         long);
+std::vector<at::Tensor> bn_bwd_sums(const at::Tensor&, const at::Tensor&,
+                                    const at::Tensor&, const at::Tensor&,
+                                    const at::Tensor&, bool);
+std::vector<at::Tensor> bn_bwd_dx_presummed(const at::Tensor&, const at::Tensor&,
+                                            const at::Tensor&, const at::Tensor&,
+                                            const at::Tensor&, const at::Tensor&,
+                                            const at::Tensor&, const at::Tensor&,
+                                            long, bool, bool);
 at::Tensor bn_fwd_infer(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                         const at::Tensor&, const at::Tensor&, double, bool,
                         const c10::optional<at::Tensor>&);
@@ -68,6 +83,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &dtmx::bn_fwd_train);
   m.def("bn_fwd_infer", &dtmx::bn_fwd_infer);
   m.def("bn_bwd", &dtmx::bn_bwd);
+  m.def("bn_local_sums", &dtmx::bn_local_sums);
+  m.def("bn_fwd_presummed", &dtmx::bn_fwd_presummed);
+  m.def("bn_bwd_sums", &dtmx::bn_bwd_sums);
+  m.def("bn_bwd_dx_presummed", &dtmx::bn_bwd_dx_presummed);
   m.def("maxpool_fwd", &dtmx::maxpool_fwd);
   m.def("maxpool_bwd", &dtmx::maxpool_bwd);
   m.def("global_avgpool_fwd", &dtmx::global_avgpool_fwd);

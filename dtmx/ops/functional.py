@@ -354,3 +354,94 @@ def layer_norm(x, gamma, beta, eps: float = 1e-5):
         return _LayerNorm.apply(x, gamma, beta, eps)
     return F.layer_norm(x.float(), (x.shape[-1],), gamma.float(), beta.float(),
                         eps).to(x.dtype)
+
+
+# ------------------------------------------------------- sync batch norm
+
+class _SyncBatchNormNHWC(torch.autograd.Function):
+    """Cross-rank BN (reference contrib/sync_batch_norm.cu): per-rank channel
+    sums -> all-reduce -> normalize with GLOBAL mean/var. dgamma/dbeta are the
+    LOCAL sums — the kvstore/bucketer gradient all-reduce makes them global
+    (same division of labor as torch.nn.SyncBatchNorm under DDP)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
+                group):
+        import torch.distributed as dist
+        world = dist.get_world_size(group)
+        rows = x.shape[0] * x.shape[2] * x.shape[3]
+        if x.is_cuda:
+            ext = require_ext()
+            s, ss = ext.bn_local_sums(x)
+        else:
+            xf = x.float()
+            s = xf.sum(dim=(0, 2, 3))
+            ss = (xf * xf).sum(dim=(0, 2, 3))
+        buf = torch.stack([s, ss])
+        dist.all_reduce(buf, group=group)
+        count = rows * world
+        if x.is_cuda:
+            y, save_mean, save_invstd = ext.bn_fwd_presummed(
+                x, gamma, beta, running_mean, running_var, momentum, eps,
+                False, None, buf[0].contiguous(), buf[1].contiguous(), count)
+        else:
+            mean = buf[0] / count
+            var = (buf[1] / count - mean * mean).clamp_min(0)
+            invstd = (var + eps).rsqrt()
+            save_mean, save_invstd = mean, invstd
+            xf = x.float()
+            y = ((xf - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
+                 * gamma.float().view(1, -1, 1, 1)
+                 + beta.float().view(1, -1, 1, 1)).to(x.dtype)
+            unbiased = var * count / (count - 1) if count > 1 else var
+            running_mean.mul_(momentum).add_(mean * (1 - momentum))
+            running_var.mul_(momentum).add_(unbiased * (1 - momentum))
+        ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+        ctx.count = count
+        ctx.group = group
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        import torch.distributed as dist
+        x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
+        if x.is_cuda:
+            ext = require_ext()
+            dy = dy.contiguous(memory_format=torch.channels_last)
+            tdb, tdg = ext.bn_bwd_sums(x, dy, y, save_mean, save_invstd, False)
+        else:
+            dyf = dy.float()
+            xhat = ((x.float() - save_mean.view(1, -1, 1, 1))
+                    * save_invstd.view(1, -1, 1, 1))
+            tdb = dyf.sum(dim=(0, 2, 3))
+            tdg = (dyf * xhat).sum(dim=(0, 2, 3))
+        dgamma = tdg.to(gamma.dtype)
+        dbeta = tdb.to(gamma.dtype)
+        buf = torch.stack([tdb, tdg])
+        dist.all_reduce(buf, group=ctx.group)
+        if x.is_cuda:
+            dx = ext.bn_bwd_dx_presummed(
+                x, dy, y, save_mean, save_invstd, gamma, buf[0].contiguous(),
+                buf[1].contiguous(), ctx.count, False, False)[0]
+        else:
+            xhat = ((x.float() - save_mean.view(1, -1, 1, 1))
+                    * save_invstd.view(1, -1, 1, 1))
+            g = gamma.float().view(1, -1, 1, 1)
+            inv = 1.0 / ctx.count
+            dx = (g * save_invstd.view(1, -1, 1, 1)
+                  * (dy.float() - buf[0].view(1, -1, 1, 1) * inv
+                     - xhat * buf[1].view(1, -1, 1, 1) * inv)).to(x.dtype)
+        return dx, dgamma, dbeta, None, None, None, None, None
+
+
+def sync_batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
+                    momentum: float = 0.9, eps: float = 1e-5, group=None):
+    """Cross-rank NHWC batch norm; falls back to local BN when not distributed
+    or in eval mode (running stats are already synchronized by construction)."""
+    import torch.distributed as dist
+    if (not training or not dist.is_available() or not dist.is_initialized()
+            or dist.get_world_size(group) == 1):
+        return batch_norm(x, gamma, beta, running_mean, running_var, training,
+                          momentum, eps)
+    return _SyncBatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
+                                    momentum, eps, group)

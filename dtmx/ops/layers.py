@@ -166,3 +166,20 @@ class LayerNorm(nn.Module):
 
     def forward(self, x):
         return DF.layer_norm(x, self.weight, self.bias, self.eps)
+
+
+class SyncBatchNorm2dNHWC(BatchNorm2dNHWC):
+    """Cross-rank BN (reference contrib/sync_batch_norm.cu): batch statistics
+    are all-reduced over the data-parallel group each step, so every rank
+    normalizes with the GLOBAL batch mean/var. Use for small per-rank batches
+    where local statistics are too noisy."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.9, process_group=None):
+        super().__init__(num_features, eps=eps, momentum=momentum, fuse_relu=False)
+        self.process_group = process_group
+
+    def forward(self, x, residual=None):
+        assert residual is None, "SyncBatchNorm: residual fusion not supported"
+        return DF.sync_batch_norm(x, self.weight, self.bias, self.running_mean,
+                                  self.running_var, self.training, self.momentum,
+                                  self.eps, self.process_group)

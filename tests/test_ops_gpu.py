@@ -412,3 +412,37 @@ def test_layer_norm_fwd_bwd():
     assert_close(x.grad, xf.grad, name="ln dx")
     assert_close(gamma.grad, gf.grad, rtol=0.03, name="ln dgamma")
     assert_close(beta.grad, bf_.grad, rtol=0.03, name="ln dbeta")
+
+
+def test_bn_presummed_path_matches_bn():
+    """SyncBN kernel seam: local_sums -> fwd_presummed(count=rows) and
+    bwd_sums -> bwd_dx_presummed must reproduce the plain BN path bit-for-bit
+    on a single rank (the all-reduce in between is the identity at world=1)."""
+    from dtmx.ops.hip import require_ext
+    ext = require_ext()
+    N, C, H, W = 8, 64, 14, 14
+    rows = N * H * W
+    x = nhwc(mk((N, C, H, W), seed=31))
+    dy = nhwc(mk((N, C, H, W), seed=32))
+    gamma = (torch.randn(C) * 0.2 + 1.0).to(torch.bfloat16).to(DEV)
+    beta = (torch.randn(C) * 0.1).to(torch.bfloat16).to(DEV)
+
+    rm1, rv1 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
+    rm2, rv2 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
+    y1, m1, i1 = ext.bn_fwd_train(x, gamma, beta, rm1, rv1, 0.9, 1e-5, False,
+                                  None, None, None)
+    s, ss = ext.bn_local_sums(x)
+    y2, m2, i2 = ext.bn_fwd_presummed(x, gamma, beta, rm2, rv2, 0.9, 1e-5,
+                                      False, None, s, ss, rows)
+    torch.testing.assert_close(y1, y2, rtol=0, atol=0)
+    torch.testing.assert_close(m1, m2)
+    torch.testing.assert_close(rm1, rm2)
+    torch.testing.assert_close(rv1, rv2)
+
+    dx1, dgamma1, dbeta1 = ext.bn_bwd(x, dy, gamma, m1, i1, False, y1, False)
+    tdb, tdg = ext.bn_bwd_sums(x, dy, y2, m2, i2, False)
+    torch.testing.assert_close(tdb, dbeta1.float(), rtol=2e-2, atol=1e-2)
+    torch.testing.assert_close(tdg, dgamma1.float(), rtol=2e-2, atol=1e-2)
+    (dx2,) = ext.bn_bwd_dx_presummed(x, dy, y2, m2, i2, gamma, tdb, tdg, rows,
+                                     False, False)
+    torch.testing.assert_close(dx1, dx2, rtol=0, atol=0)
