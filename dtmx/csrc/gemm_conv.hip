@@ -18,6 +18,9 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <map>
+#include <tuple>
+
 #include "dtmx_common.h"
 
 namespace dtmx {
@@ -702,6 +705,29 @@ __global__ void im2col_kernel(const elem_t* x, elem_t* out, uint32_t M,
   }
 }
 
+// Weight transpose for dgrad: (K,R,S,C) memory -> (C,R,S,K) memory.
+// One thread per 8-k output chunk: writes are dense 16B chunks (a wave emits
+// 1 KB contiguous); the 2B-strided reads hit the MALL — conv weights are
+// <=5 MB. Replaces at::permute().contiguous(), measured ~250 GB/s there.
+template <typename elem_t>
+__global__ void transpose_w_crsk_kernel(const elem_t* __restrict__ src,
+                                        elem_t* __restrict__ dst, uint32_t C,
+                                        uint32_t RS, uint32_t K, uint32_t total,
+                                        FastDiv dRSK8, FastDiv dK8) {
+  using V8 = typename E8<elem_t>::v8;
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  uint32_t c = dRSK8.div(i);
+  uint32_t rem = dRSK8.mod(i, c);
+  uint32_t rs = dK8.div(rem);
+  uint32_t k0 = dK8.mod(rem, rs) * 8;
+  elem_t tmp[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    tmp[j] = src[((size_t)(k0 + j) * RS + rs) * C + c];
+  *(V8*)(dst + ((size_t)c * RS + rs) * K + k0) = *(V8*)tmp;
+}
+
 template <typename elem_t>
 __global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
                                      elem_t* __restrict__ out, size_t total8) {
@@ -713,6 +739,27 @@ __global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
 #pragma unroll
     for (int e = 0; e < 8; ++e) o[e] = (elem_t)in[i * 8 + e];
     *(V8*)(out + i * 8) = o;
+  }
+}
+
+// drain-and-rezero variant for the persistent wgrad split-K accumulator:
+// casting is the only consumer, so zeroing here replaces a separate fill
+// before every wgrad launch (the fills were pure launch overhead, ~50/step).
+template <typename elem_t>
+__global__ void cast_f32_bf16_zero_kernel(float* __restrict__ in,
+                                          elem_t* __restrict__ out,
+                                          size_t total8) {
+  using V8 = typename E8<elem_t>::v8;
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    V8 o;
+    f32x4 z4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (elem_t)in[i * 8 + e];
+    *(V8*)(out + i * 8) = o;
+    *(f32x4*)(in + i * 8) = z4;
+    *(f32x4*)(in + i * 8 + 4) = z4;
   }
 }
 
@@ -728,6 +775,21 @@ static const elem_t* zero_page(const at::Tensor& like) {
 }
 
 static inline uint32_t ceil_div(uint32_t a, uint32_t b) { return (a + b - 1) / b; }
+
+// Persistent pre-zeroed fp32 accumulator for wgrad split-K atomics, keyed by
+// shape. The consumer (cast_f32_bf16_zero_kernel) re-zeroes it in the same
+// pass, so the per-call at::zeros fill disappears from the step.
+static at::Tensor wgrad_acc_ws(uint32_t Ko, uint32_t RSC, const at::Tensor& like) {
+  static std::map<std::tuple<int, long, long>, at::Tensor> cache;
+  auto key = std::make_tuple((int)like.get_device(), (long)Ko, (long)RSC);
+  auto it = cache.find(key);
+  if (it == cache.end())
+    it = cache
+             .emplace(key, at::zeros({(long)Ko, (long)RSC},
+                                     like.options().dtype(at::kFloat)))
+             .first;
+  return it->second;
+}
 
 static hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
@@ -1026,7 +1088,22 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
       Ko += padk;
     }
     auto dyc = dyk.contiguous();
-    auto wt = wtk.contiguous();  // W^T in (C,R,S,Ko) dense layout
+    // W^T in (C,R,S,Ko) dense layout — custom kernel when w is the plain
+    // channels_last parameter (the training path); at::copy otherwise
+    at::Tensor wt;
+    if (Ko == (uint32_t)w.size(0) && Ko % 8 == 0 &&
+        w.is_contiguous(at::MemoryFormat::ChannelsLast)) {
+      uint32_t RS = R * S, K8 = Ko / 8, total = C * RS * K8;
+      wt = at::empty({(long)C, (long)R, (long)S, (long)Ko}, dy.options());
+      FastDiv dRSK8, dK8;
+      dRSK8.init(RS * K8);
+      dK8.init(K8);
+      transpose_w_crsk_kernel<<<ceil_div(total, 256), 256, 0, cur_stream()>>>(
+          (const elem_t*)w.data_ptr(), (elem_t*)wt.data_ptr(), C, RS, Ko, total,
+          dRSK8, dK8);
+    } else {
+      wt = wtk.contiguous();
+    }
     auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
                         at::MemoryFormat::ChannelsLast);
     uint32_t M = N * H * W_, Ktot = R * S * Ko;
@@ -1091,7 +1168,7 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
       uint32_t ktiles = ceil_div(M, 64);
       uint32_t splitk = std::max<uint32_t>(
           1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
-      auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
+      auto dw32 = wgrad_acc_ws(Ko, RSC, x);  // pre-zeroed persistent workspace
       WgradDyA<elem_t> pa;
       pa.dy = (const elem_t*)dy.data_ptr();
       pa.zero = zero_page<elem_t>(x);
@@ -1105,10 +1182,13 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
       pb.dQ.init(Q); pb.dPQ.init(P * Q); pb.dC.init(C); pb.dS.init(S);
       EpiAtomicF32<elem_t> epi{dw32.data_ptr<float>(), Ko, RSC};
       launch_gemm_nt(pa, pb, epi, Ko, RSC, M, splitk);
-      auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
-                    .to(at::kBFloat16)
-                    .permute({0, 3, 1, 2});
-      return dw.contiguous(at::MemoryFormat::ChannelsLast);
+      // drain to 16-bit and re-zero the workspace for its next use
+      auto dwm = at::empty({(long)Ko, (long)R, (long)S, (long)C}, x.options());
+      size_t t8 = (size_t)Ko * RSC / 8;
+      cast_f32_bf16_zero_kernel<<<std::min<size_t>((t8 + 255) / 256, 2048), 256,
+                                  0, cur_stream()>>>(
+          dw32.data_ptr<float>(), (elem_t*)dwm.data_ptr(), t8);
+      return dwm.permute({0, 3, 1, 2});  // logical (K,C,R,S), channels_last
     }
 
     // dy^T : [NPQ][Ko] -> [Ko(+pad)][Mpad]
@@ -1152,12 +1232,23 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
     uint32_t ktiles = ceil_div(Mpad, 64);
     uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
 
-    auto dw32 = at::zeros({(long)Ko, (long)RSC}, x.options().dtype(at::kFloat));
+    const bool ws_ok = ((size_t)Ko * RSC) % 8 == 0;
+    auto dw32 = ws_ok ? wgrad_acc_ws(Ko, RSC, x)
+                      : at::zeros({(long)Ko, (long)RSC},
+                                  x.options().dtype(at::kFloat));
     DenseP<elem_t> pa{(const elem_t*)dyt.data_ptr(), zero_page<elem_t>(x), Ko, Mpad, Mpad};
     DenseP<elem_t> pb{(const elem_t*)xt.data_ptr(), zero_page<elem_t>(x), RSC, Mpad, Mpad};
     EpiAtomicF32<elem_t> epi{dw32.data_ptr<float>(), Ko, RSC};
     launch_gemm(pa, pb, epi, Ko, RSC, Mpad, splitk);
     // (Ko, R, S, C) fp32 -> bf16, viewed back to logical (K,C,R,S) channels_last
+    if (ws_ok) {
+      auto dwm = at::empty({(long)Ko, (long)R, (long)S, (long)C}, x.options());
+      size_t t8 = (size_t)Ko * RSC / 8;
+      cast_f32_bf16_zero_kernel<<<std::min<size_t>((t8 + 255) / 256, 2048), 256,
+                                  0, cur_stream()>>>(
+          dw32.data_ptr<float>(), (elem_t*)dwm.data_ptr(), t8);
+      return dwm.permute({0, 3, 1, 2});
+    }
     auto dw = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
                   .to(at::kBFloat16)
                   .permute({0, 3, 1, 2});
