@@ -113,6 +113,12 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
+    if os.environ.get("DTMX_PRINT_MEM") == "1" and device.type == "cuda":
+        print(f"[mem] rank {rank}: peak allocated "
+              f"{torch.cuda.max_memory_allocated() / 2**30:.1f} GiB, reserved "
+              f"{torch.cuda.max_memory_reserved() / 2**30:.1f} GiB",
+              file=sys.stderr)
+
     n_gpus = world if dist_mode else 1
     imgs = B * n_gpus * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0

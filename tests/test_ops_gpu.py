@@ -450,11 +450,12 @@ def test_bn_presummed_path_matches_bn():
 
 def test_conv_wgrad_workspace_rezero():
     """The persistent split-K accumulator must be re-zeroed by its drain pass:
-    calling wgrad twice with identical inputs must give identical results."""
+    calling wgrad twice with identical inputs must agree to within split-K
+    float-atomic reordering noise (a stale accumulator would double values)."""
     from dtmx.ops.hip import require_ext
     ext = require_ext()
     x = nhwc(mk((8, 64, 28, 28), seed=41))
     dy = nhwc(mk((8, 128, 28, 28), seed=42))
     d1 = ext.conv_wgrad(x, dy, 3, 3, 1, 1).clone()
     d2 = ext.conv_wgrad(x, dy, 3, 3, 1, 1)
-    torch.testing.assert_close(d1, d2, rtol=0, atol=0)
+    torch.testing.assert_close(d1, d2, rtol=0.02, atol=0.05)
