@@ -9,6 +9,7 @@ at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&);
 at::Tensor conv_fwd(const at::Tensor&, const at::Tensor&, long, long);
 at::Tensor conv_dgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
 at::Tensor conv_wgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
+std::vector<std::tuple<long, long, double>> wgrad_ws_stats();
 // batchnorm.hip
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor&, const at::Tensor&,
                                      const at::Tensor&, at::Tensor, at::Tensor,
@@ -80,6 +81,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd_stats", &dtmx::conv_fwd_stats);
   m.def("conv_dgrad", &dtmx::conv_dgrad);
   m.def("conv_wgrad", &dtmx::conv_wgrad);
+  m.def("wgrad_ws_stats", &dtmx::wgrad_ws_stats);
   m.def("bn_fwd_train", &dtmx::bn_fwd_train);
   m.def("bn_fwd_infer", &dtmx::bn_fwd_infer);
   m.def("bn_bwd", &dtmx::bn_bwd);

@@ -776,11 +776,21 @@ static const elem_t* zero_page(const at::Tensor& like) {
 
 static inline uint32_t ceil_div(uint32_t a, uint32_t b) { return (a + b - 1) / b; }
 
+static bool env_flag(const char* name) {
+  const char* v = getenv(name);
+  return v && v[0] == '1';
+}
+
 // Persistent pre-zeroed fp32 accumulator for wgrad split-K atomics, keyed by
 // shape. The consumer (cast_f32_bf16_zero_kernel) re-zeroes it in the same
 // pass, so the per-call at::zeros fill disappears from the step.
-static at::Tensor wgrad_acc_ws(uint32_t Ko, uint32_t RSC, const at::Tensor& like) {
+static std::map<std::tuple<int, long, long>, at::Tensor>& wgrad_ws_cache() {
   static std::map<std::tuple<int, long, long>, at::Tensor> cache;
+  return cache;
+}
+
+static at::Tensor wgrad_acc_ws(uint32_t Ko, uint32_t RSC, const at::Tensor& like) {
+  auto& cache = wgrad_ws_cache();
   auto key = std::make_tuple((int)like.get_device(), (long)Ko, (long)RSC);
   auto it = cache.find(key);
   if (it == cache.end())
@@ -789,6 +799,15 @@ static at::Tensor wgrad_acc_ws(uint32_t Ko, uint32_t RSC, const at::Tensor& like
                                      like.options().dtype(at::kFloat)))
              .first;
   return it->second;
+}
+
+// debug: max |entry| per cached workspace — all must be 0 between steps
+std::vector<std::tuple<long, long, double>> wgrad_ws_stats() {
+  std::vector<std::tuple<long, long, double>> out;
+  for (auto& kv : wgrad_ws_cache())
+    out.emplace_back(std::get<1>(kv.first), std::get<2>(kv.first),
+                     kv.second.abs().max().item<double>());
+  return out;
 }
 
 static hipStream_t cur_stream() {
@@ -1091,7 +1110,8 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
     // W^T in (C,R,S,Ko) dense layout — custom kernel when w is the plain
     // channels_last parameter (the training path); at::copy otherwise
     at::Tensor wt;
-    if (Ko == (uint32_t)w.size(0) && Ko % 8 == 0 &&
+    static const bool no_wtk = env_flag("DTMX_DISABLE_WT_KERNEL");
+    if (!no_wtk && Ko == (uint32_t)w.size(0) && Ko % 8 == 0 &&
         w.is_contiguous(at::MemoryFormat::ChannelsLast)) {
       uint32_t RS = R * S, K8 = Ko / 8, total = C * RS * K8;
       wt = at::empty({(long)C, (long)R, (long)S, (long)Ko}, dy.options());
@@ -1168,7 +1188,10 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
       uint32_t ktiles = ceil_div(M, 64);
       uint32_t splitk = std::max<uint32_t>(
           1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
-      auto dw32 = wgrad_acc_ws(Ko, RSC, x);  // pre-zeroed persistent workspace
+      static const bool no_ws = env_flag("DTMX_DISABLE_WGRAD_WS");
+      auto dw32 = no_ws ? at::zeros({(long)Ko, (long)RSC},
+                                    x.options().dtype(at::kFloat))
+                        : wgrad_acc_ws(Ko, RSC, x);  // pre-zeroed persistent
       WgradDyA<elem_t> pa;
       pa.dy = (const elem_t*)dy.data_ptr();
       pa.zero = zero_page<elem_t>(x);
@@ -1182,6 +1205,12 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
       pb.dQ.init(Q); pb.dPQ.init(P * Q); pb.dC.init(C); pb.dS.init(S);
       EpiAtomicF32<elem_t> epi{dw32.data_ptr<float>(), Ko, RSC};
       launch_gemm_nt(pa, pb, epi, Ko, RSC, M, splitk);
+      if (no_ws) {
+        auto dwo = dw32.reshape({(long)Ko, (long)R, (long)S, (long)C})
+                       .to(x.scalar_type())
+                       .permute({0, 3, 1, 2});
+        return dwo.contiguous(at::MemoryFormat::ChannelsLast);
+      }
       // drain to 16-bit and re-zero the workspace for its next use
       auto dwm = at::empty({(long)Ko, (long)R, (long)S, (long)C}, x.options());
       size_t t8 = (size_t)Ko * RSC / 8;
@@ -1232,7 +1261,8 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
     uint32_t ktiles = ceil_div(Mpad, 64);
     uint32_t splitk = std::max<uint32_t>(1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
 
-    const bool ws_ok = ((size_t)Ko * RSC) % 8 == 0;
+    static const bool no_ws2 = env_flag("DTMX_DISABLE_WGRAD_WS");
+    const bool ws_ok = !no_ws2 && ((size_t)Ko * RSC) % 8 == 0;
     auto dw32 = ws_ok ? wgrad_acc_ws(Ko, RSC, x)
                       : at::zeros({(long)Ko, (long)RSC},
                                   x.options().dtype(at::kFloat));

@@ -18,6 +18,8 @@ from __future__ import annotations
 
 from typing import Optional, Tuple
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -322,7 +324,7 @@ class _Dropout(torch.autograd.Function):
 def dropout(x, p: float, training: bool, seed: int | None = None):
     if not training or p <= 0.0:
         return x
-    if _use_hip(x):
+    if _use_hip(x) and os.environ.get("DTMX_TORCH_DROPOUT") != "1":
         if seed is None:
             seed = int(torch.randint(0, 2 ** 62, (1,)).item())
         return _Dropout.apply(x, float(p), seed)
