@@ -26,7 +26,12 @@ import torch.nn.functional as F
 from .hip import require_ext
 
 
-def _use_hip(*tensors) -> bool:
+_FALLBACK = set(os.environ.get("DTMX_FALLBACK", "").split(",")) - {""}
+
+
+def _use_hip(*tensors, op=None) -> bool:
+    if op is not None and op in _FALLBACK:  # debug: force torch fallback
+        return False
     return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
 
 
@@ -65,7 +70,7 @@ def conv2d(x, w, stride: int = 1, padding: int = 0, want_stats: bool = False):
     """want_stats: also return the fused per-block BN statistic slabs
     (psum, psumsq) computed in the conv epilogue — the following BatchNorm
     then skips its own whole-tensor stats pass."""
-    if _use_hip(x):
+    if _use_hip(x, op="conv"):
         C = x.shape[1]
         if C % 8 != 0:
             # zero-pad channels to 8 (the 3-channel stem) so the implicit-GEMM
@@ -127,7 +132,7 @@ def batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
 
     `residual`: fused y = [relu](bn(x) + residual) — the resnet block tail in
     one pass (bn_apply's extra read beats a separate add_relu round trip)."""
-    if _use_hip(x):
+    if _use_hip(x, op="bn"):
         ps, pss = pre_stats if pre_stats is not None else (None, None)
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
                                     training, momentum, eps, fuse_relu, residual,
@@ -164,7 +169,7 @@ class _MaxPoolNHWC(torch.autograd.Function):
 
 
 def max_pool2d(x, kernel: int, stride: int, padding: int = 0):
-    if _use_hip(x):
+    if _use_hip(x, op="pool"):
         return _MaxPoolNHWC.apply(x, kernel, stride, padding)
     return F.max_pool2d(x, kernel, stride, padding)
 
@@ -185,7 +190,7 @@ class _GlobalAvgPoolNHWC(torch.autograd.Function):
 
 def global_avg_pool(x):
     """(N,C,H,W) -> (N,C)"""
-    if _use_hip(x):
+    if _use_hip(x, op="gap"):
         return _GlobalAvgPoolNHWC.apply(x)
     return x.mean(dim=(2, 3))
 
@@ -208,7 +213,7 @@ class _ReLU(torch.autograd.Function):
 
 
 def relu(x):
-    if _use_hip(x):
+    if _use_hip(x, op="relu"):
         return _ReLU.apply(x)
     return F.relu(x)
 
@@ -231,7 +236,7 @@ class _AddRelu(torch.autograd.Function):
 
 def add_relu(a, b):
     """Fused residual add + ReLU (resnet hot path: one HBM round trip)."""
-    if _use_hip(a):
+    if _use_hip(a, op="relu"):
         return _AddRelu.apply(a, b)
     return F.relu(a + b)
 
@@ -267,7 +272,7 @@ class _Linear(torch.autograd.Function):
 
 def linear(x, w, bias=None):
     """x:(M,K) w:(N,K) -> (M,N)   (FullyConnected: X @ W^T + b)"""
-    if _use_hip(x):
+    if _use_hip(x, op="linear"):
         return _Linear.apply(x, w, bias)
     return F.linear(x, w, bias)
 
@@ -295,7 +300,7 @@ class _SoftmaxCE(torch.autograd.Function):
 
 
 def softmax_cross_entropy_sum(logits, label):
-    if _use_hip(logits):
+    if _use_hip(logits, op="softmax"):
         return _SoftmaxCE.apply(logits, label.to(torch.int32))
     return F.cross_entropy(logits.float(), label.long(), reduction="sum")
 
