@@ -150,19 +150,28 @@ struct EpiBF16 {
   // when bias == nullptr && !relu (stats must equal the written values).
   float* bn_psum = nullptr;
   float* bn_psumsq = nullptr;
-  // coalesced row-chunk store used by the kernel's LDS-staged epilogue
+  // coalesced row-chunk store used by the kernel's LDS-staged epilogue.
+  // N need not be a multiple of 8 (e.g. FullyConnected num_classes=100):
+  // the tail chunk is stored element-wise — a full V8 there would stomp the
+  // next row's first columns (and, on the last row, the heap past the
+  // tensor), and the bias read would run off the end of the bias vector.
   __device__ __forceinline__ void store_chunk(uint32_t m, uint32_t n0,
                                               V8 v) const {
-    if (m >= M || n0 >= N) return;  // N % 8 == 0: chunks never straddle
+    if (m >= M || n0 >= N) return;
+    const uint32_t rem = N - n0;
     if (bias || relu) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        float f = (float)v[e] + (bias ? bias[n0 + e] : 0.f);
+        float f = (float)v[e] + (bias && (uint32_t)e < rem ? bias[n0 + e] : 0.f);
         if (relu) f = fmaxf(f, 0.f);
         v[e] = (elem_t)f;
       }
     }
-    *(V8*)(c + (size_t)m * N + n0) = v;
+    if (rem >= 8) {
+      *(V8*)(c + (size_t)m * N + n0) = v;
+    } else {
+      for (uint32_t e = 0; e < rem; ++e) c[(size_t)m * N + n0 + e] = v[e];
+    }
   }
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t m0,

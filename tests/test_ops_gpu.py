@@ -459,3 +459,23 @@ def test_conv_wgrad_workspace_rezero():
     d1 = ext.conv_wgrad(x, dy, 3, 3, 1, 1).clone()
     d2 = ext.conv_wgrad(x, dy, 3, 3, 1, 1)
     torch.testing.assert_close(d1, d2, rtol=0.02, atol=0.05)
+
+
+def test_linear_unpadded_out_features_with_bias():
+    """N=100 (num_classes) regression: the epilogue's tail chunk used to write
+    a full 16B vector — stomping the next row's first columns, the heap past
+    the last row, and reading bias out of bounds. Churn the allocator with
+    garbage so OOB reads are visible."""
+    from dtmx.ops.hip import require_ext
+    ext = require_ext()
+    for (M, N, K) in [(4, 100, 2048), (8, 100, 2048), (16, 10, 504), (3, 37, 64)]:
+        g = torch.Generator().manual_seed(M * 1000 + N)
+        x = (torch.randn(M, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+        w = (torch.randn(N, K, generator=g) * 0.05).to(torch.bfloat16).cuda()
+        b = (torch.randn(N, generator=g)).float().cuda()
+        ref = x.float() @ w.float().T + b
+        for it in range(4):
+            junk = torch.full((8 << 20,), 3e4, dtype=torch.bfloat16, device="cuda")
+            del junk
+            y = ext.linear_fwd(x, w, b)
+            assert_close(y, ref.cpu(), name=f"linear M{M} N{N} K{K} it{it}")

@@ -38,14 +38,6 @@ for s in range(nsteps):
     rows.clear()
     mod.forward_backward(batch)
     print(f"--- step {s} loss={mod._loss.item():.3f} ---")
-    first = True
     for name, ty, shp, mx in rows:
-        flag = ""
-        if mx > 500 and first:
-            flag = "   <<< FIRST BIG"
-            first = False
-        if mx > 500 or flag:
-            print(f"  {name} {ty} {shp} max={mx:.4g}{flag}")
-    if first:
-        print("  all layer outputs < 500")
+        print(f"  {name}|{ty}|{mx:.6g}")
     mod.update()
