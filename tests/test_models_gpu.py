@@ -14,9 +14,7 @@ CASES = [
     ("resnet", {"num_layers": 101}, (2, 3, 224, 224), 0.01),
     ("alexnet", {}, (8, 3, 224, 224), 1e-4),
     ("vgg", {"num_layers": 16}, (4, 3, 224, 224), 1e-4),
-    # inception at lr 0.01/mom 0.9 from random init sits on the stability
-    # edge in 3 steps (split-K atomic reordering flips it run to run) — the
-    # zoo test checks kernel plumbing, so step it at a stable lr.
+    # modest lr: the zoo test checks kernel plumbing, not convergence tuning
     ("inception-v3", {}, (4, 3, 299, 299), 0.002),
     ("lenet", {}, (16, 1, 28, 28), 0.01),
     ("mlp", {}, (32, 784), 0.01),
