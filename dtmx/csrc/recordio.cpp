@@ -18,6 +18,7 @@
 #include <cstring>
 #include <deque>
 #include <fstream>
+#include <map>
 #include <mutex>
 #include <thread>
 #include <vector>
@@ -115,9 +116,9 @@ class RecordBatchLoader {
       std::shuffle(order_.begin(), order_.end(), rng);
     }
     cursor_ = 0;
-    produced_ = 0;
+    next_emit_ = 0;
     epoch_++;
-    queue_.clear();
+    ready_.clear();
     cv_space_.notify_all();
   }
 
@@ -125,31 +126,43 @@ class RecordBatchLoader {
 
   // returns (data[B,*shape] float32, label[B] float32) or empty tensors at
   // epoch end
+  // Batches are delivered IN ORDER (batch k of the epoch's shuffled order is
+  // the k-th item) regardless of worker completion order — matching the
+  // reference's ordered threadediter semantics (dmlc threadediter.h).
   std::vector<at::Tensor> next() {
     std::unique_lock<std::mutex> lk(mu_);
     cv_item_.wait(lk, [this] {
-      return stop_ || !queue_.empty() || produced_ >= batches_per_epoch();
+      return stop_ || ready_.count(next_emit_) ||
+             next_emit_ >= batches_per_epoch();
     });
-    if (queue_.empty()) return {};
-    auto out = std::move(queue_.front());
-    queue_.pop_front();
-    cv_space_.notify_one();
+    auto it = ready_.find(next_emit_);
+    if (it == ready_.end()) return {};
+    auto out = std::move(it->second);
+    ready_.erase(it);
+    next_emit_++;
+    cv_space_.notify_all();
     return out;
   }
 
  private:
   void worker() {
     while (!stop_) {
-      int64_t b;
+      int64_t b = -1, seq = 0;
       int64_t claim_epoch;
       {
-        std::lock_guard<std::mutex> lk(mu_);
+        std::unique_lock<std::mutex> lk(mu_);
+        // claim-gating bounds outstanding batches to the queue capacity so
+        // the in-order reorder buffer can never deadlock on a slow worker
+        cv_space_.wait(lk, [this] {
+          return stop_ || cursor_ + batch_ > (int64_t)order_.size() ||
+                 cursor_ / batch_ < next_emit_ + capacity_;
+        });
+        if (stop_) return;
         claim_epoch = epoch_;
-        if (cursor_ + batch_ > (int64_t)order_.size()) {
-          // wait for reset
-          b = -1;
-        } else {
+        if (cursor_ + batch_ <= (int64_t)order_.size() &&
+            cursor_ / batch_ < next_emit_ + capacity_) {
           b = cursor_;
+          seq = b / batch_;
           cursor_ += batch_;
         }
       }
@@ -172,13 +185,10 @@ class RecordBatchLoader {
         float* out = dp + i * elem_;
         for (int64_t e = 0; e < elem_; ++e) out[e] = raw[e] * (1.f / 255.f);
       }
-      std::unique_lock<std::mutex> lk(mu_);
-      cv_space_.wait(lk, [this] { return stop_ || (int64_t)queue_.size() < capacity_; });
-      if (stop_) return;
+      std::lock_guard<std::mutex> lk(mu_);
       if (epoch_ != claim_epoch) continue;  // reset() raced: drop stale batch
-      queue_.push_back({data, label});
-      produced_++;
-      cv_item_.notify_one();
+      ready_[seq] = {data, label};
+      cv_item_.notify_all();
     }
   }
 
@@ -190,10 +200,10 @@ class RecordBatchLoader {
   int64_t elem_ = 0;
   std::vector<size_t> order_;
   int64_t cursor_ = 0;
+  int64_t next_emit_ = 0;
   int64_t epoch_ = 0;
-  int64_t produced_ = 0;
   uint64_t rng_seed_ = 0;
-  std::deque<std::vector<at::Tensor>> queue_;
+  std::map<int64_t, std::vector<at::Tensor>> ready_;
   std::mutex mu_;
   std::condition_variable cv_item_, cv_space_;
   std::atomic<bool> stop_{false};
