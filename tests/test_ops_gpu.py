@@ -479,3 +479,18 @@ def test_linear_unpadded_out_features_with_bias():
             del junk
             y = ext.linear_fwd(x, w, b)
             assert_close(y, ref.cpu(), name=f"linear M{M} N{N} K{K} it{it}")
+
+
+def test_dropout_layernorm_fp16():
+    """fp16 instantiation of the counter-RNG dropout and LayerNorm kernels."""
+    from dtmx.ops import functional as DF
+    x = (torch.randn(32, 512) * 0.5).to(torch.float16).to(DEV).requires_grad_(True)
+    y = DF.dropout(x, 0.5, training=True, seed=7)
+    m = y != 0
+    assert abs(m.float().mean().item() - 0.5) < 0.05
+    assert_close(y[m], x.detach()[m] * 2.0, name="dropout fp16")
+    g = torch.ones(512, device=DEV)
+    b = torch.zeros(512, device=DEV)
+    z = DF.layer_norm(x.detach(), g, b)
+    ref = F.layer_norm(x.detach().float(), (512,), g, b, 1e-5)
+    assert_close(z, ref, name="ln fp16")
