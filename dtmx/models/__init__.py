@@ -3,7 +3,7 @@ network name -> builder, mirroring `import symbols.<net>; net.get_symbol()`
 (reference common/fit.py / train_imagenet.py)."""
 from __future__ import annotations
 
-from . import alexnet, inception_v3, lenet, mlp, resnet, vgg
+from . import alexnet, inception_bn, inception_v3, lenet, mlp, resnet, vgg
 
 _REGISTRY = {
     "resnet": resnet.get_symbol,
@@ -13,6 +13,7 @@ _REGISTRY = {
     "alexnet": alexnet.get_symbol,
     "vgg": vgg.get_symbol,
     "inception-v3": inception_v3.get_symbol,
+    "inception-bn": inception_bn.get_symbol,
     "inceptionv3": inception_v3.get_symbol,
 }
 

@@ -48,7 +48,7 @@ if __name__ == "__main__":
     ap.add_argument("--gpus", type=str, default="0" if torch.cuda.is_available() else None)
     ap.add_argument("--batch-sizes", type=str, default="1,32,128")
     ap.add_argument("--networks", type=str,
-                    default="alexnet,vgg,inception-v3,resnet,resnet-152")
+                    default="alexnet,vgg,inception-bn,inception-v3,resnet,resnet-152")
     ap.add_argument("--dtype", type=str, default="bfloat16")
     args = ap.parse_args()
     dev = dtmx.gpu(int(args.gpus.split(",")[0])) if args.gpus else dtmx.cpu()

@@ -16,6 +16,7 @@ CASES = [
     ("vgg", {"num_layers": 16}, (4, 3, 224, 224), 1e-4),
     # modest lr: the zoo test checks kernel plumbing, not convergence tuning
     ("inception-v3", {}, (4, 3, 299, 299), 0.002),
+    ("inception-bn", {}, (4, 3, 224, 224), 0.002),
     ("lenet", {}, (16, 1, 28, 28), 0.01),
     ("mlp", {}, (32, 784), 0.01),
 ]
