@@ -150,6 +150,10 @@ struct EpiBF16 {
   // when bias == nullptr && !relu (stats must equal the written values).
   float* bn_psum = nullptr;
   float* bn_psumsq = nullptr;
+  // optional elementwise accumulator: out = gemm + acc (residual-join grad
+  // fusion; acc may alias c for in-place accumulate — same thread reads and
+  // writes the same element, so aliasing is safe)
+  const elem_t* acc = nullptr;
   // coalesced row-chunk store used by the kernel's LDS-staged epilogue.
   // N need not be a multiple of 8 (e.g. FullyConnected num_classes=100):
   // the tail chunk is stored element-wise — a full V8 there would stomp the
@@ -159,6 +163,14 @@ struct EpiBF16 {
                                               V8 v) const {
     if (m >= M || n0 >= N) return;
     const uint32_t rem = N - n0;
+    if (acc && rem >= 8) {
+      V8 a = *(const V8*)(acc + (size_t)m * N + n0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v[e] = (elem_t)((float)v[e] + (float)a[e]);
+    } else if (acc) {
+      for (uint32_t e = 0; e < rem; ++e)
+        v[e] = (elem_t)((float)v[e] + (float)acc[(size_t)m * N + n0 + e]);
+    }
     if (bias || relu) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
@@ -230,6 +242,7 @@ struct EpiBF16Scatter {
   uint32_t M, N;  // M = NPQ, N = C
   uint32_t H, W, Q;
   int u, v;
+  int accumulate = 0;  // dx += val (uncovered pixels keep their value)
   FastDiv dQ, dPQ;
   __device__ __forceinline__ void store_chunk(uint32_t m, uint32_t n0,
                                               V8 val) const {
@@ -237,6 +250,11 @@ struct EpiBF16Scatter {
     uint32_t n = dPQ.div(m), pq = dPQ.mod(m, n);
     uint32_t p = dQ.div(pq), q = dQ.mod(pq, p);
     size_t off = (((size_t)n * H + p * u) * W + q * v) * N + n0;
+    if (accumulate) {
+      V8 a = *(const V8*)(dx + off);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) val[e] = (elem_t)((float)val[e] + (float)a[e]);
+    }
     *(V8*)(dx + off) = val;
   }
   template <int NJ>
