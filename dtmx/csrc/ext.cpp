@@ -7,7 +7,8 @@ at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&, const c10::optional<
 at::Tensor linear_dgrad(const at::Tensor&, const at::Tensor&);
 at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&);
 at::Tensor conv_fwd(const at::Tensor&, const at::Tensor&, long, long);
-at::Tensor conv_dgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
+at::Tensor conv_dgrad(const at::Tensor&, const at::Tensor&, long, long, long, long,
+                      const c10::optional<at::Tensor>&);
 at::Tensor conv_wgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
 std::vector<std::tuple<long, long, double>> wgrad_ws_stats();
 // batchnorm.hip
@@ -79,7 +80,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_wgrad", &dtmx::linear_wgrad);
   m.def("conv_fwd", &dtmx::conv_fwd);
   m.def("conv_fwd_stats", &dtmx::conv_fwd_stats);
-  m.def("conv_dgrad", &dtmx::conv_dgrad);
+  m.def("conv_dgrad", &dtmx::conv_dgrad, py::arg("dy"), py::arg("w"),
+        py::arg("stride"), py::arg("pad"), py::arg("H"), py::arg("W"),
+        py::arg("acc") = c10::nullopt);
   m.def("conv_wgrad", &dtmx::conv_wgrad);
   m.def("wgrad_ws_stats", &dtmx::wgrad_ws_stats);
   m.def("bn_fwd_train", &dtmx::bn_fwd_train);

@@ -1118,7 +1118,8 @@ std::vector<at::Tensor> conv_fwd_stats(const at::Tensor& x, const at::Tensor& w,
 // ------------------------------------------------------------- conv dgrad
 
 at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
-                      long pad, long H, long W_) {
+                      long pad, long H, long W_,
+                      const c10::optional<at::Tensor>& acc) {
   DTMX_DISPATCH_16(dy.scalar_type(), "conv_dgrad", {
     CHECK_BF16_CUDA(dy);
     TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy must be NHWC");
@@ -1151,8 +1152,20 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
     } else {
       wt = wtk.contiguous();
     }
-    auto dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
-                        at::MemoryFormat::ChannelsLast);
+    // residual-join fusion: dx = gemm + acc, accumulated IN PLACE into acc
+    // (one read replaces the separate read+read+write grad-sum at the fork)
+    const bool have_acc = acc.has_value() && acc->defined();
+    at::Tensor dx;
+    if (have_acc) {
+      TORCH_CHECK(acc->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                      acc->sizes() == at::IntArrayRef({(long)N, (long)C,
+                                                       (long)H, (long)W_}),
+                  "conv_dgrad: acc must be NHWC of the dx shape");
+      dx = *acc;
+    } else {
+      dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
+                     at::MemoryFormat::ChannelsLast);
+    }
     uint32_t M = N * H * W_, Ktot = R * S * Ko;
     if (R == 1 && S == 1 && pad == 0 && C % 8 == 0) {
       // 1x1 dgrad: dense dy @ W^T. Stride 1 writes rows directly; stride u>1
@@ -1164,14 +1177,16 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
       DenseP<elem_t> pbd{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy), C, Ko, Ko};
       if (stride == 1) {
         EpiBF16<elem_t> epid{(elem_t*)dx.data_ptr(), nullptr, Mn, C, 0};
-        if (!smallgrid_splitk(pad_, pbd, dx, Mn, C, Ko))
+        if (have_acc) epid.acc = (const elem_t*)dx.data_ptr();
+        if (have_acc || !smallgrid_splitk(pad_, pbd, dx, Mn, C, Ko))
           launch_gemm(pad_, pbd, epid, Mn, C, Ko);
       } else {
-        dx.zero_();
+        if (!have_acc) dx.zero_();
         EpiBF16Scatter<elem_t> epis;
         epis.dx = (elem_t*)dx.data_ptr();
         epis.M = Mn; epis.N = C; epis.H = H; epis.W = W_; epis.Q = Q;
         epis.u = stride; epis.v = stride;
+        epis.accumulate = have_acc ? 1 : 0;
         epis.dQ.init(Q); epis.dPQ.init(P * Q);
         launch_gemm(pad_, pbd, epis, Mn, C, Ko);
       }
@@ -1185,7 +1200,8 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
     pa.dW_.init(W_); pa.dHW.init(H * W_); pa.dKo.init(Ko); pa.dS.init(S);
     DenseP<elem_t> pb{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy), C, Ktot, Ktot};
     EpiBF16<elem_t> epi{(elem_t*)dx.data_ptr(), nullptr, M, C, 0};
-    if (!smallgrid_splitk(pa, pb, dx, M, C, Ktot))
+    if (have_acc) epi.acc = (const elem_t*)dx.data_ptr();
+    if (have_acc || !smallgrid_splitk(pa, pb, dx, M, C, Ktot))
       launch_gemm(pa, pb, epi, M, C, Ktot);
     return dx;
 

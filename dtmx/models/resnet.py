@@ -7,6 +7,8 @@ are bandwidth-bound, SURVEY.md §7 hard part 2).
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -18,6 +20,13 @@ from ..ops.layers import (
     LinearBF16,
     MaxPool2dNHWC,
 )
+
+
+def _use_fused_block(module, x):
+    # whole-block manual backward with residual-join grad fusion
+    # (ops/fusedblock.py); DTMX_FUSED_BLOCK=0 restores the layer-by-layer path
+    return (module.training and x.is_cuda
+            and os.environ.get("DTMX_FUSED_BLOCK", "1") == "1")
 
 
 class BasicBlock(nn.Module):
@@ -36,6 +45,9 @@ class BasicBlock(nn.Module):
             )
 
     def forward(self, x):
+        if _use_fused_block(self, x):
+            from ..ops.fusedblock import fused_basic_block
+            return fused_basic_block(x, self)
         sc = x if self.downsample is None else self.downsample(x)
         y = self.bn1(self.conv1(x))
         return self.bn2(self.conv2(y), sc)
@@ -60,6 +72,9 @@ class Bottleneck(nn.Module):
             )
 
     def forward(self, x):
+        if _use_fused_block(self, x):
+            from ..ops.fusedblock import fused_bottleneck
+            return fused_bottleneck(x, self)
         sc = x if self.downsample is None else self.downsample(x)
         y = self.bn1(self.conv1(x))
         y = self.bn2(self.conv2(y))

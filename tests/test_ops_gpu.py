@@ -494,3 +494,53 @@ def test_dropout_layernorm_fp16():
     z = DF.layer_norm(x.detach(), g, b)
     ref = F.layer_norm(x.detach().float(), (512,), g, b, 1e-5)
     assert_close(z, ref, name="ln fp16")
+
+
+def test_fused_block_matches_layerwise():
+    """The manually-backpropagated residual block (DTMX_FUSED_BLOCK) must
+    reproduce the layer-by-layer autograd path: same y, same input/param
+    grads, same running stats — it runs the same kernels in the same order,
+    with only the fork-add fused into the dgrad epilogue."""
+    import copy
+    import os
+    from dtmx.models.resnet import BasicBlock, Bottleneck
+
+    cases = [
+        (Bottleneck, dict(in_ch=256, ch=64)),                 # identity
+        (Bottleneck, dict(in_ch=256, ch=128, stride=2)),      # downsample s2
+        (Bottleneck, dict(in_ch=64, ch=64)),                  # downsample s1
+        (BasicBlock, dict(in_ch=64, ch=64)),
+        (BasicBlock, dict(in_ch=64, ch=128, stride=2)),
+    ]
+    for cls, kw in cases:
+        torch.manual_seed(0)
+        blk = cls(**kw).to(DEV).to(torch.bfloat16)
+        blk2 = copy.deepcopy(blk)
+        x = nhwc(mk((4, kw["in_ch"], 14, 14), seed=5))
+        dy_shape = None
+        results = []
+        for flag, b in (("0", blk), ("1", blk2)):
+            os.environ["DTMX_FUSED_BLOCK"] = flag
+            b.train()
+            xr = x.clone().requires_grad_(True)
+            y = b(xr)
+            if dy_shape is None:
+                dy_shape = y.shape
+            dy = nhwc(mk(tuple(dy_shape), seed=6))
+            y.backward(dy)
+            results.append({
+                "y": y.detach(), "dx": xr.grad,
+                "pg": {n: p.grad.clone() for n, p in b.named_parameters()},
+                "rm": b.bn1.running_mean.clone(),
+            })
+        os.environ.pop("DTMX_FUSED_BLOCK", None)
+        ref, fused = results
+        tag = f"{cls.__name__}{kw}"
+        # small-grid conv paths use fp32 atomics (order nondeterministic),
+        # so compare numerically, not bit-wise
+        assert_close(fused["y"], ref["y"], name=f"{tag} y")
+        torch.testing.assert_close(fused["rm"], ref["rm"], rtol=1e-3, atol=1e-4)
+        assert_close(fused["dx"], ref["dx"], rtol=0.02, name=f"{tag} dx")
+        for n, g in ref["pg"].items():
+            assert_close(fused["pg"][n], g, rtol=0.05, atol=0.05,
+                         name=f"{tag} grad {n}")
