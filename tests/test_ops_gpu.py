@@ -515,6 +515,7 @@ def test_fused_block_matches_layerwise():
     for cls, kw in cases:
         torch.manual_seed(0)
         blk = cls(**kw).to(DEV).to(torch.bfloat16)
+        blk = blk.to(memory_format=torch.channels_last)  # as Module.bind does
         blk2 = copy.deepcopy(blk)
         x = nhwc(mk((4, kw["in_ch"], 14, 14), seed=5))
         dy_shape = None
