@@ -74,12 +74,12 @@ def main():
     it = SyntheticDataIter(1000, data_shape, max_iter=10 ** 9, dtype=dtype,
                            device=device, layout="NHWC")
 
-    # hipGraph step replay is the single-GPU default (measured +5% at bs1024:
-    # launch overhead and inter-kernel gaps vanish); DTMX_HIPGRAPH=0 forces
-    # eager. Multi-rank stays eager until RCCL capture is validated at scale.
+    # hipGraph step replay measures ~5% faster at bs1024 but fails the
+    # training-trajectory validation (tools/graph_numerics.py) — it stays
+    # opt-in/experimental until the captured step is proven equivalent.
     use_graph = (
         device.type == "cuda"
-        and os.environ.get("DTMX_HIPGRAPH", "1") == "1"
+        and os.environ.get("DTMX_HIPGRAPH", "0") == "1"
         and not dist_mode
     )
 

@@ -1,3 +1,5 @@
+# KNOWN ISSUE: the graph trajectory currently DIVERGES (loss rises, then
+# NaN) — hipGraph step capture is opt-in (DTMX_HIPGRAPH=1) until fixed.
 import sys, torch
 sys.path.insert(0, ".")
 import dtmx
