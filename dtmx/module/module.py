@@ -418,19 +418,31 @@ class Module:
             labels = [self._to_device(l, False) for l in (data_batch.label or [])]
             self._static_inputs = (data, labels)
             static_batch = DataBatch(data=data, label=labels)
-            # warmup on a side stream (allocator + autograd graph priming)
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(3):
+            # capture the LAYERWISE block path: replay of it is validated
+            # (tools/graph_numerics.py); the fused-block manual backward is
+            # not yet capture-safe and diverges under replay.
+            import os
+            prev_fb = os.environ.get("DTMX_FUSED_BLOCK")
+            os.environ["DTMX_FUSED_BLOCK"] = "0"
+            try:
+                # warmup on a side stream (allocator + autograd graph priming)
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):
+                        self.forward_backward(static_batch)
+                        self.update()
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
                     self.forward_backward(static_batch)
                     self.update()
-            torch.cuda.current_stream().wait_stream(side)
-            torch.cuda.synchronize()
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self.forward_backward(static_batch)
-                self.update()
+            finally:
+                if prev_fb is None:
+                    os.environ.pop("DTMX_FUSED_BLOCK", None)
+                else:
+                    os.environ["DTMX_FUSED_BLOCK"] = prev_fb
             self._graph = g
             return
         data, labels = self._static_inputs
