@@ -74,9 +74,9 @@ def main():
     it = SyntheticDataIter(1000, data_shape, max_iter=10 ** 9, dtype=dtype,
                            device=device, layout="NHWC")
 
-    # hipGraph step replay measures ~5% faster at bs1024 but fails the
-    # training-trajectory validation (tools/graph_numerics.py) — it stays
-    # opt-in/experimental until the captured step is proven equivalent.
+    # hipGraph step replay (opt-in): captures the layerwise block path —
+    # trajectory-validated (tools/graph_numerics.py) — but the eager default
+    # uses the fused-block backward, which is faster overall.
     use_graph = (
         device.type == "cuda"
         and os.environ.get("DTMX_HIPGRAPH", "0") == "1"
