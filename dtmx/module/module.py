@@ -522,6 +522,20 @@ class Module:
             train_data = self.get_iterator(kv)
         assert train_data is not None, "no train_data and no data_iterator factory"
 
+        # A joiner's EPOCH_BEGIN env is a stale hint (written at an earlier
+        # barrier). Its rendezvous completed at the survivors' barrier, which
+        # wrote the authoritative cluster epoch — adopt it so the local epoch
+        # counter aligns and every worker terminates at the same epoch
+        # (otherwise the joiner trails and strands at a barrier after the
+        # survivors exit).
+        if (is_new_worker and isinstance(kv, DistKVStore)
+                and getattr(kv, "_elastic", None) is not None
+                and getattr(kv._elastic, "store", None) is not None):
+            try:
+                begin_epoch = int(kv._elastic.store.get("cluster/epoch"))
+            except Exception:
+                pass
+
         eval_metric = metric_mod.create(eval_metric)
         if validation_metric is None:
             validation_metric = eval_metric

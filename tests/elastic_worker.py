@@ -16,6 +16,8 @@ from dtmx.models import get_symbol
 
 def main():
     out_path = os.environ["ELASTIC_TEST_OUT"]
+    if os.environ.get("ELASTIC_TEST_OUT_PER_WID") == "1":
+        out_path += "." + os.environ.get("DMLC_WORKER_ID", "w").replace("#", "_")
     epoch_sleep = float(os.environ.get("EPOCH_SLEEP", "0.25"))
     num_epoch = int(os.environ.get("NUM_EPOCH", "8"))
     torch.manual_seed(0)
