@@ -36,6 +36,11 @@ def main():
     from dtmx.parallel.bucketer import GradBucketer
 
     backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if "RANK" not in os.environ:  # standalone single-process run
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29561")
     dist.init_process_group(backend=backend)
     rank = dist.get_rank()
     world = dist.get_world_size()
