@@ -20,7 +20,7 @@ if _os.environ.get("DTMX_BLOCKING", "0") == "1":
     _os.environ.setdefault("HIP_LAUNCH_BLOCKING", "1")
 
 from . import callback, context, initializer, io, lr_scheduler, metric, model, monitor
-from . import gluon, kvstore, models, ndarray, optimizer, parallel, profiler
+from . import gluon, kvstore, models, ndarray, optimizer, parallel, profiler, recordio
 from . import module as mod
 from .context import Context, cpu, gpu, num_gpus
 from .module import Module

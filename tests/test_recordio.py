@@ -62,3 +62,50 @@ def test_sharding(tmp_path):
     l1 = [b.label[0][0].item() for b in it1]
     assert len(l0) == 4 and len(l1) == 4
     assert l0[0] == float(y[0]) and l1[0] == float(y[32])
+
+
+def test_recordio_python_api(tmp_path):
+    """mx.recordio surface: pack/unpack + MXRecordIO + MXIndexedRecordIO
+    (reference python/mxnet/recordio.py)."""
+    import numpy as np
+    from dtmx import recordio
+
+    # scalar and array labels round-trip through pack/unpack
+    h = recordio.IRHeader(0, 7.0, 3, 0)
+    hdr, payload = recordio.unpack(recordio.pack(h, b"abc"))
+    assert hdr.label == 7.0 and payload == b"abc"
+    ha = recordio.IRHeader(0, [1.0, 2.0, 3.0], 5, 0)
+    hdr, payload = recordio.unpack(recordio.pack(ha, b"xyz"))
+    assert hdr.flag == 3 and np.allclose(hdr.label, [1, 2, 3]) and payload == b"xyz"
+
+    # sequential file
+    p = str(tmp_path / "seq.rec")
+    with recordio.MXRecordIO(p, "w") as w:
+        for i in range(5):
+            w.write(recordio.pack(recordio.IRHeader(0, float(i), i, 0),
+                                  bytes([i]) * 4))
+    r = recordio.MXRecordIO(p, "r")
+    labels = []
+    while True:
+        rec = r.read()
+        if rec is None:
+            break
+        hdr, payload = recordio.unpack(rec)
+        labels.append(hdr.label)
+        assert payload == bytes([int(hdr.label)]) * 4
+    assert labels == [0.0, 1.0, 2.0, 3.0, 4.0]
+    r.reset()
+    assert recordio.unpack(r.read())[0].label == 0.0
+
+    # indexed file
+    pi = str(tmp_path / "idx.rec")
+    ix = str(tmp_path / "idx.idx")
+    w = recordio.MXIndexedRecordIO(ix, pi, "w")
+    for i in range(4):
+        w.write_idx(10 + i, recordio.pack(recordio.IRHeader(0, float(i), i, 0),
+                                          b"p%d" % i))
+    w.close()
+    r = recordio.MXIndexedRecordIO(ix, pi, "r")
+    assert sorted(r.keys()) == [10, 11, 12, 13]
+    hdr, payload = recordio.unpack(r.read_idx(12))
+    assert hdr.label == 2.0 and payload == b"p2"
