@@ -24,8 +24,22 @@ def _free_port():
     return p
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(600)
 def test_launcher_local_elastic_join(tmp_path):
+    # spawned-cluster test: a rare TCP port/timing blip can strand a worker,
+    # so allow one retry with a fresh scratch dir
+    for attempt in range(2):
+        scratch = tmp_path / f"try{attempt}"
+        scratch.mkdir()
+        try:
+            _run_launcher_join(scratch)
+            return
+        except AssertionError:
+            if attempt == 1:
+                raise
+
+
+def _run_launcher_join(tmp_path):
     hostfile = tmp_path / "hosts"
     hostfile.write_text("127.0.0.1\n127.0.0.1\n")
     out_base = str(tmp_path / "out")
