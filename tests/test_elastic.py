@@ -145,3 +145,28 @@ def test_removal_beats_addition(tmp_path):
     assert "d#0" not in gens[0] and "e#0" not in gens[0]
     assert "e#0" in gens[1]
     assert "c#0" in sched.members  # initial worker retained
+
+
+def test_heartbeats_and_dead_node_detection(monkeypatch):
+    """Store-based heartbeats + num_dead_node (reference van.cc:686-698
+    heartbeat ledger / Postoffice dead-node accounting)."""
+    from dtmx.parallel import rendezvous as rz
+
+    monkeypatch.setattr(rz, "_HB_INTERVAL", 0.05)
+    monkeypatch.setattr(rz, "_HB_TIMEOUT", 0.5)
+    port = _free_port()
+    sched = Scheduler("127.0.0.1", port, ["a", "b"], hostfile=None)
+    ctx = rz.ElasticContext.__new__(rz.ElasticContext)
+    ctx.worker_id = "a"
+    ctx.members = ["a", "b"]
+    ctx.store = sched.store
+    ctx._hb_stop = None
+    ctx._start_heartbeat()
+    try:
+        ctx.store.set("hb/b", str(time.time()))  # peer alive
+        time.sleep(0.3)  # let our own beat land
+        assert ctx.num_dead_node() == 0
+        ctx.store.set("hb/b", str(time.time() - 100))  # peer went silent
+        assert ctx.num_dead_node() == 1
+    finally:
+        ctx._stop_heartbeat()
