@@ -202,3 +202,79 @@ def test_row_sparse_pull():
     kv.row_sparse_pull("emb", out=out, row_ids=torch.tensor([1, 3]))
     assert out[1, 0] == 4 and out[3, 3] == 15
     assert out[0].sum() == 0 and out[2].sum() == 0
+
+
+def test_speedometer_and_checkpoint_callbacks(tmp_path, caplog):
+    import logging
+
+    import dtmx
+    from dtmx.callback import BatchEndParam, Speedometer, do_checkpoint
+    from dtmx.io import DataBatch
+    from dtmx.models import get_symbol
+
+    net = get_symbol("mlp", num_classes=10, input_dim=16)
+    mod = dtmx.Module(net, context=dtmx.cpu())
+    mod.bind(data_shapes=[("data", (4, 16))], label_shapes=[("softmax_label", (4,))])
+    mod.init_params()
+    mod.init_optimizer()
+    batch = DataBatch(data=[torch.randn(4, 16)], label=[torch.randint(0, 10, (4,)).float()])
+    mod.forward_backward(batch)
+    mod.update()
+
+    speed = Speedometer(batch_size=4, frequent=2)
+    with caplog.at_level(logging.INFO):
+        for nb in range(5):
+            speed(BatchEndParam(epoch=0, nbatch=nb, eval_metric=None, locals=None))
+    assert any("samples/sec" in r.message for r in caplog.records)
+
+    cb = do_checkpoint(str(tmp_path / "ck"), period=1)
+    cb(0, mod.symbol, *mod.get_params())
+    assert (tmp_path / "ck-0001.params").exists()
+    _sym, arg, aux = dtmx.model.load_checkpoint(str(tmp_path / "ck"), 1)
+    assert set(arg) == {n for n, _ in net.named_parameters()}
+
+
+def test_monitor_collects_stats():
+    import dtmx
+    from dtmx.io import DataBatch
+    from dtmx.models import get_symbol
+    from dtmx.monitor import Monitor
+
+    net = get_symbol("mlp", num_classes=10, input_dim=16)
+    mod = dtmx.Module(net, context=dtmx.cpu())
+    mod.bind(data_shapes=[("data", (4, 16))], label_shapes=[("softmax_label", (4,))])
+    mod.init_params()
+    mod.init_optimizer()
+    mon = Monitor(interval=1)
+    mon.install(mod)
+    batch = DataBatch(data=[torch.randn(4, 16)], label=[torch.randint(0, 10, (4,)).float()])
+    mon.tic()
+    mod.forward_backward(batch)
+    rows = mon.toc()
+    assert rows and all(len(r) == 3 for r in rows)
+    assert any(name.endswith("_grad") for _, name, _ in rows)
+
+
+def test_profiler_chrome_trace(tmp_path):
+    from dtmx import profiler
+
+    out = str(tmp_path / "trace.json")
+    profiler.set_config(filename=out)
+    profiler.set_state("run")
+    torch.randn(32, 32) @ torch.randn(32, 32)
+    profiler.set_state("stop")
+    assert os.path.exists(out) and os.path.getsize(out) > 100
+
+
+def test_gluon_trainer_step():
+    from dtmx import gluon
+
+    net = torch.nn.Linear(8, 4)
+    trainer = gluon.Trainer(net.parameters(), "sgd",
+                            {"learning_rate": 0.5}, kvstore="local")
+    x = torch.randn(16, 8)
+    w0 = net.weight.detach().clone()
+    loss = gluon.L2Loss()(net(x), torch.zeros(16, 4)).mean()
+    loss.backward()
+    trainer.step(batch_size=16)
+    assert not torch.allclose(net.weight.detach(), w0)
