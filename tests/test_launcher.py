@@ -34,7 +34,7 @@ def test_launcher_local_elastic_join(tmp_path):
         try:
             _run_launcher_join(scratch)
             return
-        except AssertionError:
+        except (AssertionError, subprocess.TimeoutExpired):
             if attempt == 1:
                 raise
 
@@ -59,14 +59,22 @@ def _run_launcher_join(tmp_path):
          "--elastic-training-enabled", "True",
          "--scheduler-port", str(_free_port()), "--poll-seconds", "0.3",
          "--", sys.executable, WORKER],
-        env=env, cwd=ROOT, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+        env=env, cwd=ROOT, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+        start_new_session=True)
     try:
         time.sleep(2.5)  # let the initial pair get into the epoch loop
         hostfile.write_text("127.0.0.1\n127.0.0.1\n127.0.0.1\n")  # join
-        rc = p.wait(timeout=240)
+        rc = p.wait(timeout=200)
     finally:
         if p.poll() is None:
-            p.kill()
+            # kill the whole session (launcher + its workers) so a timed-out
+            # attempt cannot leave orphans that disturb the retry
+            import signal
+            try:
+                os.killpg(os.getpgid(p.pid), signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+            p.wait(timeout=30)
     assert rc == 0, p.stderr.read().decode()[-2000:]
 
     outs = sorted(tmp_path.glob("out.*"))
