@@ -59,7 +59,9 @@ def _run_launcher_join(tmp_path):
          "--elastic-training-enabled", "True",
          "--scheduler-port", str(_free_port()), "--poll-seconds", "0.3",
          "--", sys.executable, WORKER],
-        env=env, cwd=ROOT, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+        env=env, cwd=ROOT,
+        stdout=open(tmp_path / "launcher.out", "w"),
+        stderr=open(tmp_path / "launcher.err", "w"),
         start_new_session=True)
     try:
         time.sleep(2.5)  # let the initial pair get into the epoch loop
@@ -75,7 +77,7 @@ def _run_launcher_join(tmp_path):
             except ProcessLookupError:
                 pass
             p.wait(timeout=30)
-    assert rc == 0, p.stderr.read().decode()[-2000:]
+    assert rc == 0, (tmp_path / "launcher.err").read_text()[-2000:]
 
     outs = sorted(tmp_path.glob("out.*"))
     assert len(outs) == 3, [o.name for o in outs]
