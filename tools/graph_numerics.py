@@ -1,3 +1,7 @@
+"""hipGraph step-capture trajectory validation: eager vs graphed_step on
+ResNet-18. Both must descend identically (modulo the capture call, which
+warms up and records without replaying). Run on an MI355X box.
+"""
 import sys, torch
 sys.path.insert(0, ".")
 import dtmx
