@@ -81,10 +81,18 @@ class Launcher:
         )
         if new_worker:
             env["NEW_WORKER"] = "1"
+            # EPOCH_BEGIN is only a hint (the joiner adopts the authoritative
+            # cluster epoch at its first barrier) — bound the fetch: a blocking
+            # store.get would stall 300 s when the cluster has not reached its
+            # first epoch barrier yet.
+            from datetime import timedelta
             try:
+                self.scheduler.store.set_timeout(timedelta(seconds=2))
                 env["EPOCH_BEGIN"] = self.scheduler.store.get("cluster/env/EPOCH_BEGIN").decode()
             except Exception:
                 env["EPOCH_BEGIN"] = "0"
+            finally:
+                self.scheduler.store.set_timeout(timedelta(seconds=300))
         return env
 
     def launch_worker(self, wid: str, new_worker: bool = False):
