@@ -278,3 +278,26 @@ def test_gluon_trainer_step():
     loss.backward()
     trainer.step(batch_size=16)
     assert not torch.allclose(net.weight.detach(), w0)
+
+
+def test_prefetching_iter():
+    from dtmx.io import PrefetchingIter
+
+    base = SyntheticDataIter(10, (4, 8), max_iter=6)
+    it = PrefetchingIter(base)
+    b1 = list(it)
+    assert len(b1) == 6
+    it.reset()
+    b2 = list(it)
+    assert len(b2) == 6
+    assert it.provide_data == base.provide_data
+
+
+def test_mnist_iter_synthetic_fallback(tmp_path):
+    from dtmx.io import MNISTIter
+
+    it = MNISTIter(image=str(tmp_path / "none"), label=str(tmp_path / "nol"),
+                   batch_size=32)
+    b = it.next()
+    assert b.data[0].shape == (32, 1, 28, 28)
+    assert b.label[0].shape == (32,)
