@@ -9,6 +9,7 @@ import sys
 import time
 
 import pytest
+import torch.distributed as tdist
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 WORKER = os.path.join(ROOT, "tests", "elastic_worker.py")
@@ -24,7 +25,7 @@ def _free_port():
     return p
 
 
-@pytest.mark.timeout(600)
+@pytest.mark.timeout(900)
 def test_launcher_local_elastic_join(tmp_path):
     # spawned-cluster test: a rare TCP port/timing blip can strand a worker,
     # so allow one retry with a fresh scratch dir
@@ -54,19 +55,37 @@ def _run_launcher_join(tmp_path):
     )
     for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
         env.pop(k, None)
+    port = _free_port()
     p = subprocess.Popen(
         [sys.executable, LAUNCH, "-n", "2", "-H", str(hostfile),
          "--elastic-training-enabled", "True",
-         "--scheduler-port", str(_free_port()), "--poll-seconds", "0.3",
+         "--scheduler-port", str(port), "--poll-seconds", "0.3",
          "--", sys.executable, WORKER],
         env=env, cwd=ROOT,
         stdout=open(tmp_path / "launcher.out", "w"),
         stderr=open(tmp_path / "launcher.err", "w"),
         start_new_session=True)
     try:
-        time.sleep(2.5)  # let the initial pair get into the epoch loop
+        # event-driven join point: wait for the cluster's first epoch barrier
+        # (cluster/epoch in the scheduler store) instead of a fixed sleep —
+        # shared-host load makes absolute timings unreliable
+        from datetime import timedelta
+        store = tdist.TCPStore("127.0.0.1", port, is_master=False,
+                               timeout=timedelta(seconds=120))
+        store.set_timeout(timedelta(seconds=2))
+        deadline = time.time() + 150
+        while time.time() < deadline:
+            try:
+                if int(store.get("cluster/epoch")) >= 1:
+                    break
+            except Exception:
+                pass
+            time.sleep(0.3)
+        else:
+            raise AssertionError("cluster never reached epoch 1")
+        del store
         hostfile.write_text("127.0.0.1\n127.0.0.1\n127.0.0.1\n")  # join
-        rc = p.wait(timeout=200)
+        rc = p.wait(timeout=300)
     finally:
         if p.poll() is None:
             # kill the whole session (launcher + its workers) so a timed-out
