@@ -312,20 +312,42 @@ class PrefetchingIter(DataIter):
         self.data_iter = data_iter
         self._queue: "queue.Queue" = queue.Queue(maxsize=capacity)
         self._stop = threading.Event()
-        self._reset_req = threading.Event()
+        self._wake = threading.Event()
+        # generation counter: reset() bumps it; the worker resets the inner
+        # iterator when it observes the bump, and every queued batch carries
+        # the generation it was produced under so the consumer can discard
+        # pre-reset leftovers. This makes reset() correct even when the
+        # consumer abandons an epoch mid-way (score(num_batch=...),
+        # ResizeIter caps) — the old design only handled reset-at-
+        # StopIteration and could deliver stale batches otherwise.
+        self._gen = 0
         self._thread = threading.Thread(target=self._worker, daemon=True)
         self._thread.start()
 
     def _worker(self):
+        import queue
+
+        worker_gen = 0
         while not self._stop.is_set():
+            if worker_gen != self._gen:
+                worker_gen = self._gen
+                self.data_iter.reset()
+                continue
             try:
                 batch = self.data_iter.next()
-                self._queue.put(batch)
             except StopIteration:
-                self._queue.put(None)
-                self._reset_req.wait()
-                self._reset_req.clear()
-                self.data_iter.reset()
+                batch = None
+            # put, abandoning the item if a reset or stop arrives while full
+            while not self._stop.is_set() and worker_gen == self._gen:
+                try:
+                    self._queue.put((worker_gen, batch), timeout=0.05)
+                    break
+                except queue.Full:
+                    continue
+            if batch is None and worker_gen == self._gen:
+                # parked at epoch end until reset() or close()
+                self._wake.wait()
+                self._wake.clear()
 
     @property
     def provide_data(self):
@@ -336,16 +358,30 @@ class PrefetchingIter(DataIter):
         return self.data_iter.provide_label
 
     def reset(self):
-        # drain
-        while not self._queue.empty():
-            self._queue.get_nowait()
-        self._reset_req.set()
+        import queue
+
+        self._gen += 1
+        self._wake.set()
+        # drain stale entries so a blocked worker can make progress; any
+        # stale item that races in after this drain is filtered in next()
+        while True:
+            try:
+                self._queue.get_nowait()
+            except queue.Empty:
+                break
 
     def next(self):
-        batch = self._queue.get()
-        if batch is None:
-            raise StopIteration
-        return batch
+        while True:
+            gen, batch = self._queue.get()
+            if gen != self._gen:
+                continue  # produced before the last reset()
+            if batch is None:
+                raise StopIteration
+            return batch
+
+    def close(self):
+        self._stop.set()
+        self._wake.set()
 
 
 class ImageRecordIter(DataIter):

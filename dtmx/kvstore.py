@@ -304,7 +304,7 @@ class DistKVStore(KVStore):
             else:
                 merged = v.detach().clone()
             if self._compression is not None and not self._is_aux_key(ikey):
-                merged = self._compression.compress_decompress(merged)
+                merged = self._compression.compress_decompress(merged, key=ikey)
             dist.all_reduce(merged, op=dist.ReduceOp.SUM)
             stored = self._store[ikey]
             if self._is_aux_key(ikey):

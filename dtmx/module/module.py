@@ -154,6 +154,14 @@ class Module:
         if self.params_initialized and not force_init:
             return
         assert self.binded, "call bind before init_params"
+        # Module.load stashed checkpoint params — consume them here so
+        # load() -> bind() -> init_params()/fit() trains from the checkpoint,
+        # not fresh init (reference Module.load, module.py:131-190).
+        preloaded = getattr(self, "_preloaded", None)
+        if preloaded is not None and arg_params is None and aux_params is None:
+            arg_params, aux_params = preloaded
+            initializer = None
+            self._preloaded = None
         args, auxs = self._classified_named_tensors()
         if initializer is None and (arg_params is None and aux_params is None):
             initializer = init_mod.create("default")
@@ -453,11 +461,13 @@ class Module:
             for dst, src in zip(labels, data_batch.label or []):
                 if dst.data_ptr() != src.data_ptr():
                     dst.copy_(self._to_device(src, False), non_blocking=True)
+        # increment num_update BEFORE deriving lr, matching the eager
+        # update() path — otherwise replayed steps run one lr-update stale.
+        self._optimizer.num_update += 1
         self._update_hyper(
             self._optimizer.lr_scheduler(self._optimizer.num_update)
             if self._optimizer.lr_scheduler else self._optimizer.lr
         )
-        self._optimizer.num_update += 1
         self._graph.replay()
 
     def store_aux_params(self):

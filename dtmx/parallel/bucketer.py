@@ -220,7 +220,9 @@ class GradBucketer:
 
         ext = require_ext()
         world = dist.get_world_size()
-        flat = self.flat[bi]
+        # async mode accumulates into the shadow set — reduce THAT buffer,
+        # not the in-flight previous-step one (self.flat[bi] would be stale).
+        flat = self._cur_flats()[bi] if self.async_mode else self.flat[bi]
         packed = ext.quantize_2bit(flat.view(-1), self._gc_residual[bi].view(-1),
                                    self._gc_threshold)
         gathered = [torch.empty_like(packed) for _ in range(world)]
@@ -228,6 +230,10 @@ class GradBucketer:
         out = ext.dequantize_2bit(gathered[0], flat.numel(), self._gc_threshold)
         for g in gathered[1:]:
             out += ext.dequantize_2bit(g, flat.numel(), self._gc_threshold)
+        if self.average:
+            # finish() only averages buckets with outstanding works; the
+            # compressed path completes synchronously, so average here.
+            out /= world
         flat.view(-1).copy_(out)
 
     def _on_grad_ready_compressed(self, param):

@@ -29,9 +29,15 @@ class TwoBitCompression:
         self.threshold = float(threshold)
         self._residual = {}
 
-    def compress_decompress(self, grad: torch.Tensor) -> torch.Tensor:
-        """Quantize+dequantize with persistent per-tensor residual."""
-        key = (grad.shape, grad.device, grad.dtype)
+    def compress_decompress(self, grad: torch.Tensor, key=None) -> torch.Tensor:
+        """Quantize+dequantize with persistent per-tensor residual.
+
+        `key` should be the kvstore key: the reference keeps residual_[key]
+        per key (kvstore_dist.h:778). Falling back to tensor metadata would
+        alias the residuals of same-shaped parameters, corrupting error
+        feedback — callers that push multiple tensors MUST pass the key."""
+        if key is None:
+            key = (grad.shape, grad.device, grad.dtype)
         res = self._residual.get(key)
         if res is None or res.shape != grad.shape:
             res = torch.zeros_like(grad, dtype=torch.float32)

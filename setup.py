@@ -16,10 +16,14 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "dtmx", "csrc")
 
+# Exclude hipify-generated "*_hip.hip" twins: torch's build hipifies each
+# .hip source into a sibling <name>_hip.hip; globbing those back in would
+# feed every kernel through the compiler twice (duplicate symbols / stale
+# binaries on incremental builds).
 sources = [
     os.path.join("dtmx", "csrc", f)
     for f in sorted(os.listdir(CSRC))
-    if f.endswith((".cpp", ".hip"))
+    if f.endswith((".cpp", ".hip")) and not f.endswith("_hip.hip")
 ]
 
 setup(
