@@ -47,6 +47,11 @@ def main():
 
     torch.manual_seed(1234)
     if torch.cuda.is_available():
+        # clamp in case the launcher passes more ranks than devices (RCCL
+        # itself refuses duplicate devices in one communicator, so world>1
+        # still requires distinct GPUs; this just avoids an invalid
+        # set_device before RCCL reports the real error)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
         ctx = dtmx.gpu(local_rank)

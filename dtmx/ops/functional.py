@@ -32,7 +32,17 @@ _FALLBACK = set(os.environ.get("DTMX_FALLBACK", "").split(",")) - {""}
 def _use_hip(*tensors, op=None) -> bool:
     if op is not None and op in _FALLBACK:  # debug: force torch fallback
         return False
-    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+    ts = [t for t in tensors if isinstance(t, torch.Tensor)]
+    if not any(t.is_cuda for t in ts):
+        return False
+    # The MFMA/LDS kernels implement the training dtypes (bf16/fp16) only;
+    # fp32 work routes to torch's rocBLAS/MIOpen-free ops instead of
+    # erroring (reference supports fp32 paths, e.g. FullyConnected fp32).
+    return all(
+        t.dtype in (torch.bfloat16, torch.float16)
+        for t in ts
+        if t.is_floating_point()
+    )
 
 
 # --------------------------------------------------------------------- conv

@@ -265,7 +265,11 @@ class ElasticContext:
         version, members = obj[0]
         if members is None or version == self.version:
             return False
-        # re-form
+        # re-form; quiesce in-flight RCCL work first — destroying the NCCL
+        # communicator with collectives still on-stream is undefined
+        # (reference analog: engine WaitForAll before group teardown)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
         dist.destroy_process_group()
         if self.worker_id not in members:
             log.info("worker %s removed from roster at v%d; exiting", self.worker_id, version)

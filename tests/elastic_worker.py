@@ -23,8 +23,13 @@ def main():
     torch.manual_seed(0)
     np.random.seed(0)
 
+    # ELASTIC_TEST_DEVICE=cuda: run the same scenario on GPU with the RCCL
+    # backend (several ranks sharing one device — RCCL permits this), the
+    # hardware proof of the elastic comm destroy/re-form path.
+    use_cuda = os.environ.get("ELASTIC_TEST_DEVICE") == "cuda"
+    ctx = dtmx.gpu(0) if use_cuda else dtmx.cpu()
     net = get_symbol("mlp", num_classes=10, input_dim=32)
-    mod = dtmx.Module(net, context=dtmx.cpu())
+    mod = dtmx.Module(net, context=ctx)
     mod.bind(data_shapes=[("data", (8, 32))], label_shapes=[("softmax_label", (8,))])
     kv = dtmx.kvstore.create("dist_sync")
 
@@ -51,12 +56,17 @@ def main():
     )
     arg, aux = mod.get_params()
     param_sum = float(sum(p.double().sum().item() for p in arg.values()))
+    import torch.distributed as tdist
+
     result = {
         "wid": os.environ.get("DMLC_WORKER_ID"),
         "worker_counts": counts,
         "param_sum": param_sum,
         "final_workers": kv.num_workers,
         "final_rank": kv.rank,
+        "backend": tdist.get_backend() if tdist.is_initialized() else None,
+        "device": "cuda" if use_cuda else "cpu",
+        "generation": getattr(kv._elastic, "version", 0),
     }
     with open(out_path, "w") as f:
         json.dump(result, f)
