@@ -623,6 +623,30 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   return {dx, dgamma, dbeta};
 }
 
+// Finalize the [tiles_m][C] partial-sum slabs produced by the EpiBnBwd
+// GEMM epilogue (gemm_conv.hip conv_dgrad_bnfuse): one reduce+finalize
+// launch -> {dgamma, dbeta, tdb, tdg}. `like` supplies the elem dtype.
+std::vector<at::Tensor> bn_bwd_finalize_slabs(const at::Tensor& pdb,
+                                              const at::Tensor& pdg,
+                                              const at::Tensor& like) {
+  uint32_t nslabs = pdb.size(0), C = pdb.size(1);
+  TORCH_CHECK(C % 8 == 0, "bn: C must be a multiple of 8, got ", C);
+  uint32_t cvecs = C / 8;
+  auto opt_f = pdb.options();
+  auto tdb = at::empty({(long)C}, opt_f), tdg = at::empty({(long)C}, opt_f);
+  auto dgamma = at::empty({(long)C}, like.options());
+  auto dbeta = at::empty({(long)C}, like.options());
+  auto s = bn_stream();
+  DTMX_DISPATCH_16(like.scalar_type(), "bn_bwd_finalize_slabs", {
+    uint32_t ncv = std::min(bn_cpb(cvecs), 8u);
+    bn_bwd_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
+        pdb.data_ptr<float>(), pdg.data_ptr<float>(), (elem_t*)dgamma.data_ptr(),
+        (elem_t*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
+        C, nslabs, ncv);
+  });
+  return {dgamma, dbeta, tdb, tdg};
+}
+
 // ---- SyncBatchNorm entries (reference contrib/sync_batch_norm.cu) --------
 // Cross-rank BN = local per-channel sums -> RCCL all-reduce (Python side) ->
 // finalize/apply with the GLOBAL count. These entries split bn_fwd_train /

@@ -10,6 +10,11 @@ at::Tensor conv_fwd(const at::Tensor&, const at::Tensor&, long, long);
 at::Tensor conv_dgrad(const at::Tensor&, const at::Tensor&, long, long, long, long,
                       const c10::optional<at::Tensor>&);
 at::Tensor conv_wgrad(const at::Tensor&, const at::Tensor&, long, long, long, long);
+std::vector<at::Tensor> conv_dgrad_bnfuse(const at::Tensor&, const at::Tensor&,
+                                          long, long, long, long,
+                                          const c10::optional<at::Tensor>&,
+                                          const at::Tensor&, const at::Tensor&,
+                                          const at::Tensor&, const at::Tensor&);
 std::vector<std::tuple<long, long, double>> wgrad_ws_stats();
 // batchnorm.hip
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor&, const at::Tensor&,
@@ -30,6 +35,8 @@ std::vector<at::Tensor> bn_fwd_presummed(const at::Tensor&, const at::Tensor&,
 std::vector<at::Tensor> bn_bwd_sums(const at::Tensor&, const at::Tensor&,
                                     const at::Tensor&, const at::Tensor&,
                                     const at::Tensor&, bool);
+std::vector<at::Tensor> bn_bwd_finalize_slabs(const at::Tensor&, const at::Tensor&,
+                                              const at::Tensor&);
 std::vector<at::Tensor> bn_bwd_dx_presummed(const at::Tensor&, const at::Tensor&,
                                             const at::Tensor&, const at::Tensor&,
                                             const at::Tensor&, const at::Tensor&,
@@ -84,6 +91,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stride"), py::arg("pad"), py::arg("H"), py::arg("W"),
         py::arg("acc") = c10::nullopt);
   m.def("conv_wgrad", &dtmx::conv_wgrad);
+  m.def("conv_dgrad_bnfuse", &dtmx::conv_dgrad_bnfuse, py::arg("dy"),
+        py::arg("w"), py::arg("stride"), py::arg("pad"), py::arg("H"),
+        py::arg("W"), py::arg("acc"), py::arg("y"), py::arg("xin"),
+        py::arg("mean"), py::arg("invstd"));
+  m.def("bn_bwd_finalize_slabs", &dtmx::bn_bwd_finalize_slabs);
   m.def("wgrad_ws_stats", &dtmx::wgrad_ws_stats);
   m.def("bn_fwd_train", &dtmx::bn_fwd_train);
   m.def("bn_fwd_infer", &dtmx::bn_fwd_infer);

@@ -140,6 +140,7 @@ struct EpiBF16 {
   using elem = elem_t;
   using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
+  static constexpr bool kBnBwd = false;
   elem_t* c;
   const float* bias;  // nullable
   uint32_t M, N;
@@ -231,6 +232,35 @@ struct EpiBF16 {
   }
 };
 
+// Epilogue with fused BatchNorm-backward prologue: the
+// GEMM computes dy of a BN+ReLU layer (conv dgrad of the consumer conv, or
+// the residual-join accumulate). Instead of writing raw dy and re-reading
+// (x, dy, y) in a separate bn_bwd_stats pass (HBM-roofline, 10% of the
+// ResNet-50 step), the epilogue:
+//   - applies the ReLU mask from y in-register (g = y>0 ? dy : 0),
+//   - writes the MASKED gradient (downstream BN math never needs y again),
+//   - accumulates per-block partial sums db=Σg, dg=Σg*xhat into
+//     [tiles_m][N] slabs (no atomics: each block owns its slab row range).
+// bn_bwd_finalize_slabs + bn_bwd_dx_presummed complete the BN backward.
+// Reference analog: batch_norm.cu:314 BatchNormalizationBackwardKernel,
+// restructured as a GEMM-epilogue fusion.
+template <typename elem_t>
+struct EpiBnBwd {
+  using elem = elem_t;
+  using V8 = typename E8<elem_t>::v8;
+  static constexpr bool kLdsStage = true;
+  static constexpr bool kBnBwd = true;
+  elem_t* c;             // output: masked gradient g
+  const elem_t* acc;     // optional residual-join accumulate (may alias c)
+  const elem_t* bnb_y;   // BN(+relu) output (mask source), same [M][N] layout
+  const elem_t* bnb_x;   // pre-BN conv output (xhat source)
+  const float* bnb_mean;    // [N]
+  const float* bnb_invstd;  // [N]
+  float* bnb_pdb;           // [tiles_m][N]
+  float* bnb_pdg;           // [tiles_m][N]
+  uint32_t M, N;
+};
+
 // 1x1 stride-u dgrad: rows m = (n,p,q) of the dense dy@W^T GEMM scatter to
 // input pixels (n, p*u, q*v); everything else in dx stays zero.
 template <typename elem_t>
@@ -238,6 +268,7 @@ struct EpiBF16Scatter {
   using elem = elem_t;
   using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
+  static constexpr bool kBnBwd = false;
   elem_t* dx;
   uint32_t M, N;  // M = NPQ, N = C
   uint32_t H, W, Q;
@@ -269,6 +300,7 @@ struct EpiAtomicF32 {
   using elem = elem_t;
   using V8 = typename E8<elem_t>::v8;  // split-K partial accumulation (conv wgrad)
   static constexpr bool kLdsStage = false;
+  static constexpr bool kBnBwd = false;
   float* c;
   uint32_t M, N;
   __device__ __forceinline__ void store_chunk(uint32_t, uint32_t, V8) const {}
@@ -415,11 +447,99 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
       }
     __syncthreads();
     constexpr uint32_t CHUNKS = 128 * BN / 8;
-    for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
-      uint32_t row = idx / (BN / 8), nc = idx % (BN / 8);
-      epi.store_chunk(bm + row, bn + nc * 8, *(const V8*)(ct + row * BN + nc * 8));
+    if constexpr (EPI::kBnBwd) {
+      // BN-backward fused epilogue: mask with relu'(y), write masked g,
+      // accumulate db=Σg, dg=Σg*(x-mean)*invstd per channel. Chunk stride
+      // 256 with BN/8 | 256 keeps each thread on a FIXED 8-channel group,
+      // so the sums live in registers and one LDS tree finishes the block.
+      constexpr uint32_t NC8 = BN / 8;
+      const uint32_t nc = t % NC8;
+      const uint32_t n0 = bn + nc * 8;
+      float mean8[8], inv8[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        uint32_t n = n0 + e;
+        mean8[e] = n < epi.N ? epi.bnb_mean[n] : 0.f;
+        inv8[e] = n < epi.N ? epi.bnb_invstd[n] : 0.f;
+      }
+      float db[8] = {}, dg[8] = {};
+      for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
+        uint32_t row = idx / NC8;
+        uint32_t m = bm + row;
+        if (m >= epi.M || n0 >= epi.N) continue;
+        V8 v = *(const V8*)(ct + row * BN + nc * 8);
+        size_t off = (size_t)m * epi.N + n0;
+        uint32_t rem = epi.N - n0;
+        if (rem >= 8) {
+          V8 a{};
+          if (epi.acc) a = *(const V8*)(epi.acc + off);
+          V8 yv = *(const V8*)(epi.bnb_y + off);
+          V8 xv = *(const V8*)(epi.bnb_x + off);
+          V8 g;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            float f = (float)v[e] + (epi.acc ? (float)a[e] : 0.f);
+            if ((float)yv[e] <= 0.f) f = 0.f;
+            g[e] = (elem_t)f;
+            db[e] += f;
+            dg[e] += f * ((float)xv[e] - mean8[e]) * inv8[e];
+          }
+          *(V8*)(epi.c + off) = g;
+        } else {
+          for (uint32_t e = 0; e < rem; ++e) {
+            float f = (float)v[e] + (epi.acc ? (float)epi.acc[off + e] : 0.f);
+            if ((float)epi.bnb_y[off + e] <= 0.f) f = 0.f;
+            epi.c[off + e] = (elem_t)f;
+            db[e] += f;
+            dg[e] += f * ((float)epi.bnb_x[off + e] - mean8[e]) * inv8[e];
+          }
+        }
+      }
+      // tree-reduce the 256/NC8 threads sharing each channel group; the
+      // compute buffers are drained, so smem[1] doubles as the 8 KiB
+      // reduction scratch. Each block then writes its OWN slab row range —
+      // no atomics (vs the fwd-stats fusion's measured atomic cost).
+      float* red = (float*)&smem[1][0];
+      __syncthreads();
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] = db[e];
+      __syncthreads();
+      for (uint32_t off2 = 128; off2 >= NC8; off2 >>= 1) {
+        if (t < off2) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off2) * 8 + e];
+        }
+        __syncthreads();
+      }
+      if (t < NC8 && bn + t * 8 < epi.N) {
+        float* prow = epi.bnb_pdb + (size_t)(bm >> 7) * epi.N + bn + t * 8;
+        uint32_t remw = min(8u, epi.N - (bn + t * 8));
+        for (uint32_t e = 0; e < remw; ++e) prow[e] = red[t * 8 + e];
+      }
+      __syncthreads();
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] = dg[e];
+      __syncthreads();
+      for (uint32_t off2 = 128; off2 >= NC8; off2 >>= 1) {
+        if (t < off2) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off2) * 8 + e];
+        }
+        __syncthreads();
+      }
+      if (t < NC8 && bn + t * 8 < epi.N) {
+        float* prow = epi.bnb_pdg + (size_t)(bm >> 7) * epi.N + bn + t * 8;
+        uint32_t remw = min(8u, epi.N - (bn + t * 8));
+        for (uint32_t e = 0; e < remw; ++e) prow[e] = red[t * 8 + e];
+      }
+    } else {
+      for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
+        uint32_t row = idx / (BN / 8), nc = idx % (BN / 8);
+        epi.store_chunk(bm + row, bn + nc * 8,
+                        *(const V8*)(ct + row * BN + nc * 8));
+      }
+      epi.bn_stats(ct, bm, bn, BN, t);
     }
-    if constexpr (EPI::kLdsStage) epi.bn_stats(ct, bm, bn, BN, t);
   } else {
     epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
   }
@@ -1207,6 +1327,97 @@ at::Tensor conv_dgrad(const at::Tensor& dy, const at::Tensor& w, long stride,
 
   });
   return at::Tensor();
+}
+
+// conv dgrad with the BN-backward epilogue fusion (EpiBnBwd): computes
+// dy_bn = conv_dgrad(dy, w) [+ acc], relu-masks it with y, writes the masked
+// gradient and [tiles_m][C] partial (Σg, Σg*xhat) slabs. Returns
+// {g, pdb, pdg}; finish with bn_bwd_finalize_slabs + bn_bwd_dx_presummed.
+// Not applicable to the 1x1/stride>1 scatter path (the scatter epilogue
+// covers only 1/u^2 of the output pixels).
+std::vector<at::Tensor> conv_dgrad_bnfuse(
+    const at::Tensor& dy, const at::Tensor& w, long stride, long pad, long H,
+    long W_, const c10::optional<at::Tensor>& acc, const at::Tensor& y,
+    const at::Tensor& xin, const at::Tensor& mean, const at::Tensor& invstd) {
+  DTMX_DISPATCH_16(dy.scalar_type(), "conv_dgrad_bnfuse", {
+    CHECK_BF16_CUDA(dy);
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy must be NHWC");
+    TORCH_CHECK(y.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    xin.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "bnfuse: y/xin must be NHWC");
+    uint32_t N = dy.size(0), Ko = dy.size(1), P = dy.size(2), Q = dy.size(3);
+    uint32_t C = w.size(1), R = w.size(2), S = w.size(3);
+    TORCH_CHECK(!(R == 1 && S == 1 && stride > 1),
+                "conv_dgrad_bnfuse: strided 1x1 (scatter) is not fusable");
+    TORCH_CHECK(C % 8 == 0, "conv_dgrad_bnfuse: C must be a multiple of 8");
+    at::Tensor dyk = dy.permute({0, 2, 3, 1});
+    at::Tensor wtk = w.permute({1, 2, 3, 0});
+    if (Ko % 8) {
+      long padk = 8 - (Ko % 8);
+      dyk = at::constant_pad_nd(dyk, {0, padk}, 0.0);
+      wtk = at::constant_pad_nd(wtk, {0, padk}, 0.0);
+      Ko += padk;
+    }
+    auto dyc = dyk.contiguous();
+    at::Tensor wt;
+    if (Ko == (uint32_t)w.size(0) && Ko % 8 == 0 &&
+        w.is_contiguous(at::MemoryFormat::ChannelsLast)) {
+      uint32_t RS = R * S, K8 = Ko / 8, total = C * RS * K8;
+      wt = at::empty({(long)C, (long)R, (long)S, (long)Ko}, dy.options());
+      FastDiv dRSK8, dK8;
+      dRSK8.init(RS * K8);
+      dK8.init(K8);
+      transpose_w_crsk_kernel<<<ceil_div(total, 256), 256, 0, cur_stream()>>>(
+          (const elem_t*)w.data_ptr(), (elem_t*)wt.data_ptr(), C, RS, Ko, total,
+          dRSK8, dK8);
+    } else {
+      wt = wtk.contiguous();
+    }
+    const bool have_acc = acc.has_value() && acc->defined();
+    at::Tensor dx;
+    if (have_acc) {
+      dx = *acc;
+    } else {
+      dx = at::empty({(long)N, (long)C, (long)H, (long)W_}, dy.options(),
+                     at::MemoryFormat::ChannelsLast);
+    }
+    uint32_t M = N * H * W_, Ktot = R * S * Ko;
+    uint32_t tiles_m = ceil_div(M, 128);
+    auto opt_f = dy.options().dtype(at::kFloat);
+    // every slab element is written exactly once by its owning block
+    auto pdb = at::empty({(long)tiles_m, (long)C}, opt_f);
+    auto pdg = at::empty({(long)tiles_m, (long)C}, opt_f);
+    EpiBnBwd<elem_t> epi;
+    epi.c = (elem_t*)dx.data_ptr();
+    epi.acc = have_acc ? (const elem_t*)dx.data_ptr() : nullptr;
+    epi.bnb_y = (const elem_t*)y.data_ptr();
+    epi.bnb_x = (const elem_t*)xin.data_ptr();
+    epi.bnb_mean = mean.data_ptr<float>();
+    epi.bnb_invstd = invstd.data_ptr<float>();
+    epi.bnb_pdb = pdb.data_ptr<float>();
+    epi.bnb_pdg = pdg.data_ptr<float>();
+    epi.M = M;
+    epi.N = C;
+    if (R == 1 && S == 1 && pad == 0) {  // stride==1 guaranteed above
+      DenseP<elem_t> pa{(const elem_t*)dyc.data_ptr(), zero_page<elem_t>(dy),
+                        M, Ko, Ko};
+      DenseP<elem_t> pb{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy),
+                        C, Ko, Ko};
+      launch_gemm(pa, pb, epi, M, C, Ko);
+    } else {
+      ConvDgradA<elem_t> pa;
+      pa.dy = (const elem_t*)dyc.data_ptr();
+      pa.zero = zero_page<elem_t>(dy);
+      pa.M = M; pa.Ktot = Ktot; pa.Ko = Ko; pa.H = H; pa.W = W_; pa.P = P;
+      pa.Q = Q; pa.S = S; pa.u = stride; pa.v = stride; pa.ph = pad; pa.pw = pad;
+      pa.dW_.init(W_); pa.dHW.init(H * W_); pa.dKo.init(Ko); pa.dS.init(S);
+      DenseP<elem_t> pb{(const elem_t*)wt.data_ptr(), zero_page<elem_t>(dy),
+                        C, Ktot, Ktot};
+      launch_gemm(pa, pb, epi, M, C, Ktot);
+    }
+    return {dx, pdb, pdg};
+  });
+  return {};
 }
 
 // ------------------------------------------------------------- conv wgrad

@@ -14,6 +14,8 @@ symbols/resnet.py residual units.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from .hip import require_ext
@@ -21,6 +23,26 @@ from .hip import require_ext
 
 def _cl(t):
     return t.contiguous(memory_format=torch.channels_last)
+
+
+def _bnbwd_on() -> bool:
+    """BN-backward fusion into the dgrad epilogue (conv_dgrad_bnfuse):
+    the producing GEMM relu-masks dy and emits per-block (Σg, Σg·xhat)
+    slabs, so bn_bwd_stats never re-reads (x, dy, y) from HBM and the dx
+    pass drops its y read. Default on; DTMX_FUSE_BN_BWD=0 restores the
+    standalone bn_bwd path (numerics-identical, used by the equivalence
+    test)."""
+    return os.environ.get("DTMX_FUSE_BN_BWD", "1") == "1"
+
+
+def _bn_bwd_from_masked(ext, c, g, mean, invstd, gamma, pdb, pdg):
+    """Finish a BN backward whose relu mask + partial sums were produced by
+    the upstream GEMM epilogue. Returns (dc, dgamma, dbeta)."""
+    dgamma, dbeta, tdb, tdg = ext.bn_bwd_finalize_slabs(pdb, pdg, g)
+    rows = c.numel() // c.shape[1]
+    dc = ext.bn_bwd_dx_presummed(c, g, g, mean, invstd, gamma, tdb, tdg,
+                                 rows, False, False)[0]
+    return dc, dgamma, dbeta
 
 
 class _FusedBottleneck(torch.autograd.Function):
@@ -57,6 +79,15 @@ class _FusedBottleneck(torch.autograd.Function):
                               m1, i1, m2, i2, m3, i3, md, idn)
         ctx.stride = stride
         ctx.has_down = wd is not None
+        # cross-block BN-backward fusion handle: the NEXT block's backward
+        # produces dy3 (= its dx) and can mask+stats bn3 in its dgrad
+        # epilogue, given bn3's pre-BN output and saved stats. Passed as a
+        # tensor attribute on the block output; consumed in forward below.
+        if _bnbwd_on():
+            ctx.bn3prev = getattr(x, "_dtmx_bnout", None)
+            y3._dtmx_bnout = (c3, m3, i3)
+        else:
+            ctx.bn3prev = None
         return y3
 
     @staticmethod
@@ -66,28 +97,62 @@ class _FusedBottleneck(torch.autograd.Function):
          c1, y1, c2, y2, c3, y3, cd, sc,
          m1, i1, m2, i2, m3, i3, md, idn) = ctx.saved_tensors
         stride = ctx.stride
+        fuse = _bnbwd_on() and dy.is_cuda
         dy = _cl(dy)
         H, W = x.shape[2], x.shape[3]
         H1, W1 = y1.shape[2], y1.shape[3]
         H2, W2 = y2.shape[2], y2.shape[3]
 
-        dc3, dg3, db3, dres = ext.bn_bwd(c3, dy, g3, m3, i3, True, y3, True)
-        dy2 = ext.conv_dgrad(dc3, w3, 1, 0, H2, W2)
+        # bn3: the upstream block's dgrad epilogue may have pre-masked dy and
+        # produced the stat slabs (attribute attached by OUR return below,
+        # one block downstream)
+        slabs3 = getattr(dy, "_dtmx_bnslabs", None)
+        if slabs3 is not None:
+            dc3, dg3, db3 = _bn_bwd_from_masked(ext, c3, dy, m3, i3, g3,
+                                                slabs3[0], slabs3[1])
+            dres = dy  # already relu-masked at the residual join
+        else:
+            dc3, dg3, db3, dres = ext.bn_bwd(c3, dy, g3, m3, i3, True, y3, True)
+        if fuse:
+            dy2, pdb2, pdg2 = ext.conv_dgrad_bnfuse(dc3, w3, 1, 0, H2, W2,
+                                                    None, y2, c2, m2, i2)
+            dc2, dg2, db2 = _bn_bwd_from_masked(ext, c2, dy2, m2, i2, g2,
+                                                pdb2, pdg2)
+        else:
+            dy2 = ext.conv_dgrad(dc3, w3, 1, 0, H2, W2)
+            dc2, dg2, db2 = ext.bn_bwd(c2, dy2, g2, m2, i2, True, y2, False)
         dw3 = ext.conv_wgrad(y2, dc3, 1, 1, 1, 0)
-        dc2, dg2, db2 = ext.bn_bwd(c2, dy2, g2, m2, i2, True, y2, False)
-        dy1 = ext.conv_dgrad(dc2, w2, stride, 1, H1, W1)
+        if fuse:
+            dy1, pdb1, pdg1 = ext.conv_dgrad_bnfuse(dc2, w2, stride, 1, H1, W1,
+                                                    None, y1, c1, m1, i1)
+            dc1, dg1, db1 = _bn_bwd_from_masked(ext, c1, dy1, m1, i1, g1,
+                                                pdb1, pdg1)
+        else:
+            dy1 = ext.conv_dgrad(dc2, w2, stride, 1, H1, W1)
+            dc1, dg1, db1 = ext.bn_bwd(c1, dy1, g1, m1, i1, True, y1, False)
         dw2 = ext.conv_wgrad(y1, dc2, 3, 3, stride, 1)
-        dc1, dg1, db1 = ext.bn_bwd(c1, dy1, g1, m1, i1, True, y1, False)
         dw1 = ext.conv_wgrad(x, dc1, 1, 1, 1, 0)
+        bnp = ctx.bn3prev  # previous block's (c3, mean, invstd) or None
         if ctx.has_down:
             dcd, dgd, dbd = ext.bn_bwd(cd, dres, gd, md, idn, False, sc, False)
             dwd = ext.conv_wgrad(x, dcd, 1, 1, stride, 0)
             dx = ext.conv_dgrad(dc1, w1, 1, 0, H, W)
-            # shortcut dgrad accumulates into dx in the epilogue
-            dx = ext.conv_dgrad(dcd, wd, stride, 0, H, W, acc=dx)
+            if fuse and bnp is not None and stride == 1:
+                # last dx producer fuses the PREVIOUS block's bn3 backward
+                dx, pdb, pdg = ext.conv_dgrad_bnfuse(dcd, wd, 1, 0, H, W, dx,
+                                                     x, bnp[0], bnp[1], bnp[2])
+                dx._dtmx_bnslabs = (pdb, pdg)
+            else:
+                # shortcut dgrad accumulates into dx in the epilogue
+                dx = ext.conv_dgrad(dcd, wd, stride, 0, H, W, acc=dx)
         else:
-            # conv1 dgrad accumulates into the residual grad in place
-            dx = ext.conv_dgrad(dc1, w1, 1, 0, H, W, acc=dres)
+            if fuse and bnp is not None:
+                dx, pdb, pdg = ext.conv_dgrad_bnfuse(dc1, w1, 1, 0, H, W, dres,
+                                                     x, bnp[0], bnp[1], bnp[2])
+                dx._dtmx_bnslabs = (pdb, pdg)
+            else:
+                # conv1 dgrad accumulates into the residual grad in place
+                dx = ext.conv_dgrad(dc1, w1, 1, 0, H, W, acc=dres)
             dwd = dgd = dbd = None
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
                 dwd, dgd, dbd, None, None, None, None, None)
@@ -121,6 +186,11 @@ class _FusedBasicBlock(torch.autograd.Function):
                               sc, m1, i1, m2, i2, md, idn)
         ctx.stride = stride
         ctx.has_down = wd is not None
+        if _bnbwd_on():
+            ctx.bnprev = getattr(x, "_dtmx_bnout", None)
+            y2._dtmx_bnout = (c2, m2, i2)
+        else:
+            ctx.bnprev = None
         return y2
 
     @staticmethod
@@ -129,22 +199,47 @@ class _FusedBasicBlock(torch.autograd.Function):
         (x, w1, g1, w2, g2, wd, gd, c1, y1, c2, y2, cd, sc,
          m1, i1, m2, i2, md, idn) = ctx.saved_tensors
         stride = ctx.stride
+        fuse = _bnbwd_on() and dy.is_cuda
         dy = _cl(dy)
         H, W = x.shape[2], x.shape[3]
         H1, W1 = y1.shape[2], y1.shape[3]
 
-        dc2, dg2, db2, dres = ext.bn_bwd(c2, dy, g2, m2, i2, True, y2, True)
-        dy1 = ext.conv_dgrad(dc2, w2, 1, 1, H1, W1)
+        slabs2 = getattr(dy, "_dtmx_bnslabs", None)
+        if slabs2 is not None:
+            dc2, dg2, db2 = _bn_bwd_from_masked(ext, c2, dy, m2, i2, g2,
+                                                slabs2[0], slabs2[1])
+            dres = dy
+        else:
+            dc2, dg2, db2, dres = ext.bn_bwd(c2, dy, g2, m2, i2, True, y2, True)
+        if fuse:
+            dy1, pdb1, pdg1 = ext.conv_dgrad_bnfuse(dc2, w2, 1, 1, H1, W1,
+                                                    None, y1, c1, m1, i1)
+            dc1, dg1, db1 = _bn_bwd_from_masked(ext, c1, dy1, m1, i1, g1,
+                                                pdb1, pdg1)
+        else:
+            dy1 = ext.conv_dgrad(dc2, w2, 1, 1, H1, W1)
+            dc1, dg1, db1 = ext.bn_bwd(c1, dy1, g1, m1, i1, True, y1, False)
         dw2 = ext.conv_wgrad(y1, dc2, 3, 3, 1, 1)
-        dc1, dg1, db1 = ext.bn_bwd(c1, dy1, g1, m1, i1, True, y1, False)
         dw1 = ext.conv_wgrad(x, dc1, 3, 3, stride, 1)
+        bnp = ctx.bnprev
         if ctx.has_down:
             dcd, dgd, dbd = ext.bn_bwd(cd, dres, gd, md, idn, False, sc, False)
             dwd = ext.conv_wgrad(x, dcd, 1, 1, stride, 0)
             dx = ext.conv_dgrad(dc1, w1, stride, 1, H, W)
-            dx = ext.conv_dgrad(dcd, wd, stride, 0, H, W, acc=dx)
+            if fuse and bnp is not None and stride == 1:
+                dx, pdb, pdg = ext.conv_dgrad_bnfuse(dcd, wd, 1, 0, H, W, dx,
+                                                     x, bnp[0], bnp[1], bnp[2])
+                dx._dtmx_bnslabs = (pdb, pdg)
+            else:
+                dx = ext.conv_dgrad(dcd, wd, stride, 0, H, W, acc=dx)
         else:
-            dx = ext.conv_dgrad(dc1, w1, stride, 1, H, W, acc=dres)
+            if fuse and bnp is not None:
+                dx, pdb, pdg = ext.conv_dgrad_bnfuse(dc1, w1, stride, 1, H, W,
+                                                     dres, x, bnp[0], bnp[1],
+                                                     bnp[2])
+                dx._dtmx_bnslabs = (pdb, pdg)
+            else:
+                dx = ext.conv_dgrad(dc1, w1, stride, 1, H, W, acc=dres)
             dwd = dgd = dbd = None
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dwd, dgd, dbd,
                 None, None, None, None)

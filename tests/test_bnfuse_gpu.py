@@ -1,0 +1,158 @@
+"""BN-backward epilogue fusion equivalence (gemm_conv.hip EpiBnBwd).
+
+The fused path (conv_dgrad_bnfuse -> bn_bwd_finalize_slabs ->
+bn_bwd_dx_presummed) must reproduce the standalone path
+(conv_dgrad -> bn_bwd with fuse_relu) bit-for-bit up to fp32-reduction
+ordering. Reference semantics: relu backward then BatchNorm backward
+(reference src/operator/nn/batch_norm.cu:314).
+"""
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _ext():
+    from dtmx.ops.hip import require_ext
+
+    return require_ext()
+
+
+def _cl(t):
+    return t.contiguous(memory_format=torch.channels_last)
+
+
+def _mk_bn_layer(N, C, H, W, dtype, seed=0):
+    """Random conv output c, BN stats over c, y = relu(bn(c))."""
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    c = _cl(torch.randn((N, C, H, W), generator=g, device="cuda").to(dtype))
+    xf = c.float()
+    mean = xf.mean(dim=(0, 2, 3)).contiguous()
+    var = xf.var(dim=(0, 2, 3), unbiased=False)
+    invstd = (var + 1e-5).rsqrt().contiguous()
+    gamma = torch.randn((C,), generator=g, device="cuda").to(dtype)
+    beta = torch.randn((C,), generator=g, device="cuda").to(dtype)
+    y = _cl(
+        (
+            (xf - mean.view(1, C, 1, 1)) * invstd.view(1, C, 1, 1)
+            * gamma.float().view(1, C, 1, 1)
+            + beta.float().view(1, C, 1, 1)
+        )
+        .clamp_min(0)
+        .to(dtype)
+    )
+    return c, y, mean, invstd, gamma
+
+
+@pytest.mark.parametrize("shape", [
+    # (N, Cin(bn), Cout, H, W, R, stride, pad)
+    (4, 64, 256, 14, 14, 1, 1, 0),     # 1x1/s1 dense dgrad path
+    (4, 64, 64, 14, 14, 3, 1, 1),      # 3x3 gather dgrad path
+    (4, 128, 128, 16, 16, 3, 2, 1),    # strided 3x3 gather
+    (2, 120, 64, 10, 10, 3, 1, 1),     # non-pow2 C with N-tile tail
+])
+def test_conv_dgrad_bnfuse_matches_standalone(shape):
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    ext = _ext()
+    N, C, Ko, H, W, R, stride, pad = shape
+    dtype = torch.bfloat16
+    P = (H + 2 * pad - R) // stride + 1
+    g = torch.Generator(device="cuda").manual_seed(7)
+    dy = _cl(torch.randn((N, Ko, P, P), generator=g, device="cuda").to(dtype))
+    w = _cl(torch.randn((Ko, C, R, R), generator=g, device="cuda").to(dtype) * 0.1)
+    c, y, mean, invstd, gamma = _mk_bn_layer(N, C, H, W, dtype, seed=11)
+
+    # standalone
+    dy_bn = ext.conv_dgrad(dy, w, stride, pad, H, W)
+    dc_ref, dgamma_ref, dbeta_ref = ext.bn_bwd(c, dy_bn, gamma, mean, invstd,
+                                               True, y, False)
+
+    # fused
+    gmask, pdb, pdg = ext.conv_dgrad_bnfuse(dy, w, stride, pad, H, W, None,
+                                            y, c, mean, invstd)
+    dgamma_f, dbeta_f, tdb, tdg = ext.bn_bwd_finalize_slabs(pdb, pdg, gmask)
+    rows = N * H * W
+    dc_f = ext.bn_bwd_dx_presummed(c, gmask, gmask, mean, invstd, gamma,
+                                   tdb, tdg, rows, False, False)[0]
+
+    # masked gradient must equal relu'(y) * dy_bn exactly
+    mask = (y.float() > 0).to(dtype)
+    torch.testing.assert_close(gmask.float(), (dy_bn * mask).float(),
+                               rtol=0, atol=0)
+    torch.testing.assert_close(dgamma_f.float(), dgamma_ref.float(),
+                               rtol=0.02, atol=0.05)
+    torch.testing.assert_close(dbeta_f.float(), dbeta_ref.float(),
+                               rtol=0.02, atol=0.05)
+    torch.testing.assert_close(dc_f.float(), dc_ref.float(), rtol=0.02,
+                               atol=0.02)
+
+
+def test_conv_dgrad_bnfuse_with_acc():
+    """Residual-join accumulate + mask + stats (the cross-block bn3 seam)."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    ext = _ext()
+    N, C, Ko, H, W = 4, 64, 64, 14, 14
+    dtype = torch.bfloat16
+    g = torch.Generator(device="cuda").manual_seed(3)
+    dy = _cl(torch.randn((N, Ko, H, W), generator=g, device="cuda").to(dtype))
+    w = _cl(torch.randn((Ko, C, 1, 1), generator=g, device="cuda").to(dtype) * 0.1)
+    acc0 = _cl(torch.randn((N, C, H, W), generator=g, device="cuda").to(dtype))
+    c, y, mean, invstd, gamma = _mk_bn_layer(N, C, H, W, dtype, seed=5)
+
+    dy_sum = ext.conv_dgrad(dy, w, 1, 0, H, W, acc=acc0.clone())
+    mask = (y.float() > 0).to(dtype)
+    expected = (dy_sum.float() * mask.float()).to(dtype)
+
+    gmask, pdb, pdg = ext.conv_dgrad_bnfuse(dy, w, 1, 0, H, W, acc0.clone(),
+                                            y, c, mean, invstd)
+    torch.testing.assert_close(gmask.float(), expected.float(), rtol=0, atol=0)
+    # slabs sum to the full-tensor reductions
+    xf = c.float()
+    xhat = (xf - mean.view(1, C, 1, 1)) * invstd.view(1, C, 1, 1)
+    gm = gmask.float()
+    torch.testing.assert_close(pdb.sum(0), gm.sum(dim=(0, 2, 3)), rtol=0.01,
+                               atol=0.05)
+    torch.testing.assert_close(pdg.sum(0), (gm * xhat).sum(dim=(0, 2, 3)),
+                               rtol=0.01, atol=0.2)
+
+
+def test_fused_block_bnbwd_end_to_end(monkeypatch):
+    """Whole-model equivalence: resnet-18 fwd+bwd with the fusion on vs off
+    (exercises the interior seams AND the cross-block attribute plumbing)."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    import dtmx
+    from dtmx.models import get_symbol
+
+    def run(fuse: str):
+        monkeypatch.setenv("DTMX_FUSE_BN_BWD", fuse)
+        monkeypatch.setenv("DTMX_FUSED_BLOCK", "1")
+        torch.manual_seed(0)
+        net = get_symbol("resnet", num_layers=18, num_classes=10,
+                         image_shape="3,32,32")
+        net = net.to("cuda").to(torch.bfloat16)
+        net = net.to(memory_format=torch.channels_last)
+        torch.manual_seed(1)
+        x = torch.randn(8, 3, 32, 32, device="cuda").to(torch.bfloat16)
+        x = x.contiguous(memory_format=torch.channels_last)
+        label = torch.randint(0, 10, (8,), device="cuda")
+        net.zero_grad()
+        out = net(x)
+        loss = torch.nn.functional.cross_entropy(out.float(), label,
+                                                 reduction="sum")
+        loss.backward()
+        grads = {n: p.grad.detach().float().clone()
+                 for n, p in net.named_parameters() if p.grad is not None}
+        return loss.item(), grads
+
+    loss0, g0 = run("0")
+    loss1, g1 = run("1")
+    assert loss0 == pytest.approx(loss1, rel=1e-3)
+    assert set(g0) == set(g1)
+    for n in g0:
+        torch.testing.assert_close(g1[n], g0[n], rtol=0.03, atol=0.03,
+                                    msg=lambda m, n=n: f"{n}: {m}")
