@@ -51,7 +51,7 @@ class _FusedBottleneck(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w1, g1, b1, w2, g2, b2, w3, g3, b3, wd, gd, bd,
-                stride, bn1, bn2, bn3, bnd):
+                stride, bn1, bn2, bn3, bnd, bnprev=None, handle=None):
         ext = require_ext()
         mom, eps = bn1.momentum, bn1.eps
         c1 = ext.conv_fwd(x, w1, 1, 0)
@@ -81,13 +81,12 @@ class _FusedBottleneck(torch.autograd.Function):
         ctx.has_down = wd is not None
         # cross-block BN-backward fusion handle: the NEXT block's backward
         # produces dy3 (= its dx) and can mask+stats bn3 in its dgrad
-        # epilogue, given bn3's pre-BN output and saved stats. Passed as a
-        # tensor attribute on the block output; consumed in forward below.
-        if _bnbwd_on():
-            ctx.bn3prev = getattr(x, "_dtmx_bnout", None)
-            y3._dtmx_bnout = (c3, m3, i3)
-        else:
-            ctx.bn3prev = None
+        # epilogue, given bn3's pre-BN output and saved stats. apply() may
+        # rewrap the output tensor, so the handle travels through the
+        # `handle` dict and the WRAPPER attaches it to the real output.
+        ctx.bn3prev = bnprev
+        if handle is not None and _bnbwd_on():
+            handle["bnout"] = (c3, m3, i3)
         return y3
 
     @staticmethod
@@ -155,7 +154,7 @@ class _FusedBottleneck(torch.autograd.Function):
                 dx = ext.conv_dgrad(dc1, w1, 1, 0, H, W, acc=dres)
             dwd = dgd = dbd = None
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
-                dwd, dgd, dbd, None, None, None, None, None)
+                dwd, dgd, dbd, None, None, None, None, None, None, None)
 
 
 class _FusedBasicBlock(torch.autograd.Function):
@@ -163,7 +162,7 @@ class _FusedBasicBlock(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w1, g1, b1, w2, g2, b2, wd, gd, bd, stride,
-                bn1, bn2, bnd):
+                bn1, bn2, bnd, bnprev=None, handle=None):
         ext = require_ext()
         mom, eps = bn1.momentum, bn1.eps
         c1 = ext.conv_fwd(x, w1, stride, 1)
@@ -186,11 +185,9 @@ class _FusedBasicBlock(torch.autograd.Function):
                               sc, m1, i1, m2, i2, md, idn)
         ctx.stride = stride
         ctx.has_down = wd is not None
-        if _bnbwd_on():
-            ctx.bnprev = getattr(x, "_dtmx_bnout", None)
-            y2._dtmx_bnout = (c2, m2, i2)
-        else:
-            ctx.bnprev = None
+        ctx.bnprev = bnprev
+        if handle is not None and _bnbwd_on():
+            handle["bnout"] = (c2, m2, i2)
         return y2
 
     @staticmethod
@@ -242,7 +239,7 @@ class _FusedBasicBlock(torch.autograd.Function):
                 dx = ext.conv_dgrad(dc1, w1, stride, 1, H, W, acc=dres)
             dwd = dgd = dbd = None
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dwd, dgd, dbd,
-                None, None, None, None)
+                None, None, None, None, None, None)
 
 
 def fused_bottleneck(x, block):
@@ -251,11 +248,18 @@ def fused_bottleneck(x, block):
     gd = d[1].weight if d is not None else None
     bd = d[1].bias if d is not None else None
     bnd = d[1] if d is not None else None
-    return _FusedBottleneck.apply(
+    prev = getattr(x, "_dtmx_bnout", None) if _bnbwd_on() else None
+    handle = {}
+    out = _FusedBottleneck.apply(
         x, block.conv1.weight, block.bn1.weight, block.bn1.bias,
         block.conv2.weight, block.bn2.weight, block.bn2.bias,
         block.conv3.weight, block.bn3.weight, block.bn3.bias,
-        wd, gd, bd, block.conv2.stride, block.bn1, block.bn2, block.bn3, bnd)
+        wd, gd, bd, block.conv2.stride, block.bn1, block.bn2, block.bn3, bnd,
+        prev, handle)
+    bnout = handle.get("bnout")
+    if bnout is not None:
+        out._dtmx_bnout = bnout
+    return out
 
 
 def fused_basic_block(x, block):
@@ -264,7 +268,14 @@ def fused_basic_block(x, block):
     gd = d[1].weight if d is not None else None
     bd = d[1].bias if d is not None else None
     bnd = d[1] if d is not None else None
-    return _FusedBasicBlock.apply(
+    prev = getattr(x, "_dtmx_bnout", None) if _bnbwd_on() else None
+    handle = {}
+    out = _FusedBasicBlock.apply(
         x, block.conv1.weight, block.bn1.weight, block.bn1.bias,
         block.conv2.weight, block.bn2.weight, block.bn2.bias,
-        wd, gd, bd, block.conv1.stride, block.bn1, block.bn2, bnd)
+        wd, gd, bd, block.conv1.stride, block.bn1, block.bn2, bnd,
+        prev, handle)
+    bnout = handle.get("bnout")
+    if bnout is not None:
+        out._dtmx_bnout = bnout
+    return out

@@ -78,10 +78,13 @@ def test_conv_dgrad_bnfuse_matches_standalone(shape):
     dc_f = ext.bn_bwd_dx_presummed(c, gmask, gmask, mean, invstd, gamma,
                                    tdb, tdg, rows, False, False)[0]
 
-    # masked gradient must equal relu'(y) * dy_bn exactly
+    # masked gradient must equal relu'(y) * dy_bn (up to 1 bf16 ulp where the
+    # standalone dgrad routed through the fp32-atomic split-K path, whose
+    # k-order rounding differs); zeros at masked positions are exact.
     mask = (y.float() > 0).to(dtype)
     torch.testing.assert_close(gmask.float(), (dy_bn * mask).float(),
-                               rtol=0, atol=0)
+                               rtol=0.01, atol=0.01)
+    assert (gmask.float()[y.float() <= 0] == 0).all()
     torch.testing.assert_close(dgamma_f.float(), dgamma_ref.float(),
                                rtol=0.02, atol=0.05)
     torch.testing.assert_close(dbeta_f.float(), dbeta_ref.float(),
