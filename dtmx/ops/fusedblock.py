@@ -35,6 +35,11 @@ def _bnbwd_on() -> bool:
     return os.environ.get("DTMX_FUSE_BN_BWD", "1") == "1"
 
 
+# instrumentation: how many BN backwards took each path (read by tests and
+# the profiling harness; reset freely)
+bnbwd_stats = {"interior": 0, "cross": 0, "cross_emit": 0, "standalone": 0}
+
+
 def _bn_bwd_from_masked(ext, c, g, mean, invstd, gamma, pdb, pdg):
     """Finish a BN backward whose relu mask + partial sums were produced by
     the upstream GEMM epilogue. Returns (dc, dgamma, dbeta)."""
@@ -107,10 +112,12 @@ class _FusedBottleneck(torch.autograd.Function):
         # one block downstream)
         slabs3 = getattr(dy, "_dtmx_bnslabs", None)
         if slabs3 is not None:
+            bnbwd_stats["cross"] += 1
             dc3, dg3, db3 = _bn_bwd_from_masked(ext, c3, dy, m3, i3, g3,
                                                 slabs3[0], slabs3[1])
             dres = dy  # already relu-masked at the residual join
         else:
+            bnbwd_stats["standalone"] += 1
             dc3, dg3, db3, dres = ext.bn_bwd(c3, dy, g3, m3, i3, True, y3, True)
         if fuse:
             dy2, pdb2, pdg2 = ext.conv_dgrad_bnfuse(dc3, w3, 1, 0, H2, W2,
@@ -140,6 +147,7 @@ class _FusedBottleneck(torch.autograd.Function):
                 # last dx producer fuses the PREVIOUS block's bn3 backward
                 dx, pdb, pdg = ext.conv_dgrad_bnfuse(dcd, wd, 1, 0, H, W, dx,
                                                      x, bnp[0], bnp[1], bnp[2])
+                bnbwd_stats["cross_emit"] += 1
                 dx._dtmx_bnslabs = (pdb, pdg)
             else:
                 # shortcut dgrad accumulates into dx in the epilogue
@@ -148,6 +156,7 @@ class _FusedBottleneck(torch.autograd.Function):
             if fuse and bnp is not None:
                 dx, pdb, pdg = ext.conv_dgrad_bnfuse(dc1, w1, 1, 0, H, W, dres,
                                                      x, bnp[0], bnp[1], bnp[2])
+                bnbwd_stats["cross_emit"] += 1
                 dx._dtmx_bnslabs = (pdb, pdg)
             else:
                 # conv1 dgrad accumulates into the residual grad in place
@@ -203,10 +212,12 @@ class _FusedBasicBlock(torch.autograd.Function):
 
         slabs2 = getattr(dy, "_dtmx_bnslabs", None)
         if slabs2 is not None:
+            bnbwd_stats["cross"] += 1
             dc2, dg2, db2 = _bn_bwd_from_masked(ext, c2, dy, m2, i2, g2,
                                                 slabs2[0], slabs2[1])
             dres = dy
         else:
+            bnbwd_stats["standalone"] += 1
             dc2, dg2, db2, dres = ext.bn_bwd(c2, dy, g2, m2, i2, True, y2, True)
         if fuse:
             dy1, pdb1, pdg1 = ext.conv_dgrad_bnfuse(dc2, w2, 1, 1, H1, W1,
@@ -226,6 +237,7 @@ class _FusedBasicBlock(torch.autograd.Function):
             if fuse and bnp is not None and stride == 1:
                 dx, pdb, pdg = ext.conv_dgrad_bnfuse(dcd, wd, 1, 0, H, W, dx,
                                                      x, bnp[0], bnp[1], bnp[2])
+                bnbwd_stats["cross_emit"] += 1
                 dx._dtmx_bnslabs = (pdb, pdg)
             else:
                 dx = ext.conv_dgrad(dcd, wd, stride, 0, H, W, acc=dx)
@@ -234,6 +246,7 @@ class _FusedBasicBlock(torch.autograd.Function):
                 dx, pdb, pdg = ext.conv_dgrad_bnfuse(dc1, w1, stride, 1, H, W,
                                                      dres, x, bnp[0], bnp[1],
                                                      bnp[2])
+                bnbwd_stats["cross_emit"] += 1
                 dx._dtmx_bnslabs = (pdb, pdg)
             else:
                 dx = ext.conv_dgrad(dc1, w1, stride, 1, H, W, acc=dres)

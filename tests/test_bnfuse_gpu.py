@@ -131,9 +131,13 @@ def test_fused_block_bnbwd_end_to_end(monkeypatch):
     import dtmx
     from dtmx.models import get_symbol
 
+    from dtmx.ops import fusedblock
+
     def run(fuse: str):
         monkeypatch.setenv("DTMX_FUSE_BN_BWD", fuse)
         monkeypatch.setenv("DTMX_FUSED_BLOCK", "1")
+        for k in fusedblock.bnbwd_stats:
+            fusedblock.bnbwd_stats[k] = 0
         torch.manual_seed(0)
         net = get_symbol("resnet", num_layers=18, num_classes=10,
                          image_shape="3,32,32")
@@ -150,12 +154,19 @@ def test_fused_block_bnbwd_end_to_end(monkeypatch):
         loss.backward()
         grads = {n: p.grad.detach().float().clone()
                  for n, p in net.named_parameters() if p.grad is not None}
-        return loss.item(), grads
+        return loss.item(), grads, dict(fusedblock.bnbwd_stats)
 
-    loss0, g0 = run("0")
-    loss1, g1 = run("1")
+    loss0, g0, _ = run("0")
+    loss1, g1, stats = run("1")
+    # the cross-block seam must actually fire (attribute relay through
+    # autograd); resnet-18 on cifar shape: 8 blocks -> >=5 identity seams
+    assert stats["cross_emit"] >= 5 and stats["cross"] >= 5, stats
     assert loss0 == pytest.approx(loss1, rel=1e-3)
     assert set(g0) == set(g1)
+    # fp32 reduction order differs between the slab and standalone stats
+    # paths, and bf16 storage amplifies it element-wise — compare by relative
+    # grad-vector distance, not per-element ulps
     for n in g0:
-        torch.testing.assert_close(g1[n], g0[n], rtol=0.03, atol=0.03,
-                                    msg=lambda m, n=n: f"{n}: {m}")
+        num = (g1[n] - g0[n]).norm().item()
+        den = g0[n].norm().item() + 1e-6
+        assert num / den < 0.02, f"{n}: rel grad diff {num / den:.4f}"
