@@ -623,6 +623,30 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   return {dx, dgamma, dbeta};
 }
 
+// Wide-grid pre-reduction for very tall slab stacks: [nslabs][C] x2 ->
+// [K][C] x2 with out[k][c] = sum_{sl ≡ k (mod K)} in[sl][c]. The EpiBnBwd
+// epilogue emits one slab row per 128-row GEMM tile — up to ~25k rows at
+// bs1024 — and the channel-parallel finalize kernel launches only C/64
+// blocks, so reducing the stack there serializes on a handful of CUs
+// (measured 5.6 ms/step). Here thread (k,c) owns a strided column: K*C
+// threads fill the chip and every warp read is coalesced in c.
+__global__ void slab_prereduce2_kernel(const float* __restrict__ a,
+                                       const float* __restrict__ b,
+                                       float* __restrict__ outa,
+                                       float* __restrict__ outb, uint32_t C,
+                                       uint32_t nslabs, uint32_t K) {
+  uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= K * C) return;
+  uint32_t k = tid / C, c = tid % C;
+  float sa = 0.f, sb = 0.f;
+  for (uint32_t sl = k; sl < nslabs; sl += K) {
+    sa += a[(size_t)sl * C + c];
+    sb += b[(size_t)sl * C + c];
+  }
+  outa[(size_t)k * C + c] = sa;
+  outb[(size_t)k * C + c] = sb;
+}
+
 // Finalize the [tiles_m][C] partial-sum slabs produced by the EpiBnBwd
 // GEMM epilogue (gemm_conv.hip conv_dgrad_bnfuse): one reduce+finalize
 // launch -> {dgamma, dbeta, tdb, tdg}. `like` supplies the elem dtype.
@@ -637,10 +661,24 @@ std::vector<at::Tensor> bn_bwd_finalize_slabs(const at::Tensor& pdb,
   auto dgamma = at::empty({(long)C}, like.options());
   auto dbeta = at::empty({(long)C}, like.options());
   auto s = bn_stream();
+  const float* pa = pdb.data_ptr<float>();
+  const float* pb = pdg.data_ptr<float>();
+  at::Tensor preda, predb;
+  if (nslabs > 1024) {  // pre-reduce tall stacks across the whole chip
+    uint32_t K = 512;
+    preda = at::empty({(long)K, (long)C}, opt_f);
+    predb = at::empty({(long)K, (long)C}, opt_f);
+    uint32_t total = K * C;
+    slab_prereduce2_kernel<<<(total + 255) / 256, 256, 0, s>>>(
+        pa, pb, preda.data_ptr<float>(), predb.data_ptr<float>(), C, nslabs, K);
+    pa = preda.data_ptr<float>();
+    pb = predb.data_ptr<float>();
+    nslabs = K;
+  }
   DTMX_DISPATCH_16(like.scalar_type(), "bn_bwd_finalize_slabs", {
     uint32_t ncv = std::min(bn_cpb(cvecs), 8u);
     bn_bwd_reduce_finalize_kernel<<<(cvecs + ncv - 1) / ncv, 256, 0, s>>>(
-        pdb.data_ptr<float>(), pdg.data_ptr<float>(), (elem_t*)dgamma.data_ptr(),
+        pa, pb, (elem_t*)dgamma.data_ptr(),
         (elem_t*)dbeta.data_ptr(), tdb.data_ptr<float>(), tdg.data_ptr<float>(),
         C, nslabs, ncv);
   });
