@@ -35,6 +35,11 @@ def _bnbwd_on() -> bool:
     return os.environ.get("DTMX_FUSE_BN_BWD", "1") == "1"
 
 
+def _bnbwd_cross_on() -> bool:
+    """Separate gate for the cross-block seam (debug/bisect aid)."""
+    return _bnbwd_on() and os.environ.get("DTMX_FUSE_BN_CROSS", "1") == "1"
+
+
 # instrumentation: how many BN backwards took each path (read by tests and
 # the profiling harness; reset freely)
 bnbwd_stats = {"interior": 0, "cross": 0, "cross_emit": 0, "standalone": 0}
@@ -261,7 +266,7 @@ def fused_bottleneck(x, block):
     gd = d[1].weight if d is not None else None
     bd = d[1].bias if d is not None else None
     bnd = d[1] if d is not None else None
-    prev = getattr(x, "_dtmx_bnout", None) if _bnbwd_on() else None
+    prev = getattr(x, "_dtmx_bnout", None) if _bnbwd_cross_on() else None
     handle = {}
     out = _FusedBottleneck.apply(
         x, block.conv1.weight, block.bn1.weight, block.bn1.bias,
@@ -281,7 +286,7 @@ def fused_basic_block(x, block):
     gd = d[1].weight if d is not None else None
     bd = d[1].bias if d is not None else None
     bnd = d[1] if d is not None else None
-    prev = getattr(x, "_dtmx_bnout", None) if _bnbwd_on() else None
+    prev = getattr(x, "_dtmx_bnout", None) if _bnbwd_cross_on() else None
     handle = {}
     out = _FusedBasicBlock.apply(
         x, block.conv1.weight, block.bn1.weight, block.bn1.bias,
