@@ -77,6 +77,9 @@ std::vector<at::Tensor> layer_norm_bwd(const at::Tensor&, const at::Tensor&,
 // compress.hip
 at::Tensor quantize_2bit(const at::Tensor&, at::Tensor, double);
 at::Tensor dequantize_2bit(const at::Tensor&, long, double);
+// indexing.hip
+at::Tensor take_fwd(const at::Tensor&, const at::Tensor&);
+at::Tensor take_bwd(const at::Tensor&, const at::Tensor&, long);
 // recordio.cpp
 void register_recordio(py::module_& m);
 }  // namespace dtmx
@@ -120,6 +123,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bwd", &dtmx::dropout_bwd);
   m.def("layer_norm_fwd", &dtmx::layer_norm_fwd);
   m.def("layer_norm_bwd", &dtmx::layer_norm_bwd);
+  m.def("take_fwd", &dtmx::take_fwd);
+  m.def("take_bwd", &dtmx::take_bwd);
   m.def("quantize_2bit", &dtmx::quantize_2bit);
   m.def("dequantize_2bit", &dtmx::dequantize_2bit);
   dtmx::register_recordio(m);

@@ -93,23 +93,20 @@ class KVStore:
 
     def row_sparse_pull(self, key, out=None, priority: int = 0, row_ids=None):
         """Pull selected rows of a 2-D value (reference kvstore.py:314 /
-        kvstore_dist.h PullRowSparse_): `row_ids` selects rows of the stored
-        tensor; rows not selected are zero in `out`."""
+        kvstore_dist.h PullRowSparse_): each `row_ids` tensor selects rows
+        of the stored value; unselected rows are zero in the target."""
         if row_ids is None:
             return self.pull(key, out=out, priority=priority)
-        stored = self.pull(key)[0] if out is None else None
-        targets = _as_list(out) if out is not None else [torch.zeros(0)]
-        results = []
-        for o, rids in zip(
-            targets, _as_list(row_ids) if isinstance(row_ids, (list, tuple)) else [row_ids]
-        ):
-            src = stored if stored is not None else self.pull(key)[0]
+        stored = self.pull(key)[0]
+        rid_list = (list(row_ids) if isinstance(row_ids, (list, tuple))
+                    else [row_ids])
+        targets = (_as_list(out) if out is not None
+                   else [torch.zeros_like(stored) for _ in rid_list])
+        for dst, rids in zip(targets, rid_list):
             rids = rids.to(torch.long)
-            dst = o if out is not None else torch.zeros_like(src)
             dst.zero_()
-            dst[rids] = src[rids].to(dst.dtype)
-            results.append(dst)
-        return None if out is not None else results
+            dst[rids] = stored[rids].to(dst.dtype)
+        return None if out is not None else targets
 
     @property
     def type(self) -> str:
@@ -158,6 +155,31 @@ class KVStore:
         pass
 
 
+def _apply_row_sparse(store, ikey: int, urows: torch.Tensor,
+                      merged: torch.Tensor, world: int):
+    """Apply merged row-sparse values to a store entry: aux keys average the
+    touched rows; with an optimizer, use its lazy row update when offered
+    (reference row-sparse optimizer kernels, optimizer_op-inl.h:563-676 —
+    untouched rows stay put); otherwise densify and run the full updater."""
+    stored = store._store[ikey]
+    if store._is_aux_key(ikey):
+        stored[urows] = (merged / world).to(stored.dtype)
+    elif store._updater is not None:
+        opt = getattr(store._updater, "optimizer", None)
+        if opt is not None and hasattr(opt, "update_rows"):
+            st = store._updater.states.get(ikey)
+            if st is None:
+                st = opt.create_state(ikey, stored)
+                store._updater.states[ikey] = st
+            opt.update_rows(ikey, stored, urows, merged, st)
+        else:
+            dense = torch.zeros_like(stored, dtype=torch.float32)
+            dense[urows] = merged
+            store._updater(ikey, dense.to(stored.dtype), stored)
+    else:
+        stored[urows] = merged.to(stored.dtype)
+
+
 def _state_to_cpu(state):
     if isinstance(state, torch.Tensor):
         return state.detach().cpu()
@@ -195,7 +217,19 @@ class LocalKVStore(KVStore):
         return merged
 
     def push(self, key, value, priority: int = 0):
+        from .ops.functional import RowSparse
+
         keys = _as_list(key)
+        if isinstance(value, RowSparse):
+            ikey = self._resolve_key(keys[0])
+            urows, inverse = torch.unique(value.rows.to(torch.long),
+                                          sorted=True, return_inverse=True)
+            merged = torch.zeros(urows.numel(), value.values.shape[1],
+                                 dtype=torch.float32,
+                                 device=value.values.device)
+            merged.index_add_(0, inverse, value.values.float())
+            _apply_row_sparse(self, ikey, urows, merged, 1)
+            return
         for k, grouped in zip(keys, self._group_values(keys, value)):
             ikey = self._resolve_key(k)
             merged = self._reduce(grouped)
@@ -293,10 +327,15 @@ class DistKVStore(KVStore):
                 v.detach().copy_(t)
 
     def push(self, key, value, priority: int = 0):
+        from .ops.functional import RowSparse
+
         keys = _as_list(key)
-        values = _as_list(value)
+        values = _as_list(value) if not isinstance(value, RowSparse) else [value]
         for k, v in zip(keys, values):
             ikey = self._resolve_key(k)
+            if isinstance(v, RowSparse):
+                self._push_row_sparse(ikey, v)
+                continue
             if isinstance(v, (list, tuple)):  # per-device list: local reduce first
                 merged = v[0].detach().clone()
                 for extra in v[1:]:
@@ -328,6 +367,42 @@ class DistKVStore(KVStore):
             else:
                 o.detach().copy_(stored.to(o.device))
         return None
+
+    def _push_row_sparse(self, ikey: int, v):
+        """Row-sparse push: move only the touched rows over the wire and
+        update only those rows (reference kvstore_dist.h:452-481
+        PushRowSparse + kvstore_dist_server.h:503 DataHandleRowSparse;
+        GPU row-merge replaces kvstore_utils.cu's cub unique). Wire format:
+        length exchange -> padded all-gather of (rows, values) on the live
+        backend (RCCL on GPU, gloo on CPU) -> local unique+segment-sum."""
+        rows, vals = v.rows.to(torch.long), v.values.detach()
+        world = self.num_workers
+        if world > 1:
+            dev = vals.device
+            n = torch.tensor([rows.numel()], dtype=torch.long, device=dev)
+            ns = [torch.zeros_like(n) for _ in range(world)]
+            dist.all_gather(ns, n)
+            counts = [int(x.item()) for x in ns]
+            maxn = max(max(counts), 1)
+            D = vals.shape[1]
+            rbuf = torch.full((maxn,), -1, dtype=torch.long, device=dev)
+            vbuf = torch.zeros(maxn, D, dtype=vals.dtype, device=dev)
+            rbuf[: rows.numel()] = rows.to(dev)
+            vbuf[: rows.numel()] = vals
+            rgat = [torch.empty_like(rbuf) for _ in range(world)]
+            vgat = [torch.empty_like(vbuf) for _ in range(world)]
+            dist.all_gather(rgat, rbuf)
+            dist.all_gather(vgat, vbuf)
+            all_rows = torch.cat([r[:c] for r, c in zip(rgat, counts)])
+            all_vals = torch.cat([vv[:c] for vv, c in zip(vgat, counts)])
+        else:
+            all_rows, all_vals = rows, vals
+        urows, inverse = torch.unique(all_rows, sorted=True,
+                                      return_inverse=True)
+        merged = torch.zeros(urows.numel(), all_vals.shape[1],
+                             dtype=torch.float32, device=all_vals.device)
+        merged.index_add_(0, inverse, all_vals.float())
+        _apply_row_sparse(self, ikey, urows, merged, world)
 
     def set_gradient_compression(self, compression_params):
         from .parallel.compression import TwoBitCompression

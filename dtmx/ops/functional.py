@@ -462,3 +462,86 @@ def sync_batch_norm(x, gamma, beta, running_mean, running_var, training: bool,
                           momentum, eps)
     return _SyncBatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
                                     momentum, eps, group)
+
+
+# --------------------------------------------------- embedding / take / sparse
+
+class RowSparse:
+    """A row-sparse 2-D gradient: only `rows` carry values (reference
+    NDArray kRowSparseStorage, include/mxnet/ndarray.h; the wire unit of
+    kvstore_dist.h PushRowSparse/PullRowSparse_). `rows` is int64 sorted
+    unique, `values` is [len(rows), D]."""
+
+    __slots__ = ("rows", "values", "shape")
+
+    def __init__(self, rows: torch.Tensor, values: torch.Tensor, shape):
+        self.rows = rows
+        self.values = values
+        self.shape = tuple(shape)
+
+    def to_dense(self) -> torch.Tensor:
+        out = torch.zeros(self.shape, dtype=self.values.dtype,
+                          device=self.values.device)
+        out[self.rows] = self.values
+        return out
+
+    @staticmethod
+    def from_dense(t: torch.Tensor) -> "RowSparse":
+        nz = (t != 0).any(dim=1).nonzero(as_tuple=True)[0]
+        return RowSparse(nz, t[nz].clone(), t.shape)
+
+
+class _Take(torch.autograd.Function):
+    """Embedding/take (reference src/operator/tensor/indexing_op.cu Take /
+    AddTakeGrad): out[i] = table[idx[i]]; backward scatter-adds into an
+    fp32 accumulator (HIP atomics on GPU)."""
+
+    @staticmethod
+    def forward(ctx, table, idx):
+        ctx.save_for_backward(idx)
+        ctx.V = table.shape[0]
+        ctx.w_dtype = table.dtype
+        if _use_hip(table, op="take") and table.shape[1] % 8 == 0:
+            ext = require_ext()
+            return ext.take_fwd(table, idx.reshape(-1).long()).view(
+                *idx.shape, table.shape[1])
+        return table.index_select(0, idx.reshape(-1).long().clamp_(
+            0, table.shape[0] - 1)).view(*idx.shape, table.shape[1])
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        D = dy.shape[-1]
+        dy2 = dy.reshape(-1, D)
+        flat = idx.reshape(-1).long()
+        if dy.is_cuda and dy.dtype in (torch.bfloat16, torch.float16):
+            ext = require_ext()
+            dtab = ext.take_bwd(dy2.contiguous(), flat, ctx.V)
+        else:
+            dtab = torch.zeros(ctx.V, D, dtype=torch.float32, device=dy.device)
+            dtab.index_add_(0, flat, dy2.float())
+        return dtab.to(ctx.w_dtype), None
+
+
+def take(table: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    """out[...] = table[idx[...]] over a 2-D table (Embedding forward)."""
+    return _Take.apply(table, idx)
+
+
+def embedding_row_sparse_grad(dy: torch.Tensor, idx: torch.Tensor,
+                              vocab: int) -> RowSparse:
+    """Compute the ROW-SPARSE gradient of an embedding lookup directly —
+    only touched rows are materialized (reference kvstore row-sparse push:
+    kvstore_dist.h:452-481 moves exactly these rows). `dy` is [..., D]."""
+    D = dy.shape[-1]
+    dy2 = dy.reshape(-1, D)
+    flat = idx.reshape(-1).long()
+    rows, inverse = torch.unique(flat, sorted=True, return_inverse=True)
+    if dy.is_cuda and dy.dtype in (torch.bfloat16, torch.float16):
+        ext = require_ext()
+        vals = ext.take_bwd(dy2.contiguous(), inverse, rows.numel())
+    else:
+        vals = torch.zeros(rows.numel(), D, dtype=torch.float32,
+                           device=dy.device)
+        vals.index_add_(0, inverse, dy2.float())
+    return RowSparse(rows, vals.to(dy.dtype), (vocab, D))

@@ -183,3 +183,25 @@ class SyncBatchNorm2dNHWC(BatchNorm2dNHWC):
         return DF.sync_batch_norm(x, self.weight, self.bias, self.running_mean,
                                   self.running_var, self.training, self.momentum,
                                   self.eps, self.process_group)
+
+
+class Embedding(nn.Module):
+    """Embedding lookup over a 2-D table (reference mx.sym.Embedding /
+    src/operator/tensor/indexing_op.cu). Forward gathers rows with the HIP
+    take kernel; backward is a scatter-add. For the row-sparse gradient
+    workflow (kvstore row_sparse push of only touched rows), use
+    DF.embedding_row_sparse_grad with the saved indices."""
+
+    def __init__(self, input_dim: int, output_dim: int, dtype=None):
+        super().__init__()
+        self.input_dim = input_dim
+        self.output_dim = output_dim
+        w = torch.empty(input_dim, output_dim)
+        nn.init.uniform_(w, -0.07, 0.07)  # mxnet default embedding init scale
+        self.weight = nn.Parameter(w)
+
+    def forward(self, idx):
+        return DF.take(self.weight, idx)
+
+    def extra_repr(self):
+        return f"{self.input_dim}, {self.output_dim}"

@@ -240,6 +240,282 @@ class LBSGD(SGD):
                 weight.copy_(w32.to(weight.dtype))
 
 
+@Optimizer.register
+class NAG(Optimizer):
+    """Nesterov accelerated SGD (reference optimizer_op-inl.h NAGMomKernel /
+    python optimizer.py NAG):
+
+        g    = rescale*grad (clipped) + wd*w
+        mom  = momentum*mom + g
+        w   -= lr*(g + momentum*mom)
+    """
+
+    def __init__(self, momentum: float = 0.0, **kwargs):
+        super().__init__(**kwargs)
+        self.momentum = momentum
+
+    def create_state(self, index, weight):
+        if self.momentum != 0.0:
+            return torch.zeros_like(weight, dtype=torch.float32)
+        return None
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr = self._get_lr(index)
+        wd = self._get_wd(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            g = g.add(w32, alpha=wd)
+            if state is not None:
+                state.mul_(self.momentum).add_(g)
+                w32.add_(g.add(state, alpha=self.momentum), alpha=-lr)
+            else:
+                w32.add_(g, alpha=-lr)
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+
+@Optimizer.register
+class Signum(Optimizer):
+    """Sign-momentum SGD (reference optimizer_op-inl.h SignumKernel /
+    python optimizer.py Signum):
+
+        mom = momentum*mom - (1-momentum)*(rescale*grad, clipped)
+        w   = (1 - lr*wd_lh)*w + lr*sign(mom)
+
+    With momentum == 0 this is SignSGD: w -= lr*sign(grad)."""
+
+    def __init__(self, momentum: float = 0.9, wd_lh: float = 0.0, **kwargs):
+        super().__init__(**kwargs)
+        self.momentum = momentum
+        self.wd_lh = wd_lh
+
+    def create_state(self, index, weight):
+        if self.momentum != 0.0:
+            return torch.zeros_like(weight, dtype=torch.float32)
+        return None
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr = self._get_lr(index)
+        wd = self._get_wd(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            if state is not None:
+                # reference applies wd inside the momentum term
+                g = g.add(w32, alpha=wd)
+                state.mul_(self.momentum).add_(g, alpha=-(1.0 - self.momentum))
+                w32.mul_(1.0 - lr * self.wd_lh).add_(torch.sign(state), alpha=lr)
+            else:
+                w32.mul_(1.0 - lr * self.wd_lh).add_(torch.sign(g), alpha=-lr)
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+
+@Optimizer.register
+class RMSProp(Optimizer):
+    """RMSProp (reference optimizer_op-inl.h RMSProp{,Alex}Update / python
+    optimizer.py RMSProp). centered=False: Tieleman&Hinton; centered=True:
+    Graves' variant with gradient-mean centering and momentum gamma2."""
+
+    def __init__(self, gamma1: float = 0.9, gamma2: float = 0.9,
+                 epsilon: float = 1e-8, centered: bool = False,
+                 clip_weights: Optional[float] = None, **kwargs):
+        kwargs.setdefault("learning_rate", 0.001)
+        super().__init__(**kwargs)
+        self.gamma1 = gamma1
+        self.gamma2 = gamma2
+        self.epsilon = epsilon
+        self.centered = centered
+        self.clip_weights = clip_weights
+
+    def create_state(self, index, weight):
+        n = torch.zeros_like(weight, dtype=torch.float32)
+        if self.centered:
+            return (n, torch.zeros_like(n), torch.zeros_like(n))  # n, g, delta
+        return (n,)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr = self._get_lr(index)
+        wd = self._get_wd(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            g = g.add(w32, alpha=wd)
+            if self.centered:
+                n, gm, delta = state
+                n.mul_(self.gamma1).addcmul_(g, g, value=1 - self.gamma1)
+                gm.mul_(self.gamma1).add_(g, alpha=1 - self.gamma1)
+                denom = (n - gm * gm).add_(self.epsilon).sqrt_()
+                delta.mul_(self.gamma2).addcdiv_(g, denom, value=-lr)
+                w32.add_(delta)
+            else:
+                (n,) = state
+                n.mul_(self.gamma1).addcmul_(g, g, value=1 - self.gamma1)
+                w32.addcdiv_(g, n.sqrt().add_(self.epsilon), value=-lr)
+            if self.clip_weights:
+                w32.clamp_(-self.clip_weights, self.clip_weights)
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+
+@Optimizer.register
+class FTRL(Optimizer):
+    """FTRL-proximal (reference optimizer_op-inl.h FTRLKernel / python
+    optimizer.py Ftrl):
+
+        g  = rescale*grad (clipped)
+        z += g - (sqrt(n + g^2) - sqrt(n))/lr * w
+        n += g^2
+        w  = (sign(z)*lamda1 - z) / ((beta + sqrt(n))/lr + wd) * 1{|z|>lamda1}
+    """
+
+    def __init__(self, lamda1: float = 0.01, beta: float = 1.0, **kwargs):
+        kwargs.setdefault("learning_rate", 0.1)
+        super().__init__(**kwargs)
+        self.lamda1 = lamda1
+        self.beta = beta
+
+    def create_state(self, index, weight):
+        return (
+            torch.zeros_like(weight, dtype=torch.float32),  # z
+            torch.zeros_like(weight, dtype=torch.float32),  # n
+        )
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr = self._get_lr(index)
+        wd = self._get_wd(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        z, n = state
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            new_n = n + g * g
+            z.add_(g - (new_n.sqrt() - n.sqrt()) / lr * w32)
+            n.copy_(new_n)
+            denom = (self.beta + n.sqrt()) / lr + wd
+            w32.copy_((torch.sign(z) * self.lamda1 - z) / denom
+                      * (z.abs() > self.lamda1))
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+
+@Optimizer.register
+class AdaGrad(Optimizer):
+    """AdaGrad (reference python optimizer.py AdaGrad):
+        hist += g^2;  w -= lr * (g / (sqrt(hist) + eps) + wd*w)
+    """
+
+    def __init__(self, eps: float = 1e-7, **kwargs):
+        super().__init__(**kwargs)
+        self.float_stable_eps = eps
+
+    def create_state(self, index, weight):
+        return torch.zeros_like(weight, dtype=torch.float32)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr = self._get_lr(index)
+        wd = self._get_wd(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            state.addcmul_(g, g, value=1.0)
+            adj = g / (state.sqrt() + self.float_stable_eps)
+            if wd > 0:
+                adj = adj.add(w32, alpha=wd)
+            w32.add_(adj, alpha=-lr)
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+
+@Optimizer.register
+class GroupAdagrad(Optimizer):
+    """Row-wise AdaGrad (reference src/operator/contrib/optimizer_op.cu
+    GroupAdagrad / python contrib optimizer): one shared history entry per
+    ROW of a 2-D weight (the embedding-table optimizer — pairs with
+    row-sparse gradients where only touched rows update):
+
+        hist[row] += mean_j(g[row,j]^2)
+        w[row]    -= lr * g[row] / sqrt(hist[row] + eps)
+
+    wd is not supported (reference asserts wd == 0)."""
+
+    def __init__(self, eps: float = 1e-5, **kwargs):
+        super().__init__(**kwargs)
+        self.epsilon = eps
+
+    def create_state(self, index, weight):
+        assert weight.dim() == 2, "GroupAdagrad expects 2-D (row) weights"
+        return torch.zeros(weight.shape[0], dtype=torch.float32,
+                           device=weight.device)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr = self._get_lr(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            state.add_((g * g).mean(dim=1))
+            w32.add_(g / (state + self.epsilon).sqrt().unsqueeze(1), alpha=-lr)
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+    def update_rows(self, index, weight, grad_rows: torch.Tensor,
+                    grad_vals: torch.Tensor, state):
+        """Row-sparse update: only `grad_rows` change (reference
+        GroupAdagrad FComputeEx on row_sparse grads)."""
+        self._update_count(index)
+        lr = self._get_lr(index)
+        with torch.no_grad():
+            g = (grad_vals.float() * self.rescale_grad)
+            if self.clip_gradient is not None:
+                g = g.clamp_(-self.clip_gradient, self.clip_gradient)
+            state[grad_rows] += (g * g).mean(dim=1)
+            upd = g / (state[grad_rows] + self.epsilon).sqrt().unsqueeze(1)
+            weight[grad_rows] -= (lr * upd).to(weight.dtype)
+
+
+@Optimizer.register
+class AdaDelta(Optimizer):
+    """AdaDelta (reference python optimizer.py AdaDelta):
+        acc_g = rho*acc_g + (1-rho)*g^2
+        d     = sqrt(acc_d + eps)/sqrt(acc_g + eps) * g
+        acc_d = rho*acc_d + (1-rho)*d^2
+        w    -= d  (wd applied to g)
+    """
+
+    def __init__(self, rho: float = 0.90, epsilon: float = 1e-5, **kwargs):
+        kwargs.setdefault("learning_rate", 1.0)
+        super().__init__(**kwargs)
+        self.rho = rho
+        self.epsilon = epsilon
+
+    def create_state(self, index, weight):
+        return (
+            torch.zeros_like(weight, dtype=torch.float32),
+            torch.zeros_like(weight, dtype=torch.float32),
+        )
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        wd = self._get_wd(index)
+        g = self._preprocess(index, grad).to(torch.float32)
+        acc_g, acc_d = state
+        with torch.no_grad():
+            w32 = weight if weight.dtype == torch.float32 else weight.float()
+            g = g.add(w32, alpha=wd)
+            acc_g.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
+            d = (acc_d + self.epsilon).sqrt() / (acc_g + self.epsilon).sqrt() * g
+            acc_d.mul_(self.rho).addcmul_(d, d, value=1 - self.rho)
+            w32.sub_(d)
+            if w32 is not weight:
+                weight.copy_(w32.to(weight.dtype))
+
+
 class Updater:
     """Callable (index, grad, weight) updater with per-index state
     (reference optimizer.py get_updater / Updater class). This is what runs
