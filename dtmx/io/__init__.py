@@ -387,14 +387,17 @@ class PrefetchingIter(DataIter):
 class ImageRecordIter(DataIter):
     """RecordIO-backed image iterator on the native C++ pipeline
     (reference src/io/iter_image_recordio_2.cc; dtmx/csrc/recordio.cpp):
-    threaded record parsing + batch assembly + bounded prefetch queue,
-    sharded by (part_index, num_parts). Records carry raw uint8 HWC payloads
-    packed by tools/im2rec.py (no JPEG stage in this environment)."""
+    threaded record parsing + JPEG decode (libjpeg) + resize/random-crop/
+    mirror augmentation + batch assembly + bounded prefetch queue, sharded
+    by (part_index, num_parts). Records carry JPEG payloads or raw uint8
+    HWC of the target shape (tools/im2rec.py packs both)."""
 
     def __init__(self, path_imgrec: str, data_shape, batch_size: int,
                  shuffle: bool = False, part_index: int = 0, num_parts: int = 1,
                  preprocess_threads: int = 4, prefetch_buffer: int = 4,
-                 seed: int = 0, label_name: str = "softmax_label", **kwargs):
+                 seed: int = 0, label_name: str = "softmax_label",
+                 rand_crop: bool = False, rand_mirror: bool = False,
+                 resize: int = 0, **kwargs):
         super().__init__(batch_size)
         from dtmx.ops.hip import require_ext
 
@@ -406,6 +409,7 @@ class ImageRecordIter(DataIter):
         self._loader = ext.RecordBatchLoader(
             self._reader, batch_size, [h, w, c], part_index, num_parts,
             shuffle, preprocess_threads, prefetch_buffer, seed,
+            rand_crop, rand_mirror, resize,
         )
         self.label_name = label_name
 

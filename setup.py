@@ -34,6 +34,15 @@ setup(
         CUDAExtension(
             name="dtmx._C",
             sources=sources,
+            # libjpeg (v9, /opt/conda ships the only dev copy in this image)
+            # powers the data pipeline's decode stage (recordio.cpp); rpath
+            # pins the matching runtime .so.9. Torch/stdc++ sonames are
+            # already loaded by `import torch` before this extension, so the
+            # extra rpath entry cannot redirect them.
+            include_dirs=["/opt/conda/include"],
+            library_dirs=["/opt/conda/lib"],
+            runtime_library_dirs=["/opt/conda/lib"],
+            libraries=["jpeg"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

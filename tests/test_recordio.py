@@ -109,3 +109,129 @@ def test_recordio_python_api(tmp_path):
     assert sorted(r.keys()) == [10, 11, 12, 13]
     hdr, payload = recordio.unpack(r.read_idx(12))
     assert hdr.label == 2.0 and payload == b"p2"
+
+
+# ------------------------------------------- JPEG decode + augmentation stage
+
+def _pack(tmp_path, imgs, labels, jpeg=False, quality=95):
+    import struct
+
+    from dtmx.ops.hip import require_ext
+
+    ext = require_ext()
+    recs = []
+    for i, (im, y) in enumerate(zip(imgs, labels)):
+        h, w, c = im.shape
+        payload = (ext.encode_jpeg(im.tobytes(), h, w, c, quality)
+                   if jpeg else im.tobytes())
+        recs.append(struct.pack("<IfQQ", 0, float(y), i, 0) + payload)
+    path = str(tmp_path / ("aug_j.rec" if jpeg else "aug_r.rec"))
+    ext.write_recordio(path, recs)
+    return path
+
+
+def _smooth_images(n, h, w, c=3):
+    """Gradient images survive JPEG q95 within ~2/255 — usable as oracles."""
+    import numpy as np
+
+    ys, xs = np.mgrid[0:h, 0:w]
+    out = []
+    for i in range(n):
+        im = np.stack([(ys * 2 + i * 7) % 256, (xs * 3 + i * 11) % 256,
+                       ((ys + xs) * 2 + i * 5) % 256], axis=-1)[:, :, :c]
+        out.append(im.astype(np.uint8))
+    return out
+
+
+def test_jpeg_roundtrip_smooth():
+    import numpy as np
+
+    from dtmx.ops.hip import require_ext
+
+    ext = require_ext()
+    (im,) = _smooth_images(1, 24, 32)
+    enc = ext.encode_jpeg(im.tobytes(), 24, 32, 3, 95)
+    dec = ext.decode_jpeg(enc, 3).numpy()
+    assert dec.shape == (24, 32, 3)
+    assert np.abs(dec.astype(int) - im.astype(int)).max() <= 12
+
+
+def test_jpeg_records_decode_in_loader(tmp_path):
+    import numpy as np
+
+    from dtmx.io import ImageRecordIter
+
+    imgs = _smooth_images(8, 16, 16)
+    path = _pack(tmp_path, imgs, list(range(8)), jpeg=True)
+    it = ImageRecordIter(path, (3, 16, 16), batch_size=4,
+                         preprocess_threads=2)
+    batch = it.next()
+    x = batch.data[0].permute(0, 2, 3, 1).numpy()  # NCHW view -> NHWC
+    y = batch.label[0].numpy()
+    for i in range(4):
+        ref = imgs[int(y[i])].astype(np.float32) / 255.0
+        assert np.abs(x[i] - ref).max() < 0.06  # JPEG q95 tolerance
+
+
+def test_rand_crop_is_a_valid_crop(tmp_path):
+    import numpy as np
+
+    from dtmx.io import ImageRecordIter
+
+    imgs = _smooth_images(6, 12, 12)
+    path = _pack(tmp_path, imgs, list(range(6)), jpeg=True)
+    it = ImageRecordIter(path, (3, 8, 8), batch_size=6, rand_crop=True,
+                         rand_mirror=True, preprocess_threads=3, seed=5)
+    batch = it.next()
+    x = (batch.data[0].permute(0, 2, 3, 1).numpy() * 255.0)
+    y = batch.label[0].numpy()
+    from dtmx.ops.hip import require_ext
+
+    ext = require_ext()
+    for i in range(6):
+        im = imgs[int(y[i])]
+        # decode oracle (the loader crops the DECODED image)
+        dec = ext.decode_jpeg(ext.encode_jpeg(im.tobytes(), 12, 12, 3, 95),
+                              3).numpy().astype(np.float32)
+        best = 1e9
+        for y0 in range(5):
+            for x0 in range(5):
+                for mirror in (False, True):
+                    cand = dec[y0:y0 + 8, x0:x0 + 8]
+                    if mirror:
+                        cand = cand[:, ::-1]
+                    best = min(best, np.abs(cand - x[i]).max())
+        assert best < 1.0, f"output {i} is not any crop/mirror (err {best})"
+
+
+def test_augmentation_deterministic_per_seed(tmp_path):
+    import numpy as np
+
+    from dtmx.io import ImageRecordIter
+
+    imgs = _smooth_images(8, 14, 14)
+    path = _pack(tmp_path, imgs, list(range(8)), jpeg=True)
+
+    def first_batch(seed):
+        it = ImageRecordIter(path, (3, 10, 10), batch_size=8, rand_crop=True,
+                             rand_mirror=True, preprocess_threads=3, seed=seed)
+        return it.next().data[0].numpy()
+
+    a1, a2, b = first_batch(3), first_batch(3), first_batch(4)
+    assert np.array_equal(a1, a2)  # same seed -> same augmentation
+    assert not np.array_equal(a1, b)  # different seed -> different crops
+
+
+def test_resize_shorter_side(tmp_path):
+    import numpy as np
+
+    from dtmx.io import ImageRecordIter
+
+    imgs = _smooth_images(4, 32, 20)  # portrait: shorter side = W
+    path = _pack(tmp_path, imgs, list(range(4)), jpeg=True)
+    it = ImageRecordIter(path, (3, 12, 12), batch_size=4, resize=12,
+                         preprocess_threads=2)
+    batch = it.next()
+    x = batch.data[0]
+    assert tuple(x.shape) == (4, 3, 12, 12)
+    assert 0.0 <= float(x.min()) and float(x.max()) <= 1.0

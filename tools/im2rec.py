@@ -1,11 +1,12 @@
 #!/usr/bin/env python3
 """Pack arrays into a RecordIO file (reference tools/im2rec.py / im2rec.cc).
 
-No image decode libraries ship in this environment, so records carry raw
-uint8 HWC payloads behind the reference IRHeader (flag,label,id,id2):
+Records carry either JPEG payloads (libjpeg encode; --jpeg) or raw uint8
+HWC behind the reference IRHeader (flag,label,id,id2):
 
     python tools/im2rec.py out.rec --from-npz data.npz          # x:[N,H,W,C] u8, y:[N]
     python tools/im2rec.py out.rec --synthetic N H W C CLASSES  # random data
+    python tools/im2rec.py out.rec --synthetic ... --jpeg --quality 90
 """
 import argparse
 import os
@@ -28,6 +29,9 @@ def main():
     ap.add_argument("--from-npz", type=str, default=None)
     ap.add_argument("--synthetic", type=int, nargs=5, default=None,
                     metavar=("N", "H", "W", "C", "CLASSES"))
+    ap.add_argument("--jpeg", action="store_true",
+                    help="JPEG-encode payloads (reference im2rec default)")
+    ap.add_argument("--quality", type=int, default=95)
     args = ap.parse_args()
     from dtmx.ops.hip import require_ext
 
@@ -40,7 +44,13 @@ def main():
         rng = np.random.RandomState(0)
         x = rng.randint(0, 256, (n, h, w, c), dtype=np.uint8)
         y = rng.randint(0, k, (n,))
-    records = [pack_record(y[i], x[i].tobytes(), i) for i in range(len(x))]
+    if args.jpeg:
+        h, w, c = x.shape[1:]
+        records = [pack_record(y[i], ext.encode_jpeg(x[i].tobytes(), h, w, c,
+                                                     args.quality), i)
+                   for i in range(len(x))]
+    else:
+        records = [pack_record(y[i], x[i].tobytes(), i) for i in range(len(x))]
     ext.write_recordio(args.out, records)
     print(f"wrote {len(records)} records to {args.out}")
 
