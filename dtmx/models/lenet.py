@@ -29,5 +29,6 @@ class LeNet(nn.Module):
         return self.classifier(x)
 
 
-def get_symbol(num_classes=10, **kwargs):
-    return LeNet(num_classes=num_classes)
+def get_symbol(num_classes=10, image_shape="1,28,28", **kwargs):
+    c, h, _ = (int(x) for x in image_shape.split(","))
+    return LeNet(num_classes=num_classes, in_channels=c, image_hw=h)

@@ -146,16 +146,34 @@ class Scheduler:
             time.sleep(poll_seconds)
 
     def num_dead(self) -> int:
-        n = 0
+        return len(self.dead_members())
+
+    def dead_members(self) -> List[str]:
+        out = []
         now = time.time()
         for m in self.members:
             try:
                 ts = float(self.store.get(f"hb/{m}"))
                 if now - ts > _HB_TIMEOUT:
-                    n += 1
+                    out.append(m)
             except Exception:
-                n += 1
-        return n
+                out.append(m)
+        return out
+
+    def prune_dead(self) -> List[str]:
+        """Publish a roster without heartbeat-expired members (UNPLANNED
+        failure handling; reference postoffice.cc:410-429 GetDeadNodes +
+        van.cc dead-node accounting). Unlike planned removal, death pruning
+        may remove initial workers — a dead node cannot object — so this
+        emits directly rather than via publish()'s policy filter. Survivors
+        blocked on a timed-out collective pick up the new generation at
+        their next barrier (or exit loudly mid-epoch)."""
+        dead = self.dead_members()
+        if not dead or len(dead) >= len(self.members):
+            return []
+        survivors = sorted(set(self.members) - set(dead))
+        self._emit(survivors, removed=sorted(dead))
+        return dead
 
 
 def read_hostfile(path: str) -> List[str]:
@@ -231,8 +249,14 @@ class ElasticContext:
         rank = members.index(self.worker_id)
         world = len(members)
         prefix = dist.PrefixStore(f"gen{version}", self.store)
+        # bounded collectives: when a peer dies MID-epoch (no barrier in
+        # sight), the next all-reduce must fail loudly after DTMX_PG_TIMEOUT
+        # instead of hanging the survivors forever (reference analog: ps-lite
+        # heartbeat timeout surfacing as van.cc dead-node handling)
+        timeout = datetime.timedelta(
+            seconds=float(os.environ.get("DTMX_PG_TIMEOUT", "300")))
         dist.init_process_group(backend=_backend(), store=prefix,
-                                rank=rank, world_size=world)
+                                rank=rank, world_size=world, timeout=timeout)
         self.version = version
         self.members = members
         if torch.cuda.is_available():
@@ -295,9 +319,18 @@ class ElasticContext:
     # -- heartbeats (reference van.cc:686-698) ------------------------------
     def _start_heartbeat(self):
         self._hb_stop = threading.Event()
+        # PS_DROP_MSG fault injection (reference ps-lite van.cc:430-432:
+        # random message drop to exercise failure detection): probability of
+        # dropping each heartbeat. PS_DROP_MSG=1 makes this worker appear
+        # dead after PS_HEARTBEAT_TIMEOUT without killing it.
+        drop_p = float(os.environ.get("PS_DROP_MSG", "0"))
 
         def beat():
+            import random
+
             while not self._hb_stop.wait(_HB_INTERVAL):
+                if drop_p > 0 and random.random() < drop_p:
+                    continue
                 try:
                     self.store.set(f"hb/{self.worker_id}", str(time.time()))
                 except Exception:

@@ -2,12 +2,15 @@
 test_io, test_metric, test_lr_scheduler, test_initializer subsets)."""
 import math
 import os
+import sys
 
 import numpy as np
 import pytest
 import torch
 
 import dtmx
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 from dtmx import initializer, lr_scheduler, metric
 from dtmx.io import CSVIter, DataBatch, NDArrayIter, ResizeIter, SyntheticDataIter
 from dtmx.optimizer import SGD, Adam, LBSGD, create as opt_create, get_updater
@@ -208,6 +211,7 @@ def test_speedometer_and_checkpoint_callbacks(tmp_path, caplog):
     import logging
 
     import dtmx
+
     from dtmx.callback import BatchEndParam, Speedometer, do_checkpoint
     from dtmx.io import DataBatch
     from dtmx.models import get_symbol
@@ -236,6 +240,7 @@ def test_speedometer_and_checkpoint_callbacks(tmp_path, caplog):
 
 def test_monitor_collects_stats():
     import dtmx
+
     from dtmx.io import DataBatch
     from dtmx.models import get_symbol
     from dtmx.monitor import Monitor
@@ -301,3 +306,42 @@ def test_mnist_iter_synthetic_fallback(tmp_path):
     b = it.next()
     assert b.data[0].shape == (32, 1, 28, 28)
     assert b.label[0].shape == (32,)
+
+
+# ----------------------------------------------------- cluster tools tail
+
+def test_parse_log(tmp_path):
+    sys.path.insert(0, os.path.join(ROOT, "tools"))
+    import parse_log
+
+    lines = [
+        "INFO Epoch[0] Train-accuracy=0.512000\n",
+        "INFO Epoch[0] Time cost=12.345\n",
+        "INFO Epoch[0] Validation-accuracy=0.423000\n",
+        "INFO Epoch[1] Train-accuracy=0.734000\n",
+        "INFO Epoch[1] Time cost=11.002\n",
+    ]
+    rows = parse_log.parse(lines)
+    assert rows[0]["train-accuracy"] == 0.512
+    assert rows[0]["valid-accuracy"] == 0.423
+    assert rows[1]["time"] == 11.002
+    md = parse_log.to_markdown(rows)
+    assert md.count("|") > 10 and "0.734" in md
+
+
+def test_kill_dtmx_pidfiles(tmp_path):
+    import subprocess
+    import sys as _sys
+
+    sys.path.insert(0, os.path.join(ROOT, "tools"))
+    import kill_dtmx
+
+    rundir = tmp_path / "run"
+    rundir.mkdir()
+    p = subprocess.Popen([_sys.executable, "-c", "import time; time.sleep(60)"])
+    (rundir / "127.0.0.1#0.pid").write_text(str(p.pid))
+    (rundir / "stale.pid").write_text("999999")  # no such pid: skipped
+    n = kill_dtmx.kill_local(str(rundir))
+    assert n == 1
+    assert p.wait(timeout=15) != 0  # SIGTERM'd
+    assert not list(rundir.glob("*.pid"))  # ledger cleaned

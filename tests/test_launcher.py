@@ -115,3 +115,58 @@ def _run_launcher_join(tmp_path):
     # scheduler audit log recorded the addition
     log = (tmp_path / "hosts_log").read_text()
     assert "ADDED" in log, log
+
+
+def test_ssh_launch_path(tmp_path, monkeypatch):
+    """Joiner-via-ssh command path (reference elastic_training.cc:26-62
+    launchCommandOnNewWorker / dmlc_tracker ssh.submit): exercised against a
+    MOCK ssh on PATH that executes the remote command locally — verifies the
+    command string, env forwarding and cd-to-workdir without needing sshd."""
+    import subprocess
+    import sys
+    import time
+
+    # mock ssh: drop the option args, run the final command with bash
+    mock = tmp_path / "bin"
+    mock.mkdir()
+    ssh = mock / "ssh"
+    ssh.write_text("#!/bin/bash\n"
+                   "# args: -o StrictHostKeyChecking=no <host> <cmd>\n"
+                   'echo "$3" > "%s/ssh_host"\n'
+                   'exec bash -c "$4"\n' % tmp_path)
+    ssh.chmod(0o755)
+    monkeypatch.setenv("PATH", f"{mock}:{os.environ['PATH']}")
+
+    marker = tmp_path / "marker"
+    sys.path.insert(0, os.path.join(ROOT, "tools"))
+    import importlib
+
+    import launch as launch_mod
+    importlib.reload(launch_mod)
+
+    class A:
+        hostfile = None
+        num_workers = 1
+        scheduler_host = "127.0.0.1"
+        scheduler_port = _free_port()
+        elastic_training_enabled = False
+        poll_seconds = 0.2
+        sync_dst_dir = None
+
+    # worker command writes its DMLC env + cwd to the marker
+    cmd = [sys.executable, "-c",
+           "import os,sys; open(%r,'w').write("
+           "os.environ.get('DMLC_WORKER_ID','')+'\\n'+os.getcwd())" % str(marker)]
+    lau = launch_mod.Launcher(A(), cmd)
+    lau.launch_worker("fakehost#0", new_worker=False)
+    p = lau.procs["fakehost#0"]
+    assert p.wait(timeout=60) == 0
+    for _ in range(100):
+        if marker.exists():
+            break
+        time.sleep(0.1)
+    host_seen = (tmp_path / "ssh_host").read_text().strip()
+    assert host_seen == "fakehost"
+    wid, cwd = marker.read_text().split("\n")
+    assert wid == "fakehost#0"
+    assert cwd == os.getcwd()  # launcher cd's to the working dir

@@ -112,6 +112,15 @@ class Launcher:
             )
             p = subprocess.Popen(["ssh", "-o", "StrictHostKeyChecking=no", host, cmd])
         self.procs[wid] = p
+        # exact-PID ledger for tools/kill_dtmx.py (reference kill-mxnet.py
+        # redesigned: pidfiles instead of process-name pattern kills)
+        rundir = os.path.expanduser(os.environ.get("DTMX_RUN_DIR", "~/.dtmx/run"))
+        try:
+            os.makedirs(rundir, exist_ok=True)
+            with open(os.path.join(rundir, wid.replace("/", "_") + ".pid"), "w") as f:
+                f.write(str(p.pid))
+        except OSError:
+            pass
         logging.info("launched worker %s (pid %d, new=%s)", wid, p.pid, new_worker)
 
     def watch(self, stop: threading.Event):
@@ -119,6 +128,12 @@ class Launcher:
         last = list(self.initial_members)
         while not stop.is_set():
             time.sleep(self.args.poll_seconds)
+            # unplanned-death pruning: heartbeat-expired members leave the
+            # roster so survivors re-form at their next barrier
+            # (reference postoffice.cc:410-429 dead-node accounting)
+            pruned = self.scheduler.prune_dead()
+            for wid in pruned:
+                logging.warning("pruned dead worker %s from roster", wid)
             if not self.args.hostfile:
                 continue
             try:
