@@ -16,6 +16,10 @@ namespace dtmx {
 
 static hipStream_t bn_stream() { return at::hip::getCurrentHIPStream().stream(); }
 
+__global__ void slab_prereduce2_kernel(const float* a, const float* b,
+                                       float* outa, float* outb, uint32_t C,
+                                       uint32_t nslabs, uint32_t K);
+
 // tree-reduce 8 per-thread floats across the threads sharing a channel
 // vector (stride cvecs in the block), then write the block's partial row to
 // a [gridDim.y][C] slab. No atomics: same-address fp32 atomicAdd serializes
@@ -509,6 +513,18 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
     psum = *pre_psum;
     psumsq = *pre_psumsq;
     nslabs = psum.size(0);
+    if (nslabs > 1024) {  // epilogue slabs have tiles_m rows: pre-reduce
+      uint32_t K = 512;
+      auto pa = at::empty({(long)K, (long)C}, opt_f);
+      auto pb = at::empty({(long)K, (long)C}, opt_f);
+      uint32_t total = K * C;
+      slab_prereduce2_kernel<<<(total + 255) / 256, 256, 0, bn_stream()>>>(
+          psum.data_ptr<float>(), psumsq.data_ptr<float>(),
+          pa.data_ptr<float>(), pb.data_ptr<float>(), C, nslabs, K);
+      psum = pa;
+      psumsq = pb;
+      nslabs = K;
+    }
   } else {
     psum = at::empty({(long)nslabs, (long)C}, opt_f);
     psumsq = at::empty({(long)nslabs, (long)C}, opt_f);

@@ -141,16 +141,11 @@ struct EpiBF16 {
   using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
   static constexpr bool kBnBwd = false;
+  static constexpr bool kFwdStats = false;
   elem_t* c;
   const float* bias;  // nullable
   uint32_t M, N;
   int relu;
-  // optional fused BN statistics: per-block column sums/sumsq of the output
-  // tile into [tiles_m][N] slabs (zeroed by the launcher) — the following
-  // BatchNorm then skips its own stats read of the whole tensor. Only set
-  // when bias == nullptr && !relu (stats must equal the written values).
-  float* bn_psum = nullptr;
-  float* bn_psumsq = nullptr;
   // optional elementwise accumulator: out = gemm + acc (residual-join grad
   // fusion; acc may alias c for in-place accumulate — same thread reads and
   // writes the same element, so aliasing is safe)
@@ -207,29 +202,6 @@ struct EpiBF16 {
       }
   }
 
-  // column sums/sumsq of the block's LDS-staged tile; distinct (slab_row, n)
-  // targets per block, 256/BN stripes share one via atomicAdd.
-  __device__ __forceinline__ void bn_stats(const elem_t* ct, uint32_t bm,
-                                           uint32_t bn, uint32_t BN,
-                                           uint32_t t) const {
-    if (!bn_psum) return;
-    const uint32_t stripes = 256 / BN;
-    const uint32_t col = t % BN;
-    const uint32_t n = bn + col;
-    if (n >= N) return;
-    const uint32_t rows_per = 128 / stripes;
-    const uint32_t r0 = (t / BN) * rows_per;
-    float sum = 0.f, sumsq = 0.f;
-    for (uint32_t r = 0; r < rows_per; ++r) {
-      if (bm + r0 + r >= M) break;
-      float v = (float)ct[(r0 + r) * BN + col];
-      sum += v;
-      sumsq += v * v;
-    }
-    const size_t slab = (size_t)(bm >> 7) * N + n;
-    atomicAdd(&bn_psum[slab], sum);
-    atomicAdd(&bn_psumsq[slab], sumsq);
-  }
 };
 
 // Epilogue with fused BatchNorm-backward prologue: the
@@ -250,6 +222,7 @@ struct EpiBnBwd {
   using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
   static constexpr bool kBnBwd = true;
+  static constexpr bool kFwdStats = false;
   elem_t* c;             // output: masked gradient g
   const elem_t* acc;     // optional residual-join accumulate (may alias c)
   const elem_t* bnb_y;   // BN(+relu) output (mask source), same [M][N] layout
@@ -269,6 +242,7 @@ struct EpiBF16Scatter {
   using V8 = typename E8<elem_t>::v8;
   static constexpr bool kLdsStage = true;
   static constexpr bool kBnBwd = false;
+  static constexpr bool kFwdStats = false;
   elem_t* dx;
   uint32_t M, N;  // M = NPQ, N = C
   uint32_t H, W, Q;
@@ -291,8 +265,6 @@ struct EpiBF16Scatter {
   template <int NJ>
   __device__ __forceinline__ void store(const f32x4 (&acc)[4][NJ], uint32_t,
                                         uint32_t, uint32_t) const {}
-  __device__ __forceinline__ void bn_stats(const elem_t*, uint32_t, uint32_t,
-                                           uint32_t, uint32_t) const {}
 };
 
 template <typename elem_t>
@@ -301,6 +273,7 @@ struct EpiAtomicF32 {
   using V8 = typename E8<elem_t>::v8;  // split-K partial accumulation (conv wgrad)
   static constexpr bool kLdsStage = false;
   static constexpr bool kBnBwd = false;
+  static constexpr bool kFwdStats = false;
   float* c;
   uint32_t M, N;
   __device__ __forceinline__ void store_chunk(uint32_t, uint32_t, V8) const {}
@@ -322,6 +295,52 @@ struct EpiAtomicF32 {
       }
   }
 };
+
+// conv fwd epilogue with fused BN statistics — the SECOND attempt at this
+// fusion: round 1 used per-stripe atomicAdd into shared slab rows and
+// measured -3.5% end to end; this version reuses the EpiBnBwd scheme
+// (fixed 8-channel group per thread + one LDS tree + exclusive slab-row
+// writes, no atomics) that measured +6% on the backward side. The slabs are
+// column sums/sumsq of the values the epilogue itself writes, so the
+// following BatchNorm skips its whole-tensor stats read.
+template <typename elem_t>
+struct EpiBF16FwdStats {
+  using elem = elem_t;
+  using V8 = typename E8<elem_t>::v8;
+  static constexpr bool kLdsStage = true;
+  static constexpr bool kBnBwd = false;
+  static constexpr bool kFwdStats = true;
+  elem_t* c;
+  float* bn_psum;    // [tiles_m][N]
+  float* bn_psumsq;  // [tiles_m][N]
+  uint32_t M, N;
+};
+
+// column-group tree reduction + exclusive slab-row write shared by the
+// fused-BN epilogues: threads t ≡ nc (mod NC8) hold partials for channel
+// group nc; after the tree, thread t < NC8 writes its 8 channels of the
+// block's own slab row. `red` is 8 KiB of post-compute LDS.
+template <int NC8>
+__device__ __forceinline__ void epi_colreduce_write(
+    float* red, const float (&v)[8], float* slab, uint32_t bm, uint32_t bn,
+    uint32_t N, uint32_t t) {
+  __syncthreads();
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[t * 8 + e] = v[e];
+  __syncthreads();
+  for (uint32_t off = 128; off >= NC8; off >>= 1) {
+    if (t < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (t < NC8 && bn + t * 8 < N) {
+    float* prow = slab + (size_t)(bm >> 7) * N + bn + t * 8;
+    uint32_t remw = min(8u, N - (bn + t * 8));
+    for (uint32_t e = 0; e < remw; ++e) prow[e] = red[t * 8 + e];
+  }
+}
 
 // ------------------------------------------------------------------- kernel
 
@@ -495,50 +514,51 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
           }
         }
       }
-      // tree-reduce the 256/NC8 threads sharing each channel group; the
-      // compute buffers are drained, so smem[1] doubles as the 8 KiB
-      // reduction scratch. Each block then writes its OWN slab row range —
-      // no atomics (vs the fwd-stats fusion's measured atomic cost).
+      // tree-reduce + exclusive slab-row writes — no atomics (vs round 1's
+      // measured atomic cost); smem[1] is drained, reuse as 8 KiB scratch
       float* red = (float*)&smem[1][0];
-      __syncthreads();
+      epi_colreduce_write<(int)NC8>(red, db, epi.bnb_pdb, bm, bn, epi.N, t);
+      epi_colreduce_write<(int)NC8>(red, dg, epi.bnb_pdg, bm, bn, epi.N, t);
+    } else if constexpr (EPI::kFwdStats) {
+      // fused forward BN statistics: column sums/sumsq of the tile the
+      // epilogue writes — the following BatchNorm skips its stats pass
+      constexpr uint32_t NC8 = BN / 8;
+      const uint32_t nc = t % NC8;
+      const uint32_t n0 = bn + nc * 8;
+      float s[8] = {}, ss[8] = {};
+      for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
+        uint32_t row = idx / NC8;
+        uint32_t m = bm + row;
+        if (m >= epi.M || n0 >= epi.N) continue;
+        V8 v = *(const V8*)(ct + row * BN + nc * 8);
+        size_t off = (size_t)m * epi.N + n0;
+        uint32_t rem = epi.N - n0;
+        if (rem >= 8) {
+          *(V8*)(epi.c + off) = v;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) red[t * 8 + e] = db[e];
-      __syncthreads();
-      for (uint32_t off2 = 128; off2 >= NC8; off2 >>= 1) {
-        if (t < off2) {
-#pragma unroll
-          for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off2) * 8 + e];
+          for (int e = 0; e < 8; ++e) {
+            float f = (float)v[e];
+            s[e] += f;
+            ss[e] += f * f;
+          }
+        } else {
+          for (uint32_t e = 0; e < rem; ++e) {
+            epi.c[off + e] = v[e];
+            float f = (float)v[e];
+            s[e] += f;
+            ss[e] += f * f;
+          }
         }
-        __syncthreads();
       }
-      if (t < NC8 && bn + t * 8 < epi.N) {
-        float* prow = epi.bnb_pdb + (size_t)(bm >> 7) * epi.N + bn + t * 8;
-        uint32_t remw = min(8u, epi.N - (bn + t * 8));
-        for (uint32_t e = 0; e < remw; ++e) prow[e] = red[t * 8 + e];
-      }
-      __syncthreads();
-#pragma unroll
-      for (int e = 0; e < 8; ++e) red[t * 8 + e] = dg[e];
-      __syncthreads();
-      for (uint32_t off2 = 128; off2 >= NC8; off2 >>= 1) {
-        if (t < off2) {
-#pragma unroll
-          for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off2) * 8 + e];
-        }
-        __syncthreads();
-      }
-      if (t < NC8 && bn + t * 8 < epi.N) {
-        float* prow = epi.bnb_pdg + (size_t)(bm >> 7) * epi.N + bn + t * 8;
-        uint32_t remw = min(8u, epi.N - (bn + t * 8));
-        for (uint32_t e = 0; e < remw; ++e) prow[e] = red[t * 8 + e];
-      }
+      float* red = (float*)&smem[1][0];
+      epi_colreduce_write<(int)NC8>(red, s, epi.bn_psum, bm, bn, epi.N, t);
+      epi_colreduce_write<(int)NC8>(red, ss, epi.bn_psumsq, bm, bn, epi.N, t);
     } else {
       for (uint32_t idx = t; idx < CHUNKS; idx += 256) {
         uint32_t row = idx / (BN / 8), nc = idx % (BN / 8);
         epi.store_chunk(bm + row, bn + nc * 8,
                         *(const V8*)(ct + row * BN + nc * 8));
       }
-      epi.bn_stats(ct, bm, bn, BN, t);
     }
   } else {
     epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
@@ -1212,10 +1232,15 @@ std::vector<at::Tensor> conv_fwd_stats(const at::Tensor& x, const at::Tensor& w,
     auto y = at::empty({(long)N, (long)Ko, (long)P, (long)Q}, x.options(),
                        at::MemoryFormat::ChannelsLast);
     auto opt_f = x.options().dtype(at::kFloat);
-    auto psum = at::zeros({(long)tiles_m, (long)Ko}, opt_f);
-    auto psumsq = at::zeros({(long)tiles_m, (long)Ko}, opt_f);
-    EpiBF16<elem_t> epi{(elem_t*)y.data_ptr(), nullptr, M, Ko, 0,
-                        psum.data_ptr<float>(), psumsq.data_ptr<float>()};
+    // every slab element is written exactly once by its owning block
+    auto psum = at::empty({(long)tiles_m, (long)Ko}, opt_f);
+    auto psumsq = at::empty({(long)tiles_m, (long)Ko}, opt_f);
+    EpiBF16FwdStats<elem_t> epi;
+    epi.c = (elem_t*)y.data_ptr();
+    epi.bn_psum = psum.data_ptr<float>();
+    epi.bn_psumsq = psumsq.data_ptr<float>();
+    epi.M = M;
+    epi.N = Ko;
     if (R == 1 && S == 1 && stride == 1 && pad == 0) {
       DenseP<elem_t> pa{(const elem_t*)x.data_ptr(), zero_page<elem_t>(x), M, C, C};
       DenseP<elem_t> pb{(const elem_t*)w.data_ptr(), zero_page<elem_t>(x), Ko, C, C};

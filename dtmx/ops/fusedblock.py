@@ -40,6 +40,24 @@ def _bnbwd_cross_on() -> bool:
     return _bnbwd_on() and os.environ.get("DTMX_FUSE_BN_CROSS", "1") == "1"
 
 
+def _bnstats_on() -> bool:
+    """Forward BN-stats fusion: the producing conv's epilogue emits the
+    (sum, sumsq) slabs, so bn_fwd_train skips its whole-tensor stats read.
+    No-atomic scheme (round 1's atomicAdd version measured -3.5% and was
+    reverted; see EpiBF16FwdStats, gemm_conv.hip)."""
+    return os.environ.get("DTMX_FUSE_BN_STATS", "1") == "1"
+
+
+def _conv_fs(ext, x, w, stride, pad, want):
+    """conv forward, optionally with fused BN-stat slabs."""
+    if want:
+        c, ps, pss = ext.conv_fwd_stats(x, w, stride, pad)
+        if ps.numel():
+            return c, ps, pss
+        return c, None, None
+    return ext.conv_fwd(x, w, stride, pad), None, None
+
+
 # instrumentation: how many BN backwards took each path (read by tests and
 # the profiling harness; reset freely)
 bnbwd_stats = {"interior": 0, "cross": 0, "cross_emit": 0, "standalone": 0}
@@ -64,26 +82,27 @@ class _FusedBottleneck(torch.autograd.Function):
                 stride, bn1, bn2, bn3, bnd, bnprev=None, handle=None):
         ext = require_ext()
         mom, eps = bn1.momentum, bn1.eps
-        c1 = ext.conv_fwd(x, w1, 1, 0)
+        fs = _bnstats_on()
+        c1, ps1, pss1 = _conv_fs(ext, x, w1, 1, 0, fs)
         y1, m1, i1 = ext.bn_fwd_train(c1, g1, b1, bn1.running_mean,
                                       bn1.running_var, mom, eps, True, None,
-                                      None, None)
-        c2 = ext.conv_fwd(y1, w2, stride, 1)
+                                      ps1, pss1)
+        c2, ps2, pss2 = _conv_fs(ext, y1, w2, stride, 1, fs)
         y2, m2, i2 = ext.bn_fwd_train(c2, g2, b2, bn2.running_mean,
                                       bn2.running_var, mom, eps, True, None,
-                                      None, None)
-        c3 = ext.conv_fwd(y2, w3, 1, 0)
+                                      ps2, pss2)
+        c3, ps3, pss3 = _conv_fs(ext, y2, w3, 1, 0, fs)
         if wd is not None:
-            cd = ext.conv_fwd(x, wd, stride, 0)
+            cd, psd, pssd = _conv_fs(ext, x, wd, stride, 0, fs)
             sc, md, idn = ext.bn_fwd_train(cd, gd, bd, bnd.running_mean,
                                            bnd.running_var, mom, eps, False,
-                                           None, None, None)
+                                           None, psd, pssd)
         else:
             cd = sc = x
             md = idn = m1  # placeholders (unused)
         y3, m3, i3 = ext.bn_fwd_train(c3, g3, b3, bn3.running_mean,
                                       bn3.running_var, mom, eps, True, sc,
-                                      None, None)
+                                      ps3, pss3)
         ctx.save_for_backward(x, w1, g1, w2, g2, w3, g3, wd, gd,
                               c1, y1, c2, y2, c3, y3, cd, sc,
                               m1, i1, m2, i2, m3, i3, md, idn)
@@ -179,22 +198,23 @@ class _FusedBasicBlock(torch.autograd.Function):
                 bn1, bn2, bnd, bnprev=None, handle=None):
         ext = require_ext()
         mom, eps = bn1.momentum, bn1.eps
-        c1 = ext.conv_fwd(x, w1, stride, 1)
+        fs = _bnstats_on()
+        c1, ps1, pss1 = _conv_fs(ext, x, w1, stride, 1, fs)
         y1, m1, i1 = ext.bn_fwd_train(c1, g1, b1, bn1.running_mean,
                                       bn1.running_var, mom, eps, True, None,
-                                      None, None)
-        c2 = ext.conv_fwd(y1, w2, 1, 1)
+                                      ps1, pss1)
+        c2, ps2, pss2 = _conv_fs(ext, y1, w2, 1, 1, fs)
         if wd is not None:
-            cd = ext.conv_fwd(x, wd, stride, 0)
+            cd, psd, pssd = _conv_fs(ext, x, wd, stride, 0, fs)
             sc, md, idn = ext.bn_fwd_train(cd, gd, bd, bnd.running_mean,
                                            bnd.running_var, mom, eps, False,
-                                           None, None, None)
+                                           None, psd, pssd)
         else:
             cd = sc = x
             md = idn = m1
         y2, m2, i2 = ext.bn_fwd_train(c2, g2, b2, bn2.running_mean,
                                       bn2.running_var, mom, eps, True, sc,
-                                      None, None)
+                                      ps2, pss2)
         ctx.save_for_backward(x, w1, g1, w2, g2, wd, gd, c1, y1, c2, y2, cd,
                               sc, m1, i1, m2, i2, md, idn)
         ctx.stride = stride

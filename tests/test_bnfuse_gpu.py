@@ -173,3 +173,40 @@ def test_fused_block_bnbwd_end_to_end(monkeypatch):
         diff = ((g1[n] - g0[n]).norm() / (g0[n].norm() + 1e-6)).item()
         assert diff < 3 * max(noise, 0.02), (
             f"{n}: fused-vs-standalone {diff:.4f} vs noise floor {noise:.4f}")
+
+
+def test_conv_fwd_stats_slabs_match_tensor_sums():
+    """Forward-stats epilogue (EpiBF16FwdStats): slab column sums equal the
+    full-tensor sums of the conv output, and bn_fwd_train with the slabs
+    reproduces the unfused BN exactly."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    ext = _ext()
+    torch.manual_seed(9)
+    dtype = torch.bfloat16
+    N, C, Ko, H = 8, 64, 128, 30
+    x = _cl(torch.randn((N, C, H, H), device="cuda").to(dtype))
+    w = _cl(torch.randn((Ko, C, 3, 3), device="cuda").to(dtype) * 0.05)
+    y, ps, pss = ext.conv_fwd_stats(x, w, 1, 1)
+    assert ps.numel(), "fusable shape routed to the fallback"
+    y_ref = ext.conv_fwd(x, w, 1, 1)
+    torch.testing.assert_close(y.float(), y_ref.float(), rtol=0, atol=0)
+    yf = y.float()
+    torch.testing.assert_close(ps.sum(0), yf.sum(dim=(0, 2, 3)), rtol=1e-4,
+                               atol=0.5)
+    torch.testing.assert_close(pss.sum(0), (yf * yf).sum(dim=(0, 2, 3)),
+                               rtol=1e-4, atol=0.5)
+
+    gamma = torch.randn(Ko, device="cuda").to(dtype)
+    beta = torch.randn(Ko, device="cuda").to(dtype)
+    rm1 = torch.zeros(Ko, device="cuda")
+    rv1 = torch.ones(Ko, device="cuda")
+    rm2, rv2 = rm1.clone(), rv1.clone()
+    o1, m1, i1 = ext.bn_fwd_train(y, gamma, beta, rm1, rv1, 0.9, 1e-5, True,
+                                  None, ps, pss)
+    o2, m2, i2 = ext.bn_fwd_train(y, gamma, beta, rm2, rv2, 0.9, 1e-5, True,
+                                  None, None, None)
+    torch.testing.assert_close(m1, m2, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(i1, i2, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(o1.float(), o2.float(), rtol=0.01, atol=0.01)
+    torch.testing.assert_close(rm1, rm2, rtol=1e-5, atol=1e-5)
