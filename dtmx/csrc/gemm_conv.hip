@@ -607,24 +607,30 @@ struct WgradXcolB {
   }
 };
 
-template <int NJ, class PA, class PB, class EPI>
+// WM = waves along M (wave grid WM x (4/WM)): WM=2 is the square 128xBN
+// tile; WM=1 is the flat 64x(NJ*64) tile for small-M work — conv wgrad's
+// M = K_out is 64 on ResNet's narrow layers, where the 128-row tile wastes
+// HALF its MFMA work on zero-padded rows (measured 222 TF on l1.conv2).
+template <int NJ, int WM, class PA, class PB, class EPI>
 __launch_bounds__(256, 2) __global__
 void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
                     uint32_t tiles_n, uint32_t kt_per_slice) {
   using elem_t = typename PA::elem;
   using V8 = typename E8<elem_t>::v8;
-  constexpr uint32_t BN = NJ * 32;
+  constexpr uint32_t WN = 4 / WM;
+  constexpr uint32_t BM = WM * 64;
+  constexpr uint32_t BN = WN * NJ * 16;
   const uint32_t kt0 = blockIdx.y * kt_per_slice;
   const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
   if (kt0 >= ktiles_total) return;
   // 16-B alignment: ds_read_b64_tr_b16 at a misaligned address silently
   // returns the 8-aligned address's data (G17)
-  __shared__ __attribute__((aligned(16))) elem_t smem[2][64 * (128 + BN)];
-  constexpr uint32_t B_OFF = 64 * 128;
+  __shared__ __attribute__((aligned(16))) elem_t smem[2][64 * (BM + BN)];
+  constexpr uint32_t B_OFF = 64 * BM;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
   const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  const uint32_t bm = (bid / tiles_n) * 128, bn = (bid % tiles_n) * BN;
+  const uint32_t bm = (bid / tiles_n) * BM, bn = (bid % tiles_n) * BN;
 
   // staging: A rows are 256 B (128 m), chunk = 16 kd-rows per 4 KiB;
   //          B rows are BN*2 B, NJ 4-KiB chunks total.
@@ -633,31 +639,38 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   // a linear image (2-way conflict, SQ_LDS_BANK_CONFLICT ~3e9/step measured)
   // — flipping 16 columns on odd (kd>>3) separates them. Applied on the
   // glds SOURCE (lane-linear dest, rule 21) and un-applied on the reads.
+  // ITS_A/ITS_B glds instructions per tile: each covers 256 lanes x 16 B
+  constexpr uint32_t PPR_A = BM / 8;   // 16-B pieces per kd-row of A
+  constexpr uint32_t PPR_B = BN / 8;
+  constexpr uint32_t ITS_A = 64 * PPR_A / 256;
+  constexpr uint32_t ITS_B = 64 * PPR_B / 256;
   auto stage = [&](int buf, uint32_t kt) {
     const uint32_t kd0 = kt * 64;
 #pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      uint32_t krow = it * 16 + (t >> 4);
-      uint32_t piece = (t & 15) ^ (((krow >> 3) & 1) << 1);
+    for (uint32_t it = 0; it < ITS_A; ++it) {
+      uint32_t krow = it * (256 / PPR_A) + t / PPR_A;
+      uint32_t piece = (t % PPR_A) ^ (((krow >> 3) & 1) << 1);
       glds16(pa.addr(kd0 + krow, bm / 8 + piece),
              &smem[buf][it * 2048 + wave * 512]);
     }
 #pragma unroll
-    for (int it = 0; it < NJ; ++it) {
-      uint32_t krow = it * (256 / (NJ * 4)) + t / (NJ * 4);
-      uint32_t piece = (t % (NJ * 4)) ^ (((krow >> 3) & 1) << 1);
+    for (uint32_t it = 0; it < ITS_B; ++it) {
+      uint32_t krow = it * (256 / PPR_B) + t / PPR_B;
+      uint32_t piece = (t % PPR_B) ^ (((krow >> 3) & 1) << 1);
       glds16(pb.addr(kd0 + krow, bn / 8 + piece),
              &smem[buf][B_OFF + it * 2048 + wave * 512]);
     }
   };
 
-  const uint32_t wr = (wave >> 1) * 64, wc = (wave & 1) * NJ * 16;
+  const uint32_t wr = (wave / WN) * 64, wc = (wave % WN) * NJ * 16;
   f32x4 acc[4][NJ] = {};
 
-  constexpr int G = 4 + NJ;
+  constexpr int G = (int)(ITS_A + ITS_B);
   auto wait_tile = [&](bool one_in_flight) {
     if (one_in_flight) {
-      if constexpr (NJ == 4)
+      if constexpr (G == 10)
+        asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+      else if constexpr (G == 8)
         asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       else
         asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
@@ -692,8 +705,8 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
         unsigned a0, a1, a2, a3, a4, a5, a6, a7;
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          uint32_t lo = (kg)*128 + ((wr + i * 16 + cq) ^ (((kg >> 3) & 1) << 4));
-          uint32_t hi = (kg + 4) * 128 +
+          uint32_t lo = (kg)*BM + ((wr + i * 16 + cq) ^ (((kg >> 3) & 1) << 4));
+          uint32_t hi = (kg + 4) * BM +
                         ((wr + i * 16 + cq) ^ ((((kg + 4) >> 3) & 1) << 4));
           unsigned alo = (unsigned)(uintptr_t)&smem[cur][lo];
           unsigned ahi = (unsigned)(uintptr_t)&smem[cur][hi];
@@ -1015,16 +1028,25 @@ static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
   uint32_t ktiles_total = ceil_div(K, 64);
   splitk = std::min(splitk, ktiles_total);
   uint32_t kt_per = ceil_div(ktiles_total, splitk);
+  if (M <= 64 && N > 64) {
+    // flat 64x256 tile: ResNet's K_out=64 wgrads waste half the 128-row
+    // tile on zero rows (see gemm_nt_kernel WM doc)
+    uint32_t tiles_n = ceil_div(N, 256);
+    dim3 grid(tiles_n, ceil_div(ktiles_total, kt_per));
+    gemm_nt_kernel<4, 1, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    return;
+  }
   uint32_t tiles_m = ceil_div(M, 128);
   if (N <= 64) {
     uint32_t tiles_n = ceil_div(N, 64);
     dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-    gemm_nt_kernel<2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+    gemm_nt_kernel<2, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
         pa, pb, epi, ktiles_total, tiles_n, kt_per);
   } else {
     uint32_t tiles_n = ceil_div(N, 128);
     dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-    gemm_nt_kernel<4, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+    gemm_nt_kernel<4, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
         pa, pb, epi, ktiles_total, tiles_n, kt_per);
   }
 }
@@ -1463,7 +1485,10 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
     if (C % 8 == 0 && Ko % 8 == 0) {
       // transpose-free: dy and the im2col view of x are read NHWC-native by
       // the contraction-major (NT) kernel
-      uint32_t tiles_mn = ceil_div(Ko, 128) * ceil_div(RSC, RSC <= 64 ? 64 : 128);
+      uint32_t tiles_mn =
+          (Ko <= 64 && RSC > 64)
+              ? ceil_div(RSC, 256)  // flat 64x256 tile (launch_gemm_nt)
+              : ceil_div(Ko, 128) * ceil_div(RSC, RSC <= 64 ? 64 : 128);
       uint32_t ktiles = ceil_div(M, 64);
       uint32_t splitk = std::max<uint32_t>(
           1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
