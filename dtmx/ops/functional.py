@@ -545,3 +545,43 @@ def embedding_row_sparse_grad(dy: torch.Tensor, idx: torch.Tensor,
                            device=dy.device)
         vals.index_add_(0, inverse, dy2.float())
     return RowSparse(rows, vals.to(dy.dtype), (vocab, D))
+
+
+# ------------------------------------------------------------ deconvolution
+
+class _DeconvNHWC(torch.autograd.Function):
+    """Transposed convolution (reference src/operator/nn/deconvolution.cu) —
+    no new kernels: deconv forward IS conv dgrad, its data-grad IS conv
+    forward, and its weight-grad is conv wgrad with the operands swapped.
+    Weight layout (Cin, Cout, R, S) channels_last, mxnet convention."""
+
+    @staticmethod
+    def forward(ctx, x, w, stride, padding):
+        ext = require_ext()
+        ctx.save_for_backward(x, w)
+        ctx.stride, ctx.padding = stride, padding
+        R = w.shape[2]
+        H = (x.shape[2] - 1) * stride - 2 * padding + R
+        W = (x.shape[3] - 1) * stride - 2 * padding + R
+        return ext.conv_dgrad(x, w, stride, padding, H, W, None)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = (ext.conv_fwd(dy, w, ctx.stride, ctx.padding)
+              if ctx.needs_input_grad[0] else None)
+        dw = None
+        if ctx.needs_input_grad[1]:
+            R = w.shape[2]
+            dw = ext.conv_wgrad(dy, x, R, R, ctx.stride, ctx.padding)
+        return dx, dw, None, None
+
+
+def deconv2d(x, w, stride: int = 1, padding: int = 0):
+    """Transposed conv, NHWC HIP path on GPU; torch fallback elsewhere.
+    `w` is (Cin, Cout, R, S)."""
+    if _use_hip(x, op="deconv") and w.shape[1] % 8 == 0 and w.shape[0] % 8 == 0:
+        return _DeconvNHWC.apply(x, w, stride, padding)
+    return F.conv_transpose2d(x, w, None, stride, padding)

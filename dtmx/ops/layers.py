@@ -205,3 +205,19 @@ class Embedding(nn.Module):
 
     def extra_repr(self):
         return f"{self.input_dim}, {self.output_dim}"
+
+
+class Deconv2dNHWC(nn.Module):
+    """Transposed convolution (reference mx.sym.Deconvolution)."""
+
+    def __init__(self, in_channels, out_channels, kernel_size, stride=1,
+                 padding=0):
+        super().__init__()
+        self.stride = stride
+        self.padding = padding
+        w = torch.empty(in_channels, out_channels, kernel_size, kernel_size)
+        nn.init.kaiming_normal_(w, mode="fan_in", nonlinearity="relu")
+        self.weight = nn.Parameter(w)
+
+    def forward(self, x):
+        return DF.deconv2d(x, self.weight, self.stride, self.padding)

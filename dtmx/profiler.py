@@ -42,5 +42,34 @@ def set_state(state: str = "stop"):
             logging.info("profiler trace written to %s", _config["filename"])
 
 
-def dump():
+def dumps(reset: bool = False, sort_by: str = "self_device_time",
+          max_rows: int = 40) -> str:
+    """Aggregate per-op statistics table (reference profiler
+    aggregate_stats.cc / mx.profiler.dumps): op name, call count, total and
+    mean device/host time. Call between set_state('run') and 'stop', or
+    after 'stop' on the captured events."""
+    if _prof is None:
+        return "(profiler not running; set_state('run') first)"
+    events = _prof.key_averages()
+    key = {"self_device_time": "self_device_time_total",
+           "device_time": "device_time_total",
+           "cpu_time": "self_cpu_time_total"}.get(sort_by, sort_by)
+    rows = sorted(events, key=lambda e: -getattr(e, key, 0))[:max_rows]
+    out = [f"{'op':60s} {'calls':>7s} {'dev_ms':>10s} {'mean_us':>9s} "
+           f"{'cpu_ms':>10s}"]
+    for e in rows:
+        dev_ms = getattr(e, "self_device_time_total", 0) / 1e3
+        cpu_ms = e.self_cpu_time_total / 1e3
+        mean_us = (getattr(e, "self_device_time_total", 0) / e.count
+                   if e.count else 0.0)
+        out.append(f"{e.key[:60]:60s} {e.count:7d} {dev_ms:10.3f} "
+                   f"{mean_us:9.1f} {cpu_ms:10.3f}")
+    return "\n".join(out)
+
+
+def dump(aggregate: bool = False):
+    """Stop and write the chrome trace; with aggregate=True also log the
+    per-op summary table (reference MXDumpProfile + aggregate stats)."""
+    if aggregate and _prof is not None:
+        logging.info("\n%s", dumps())
     set_state("stop")

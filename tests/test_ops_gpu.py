@@ -656,3 +656,32 @@ def test_linear_gemms_exact_on_integer_inputs():
     dw = ext.linear_wgrad(dy, x)
     torch.testing.assert_close(dw.double(), bf(dy.double().T @ x.double()),
                                rtol=0, atol=0)
+
+
+@pytest.mark.gpu
+def test_deconv2d_matches_torch():
+    """Deconvolution built from the conv kernels (fwd = dgrad, etc.)."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from dtmx.ops import functional as DF
+
+    torch.manual_seed(0)
+    x = (torch.randn(4, 16, 9, 9, device="cuda").to(torch.bfloat16)
+         .contiguous(memory_format=torch.channels_last).requires_grad_())
+    w = (torch.randn(16, 32, 4, 4, device="cuda").to(torch.bfloat16)
+         .contiguous(memory_format=torch.channels_last).requires_grad_() )
+    y = DF.deconv2d(x, w, 2, 1)
+    assert y.shape == (4, 32, 18, 18)
+    ref = torch.nn.functional.conv_transpose2d(x.detach().float(),
+                                               w.detach().float(), None, 2, 1)
+    torch.testing.assert_close(y.float(), ref, rtol=0.02,
+                               atol=0.02 * ref.abs().mean().item())
+    g = torch.randn_like(y)
+    y.backward(g)
+    xr = x.detach().float().requires_grad_()
+    wr = w.detach().float().requires_grad_()
+    torch.nn.functional.conv_transpose2d(xr, wr, None, 2, 1).backward(g.float())
+    torch.testing.assert_close(x.grad.float(), xr.grad, rtol=0.03,
+                               atol=0.05 * xr.grad.abs().mean().item())
+    torch.testing.assert_close(w.grad.float(), wr.grad, rtol=0.03,
+                               atol=0.05 * wr.grad.abs().mean().item())
