@@ -21,6 +21,11 @@ def add_data_args(parser):
     data.add_argument("--num-examples", type=int, default=1281167)
     data.add_argument("--image-shape", type=str, default="3,224,224")
     data.add_argument("--data-nthreads", type=int, default=4)
+    # augmentation flags (reference common/data.py add_data_aug_args subset)
+    data.add_argument("--random-crop", type=int, default=1)
+    data.add_argument("--random-mirror", type=int, default=1)
+    data.add_argument("--resize", type=int, default=0,
+                      help="resize shorter side before crop (0 = off)")
     return data
 
 
@@ -42,6 +47,25 @@ def get_rec_iter(args, kv):
     image_shape = tuple(int(x) for x in args.image_shape.split(","))
     if getattr(args, "benchmark", 0) or not args.data_train:
         return _synthetic_pair(args, kv, image_shape)
+    if args.data_train.endswith(".rec") and os.path.exists(args.data_train):
+        from dtmx.io import ImageRecordIter
+
+        train = ImageRecordIter(
+            args.data_train, image_shape, args.batch_size, shuffle=True,
+            part_index=kv.rank, num_parts=kv.num_workers,
+            preprocess_threads=args.data_nthreads,
+            rand_crop=bool(getattr(args, "random_crop", 1)),
+            rand_mirror=bool(getattr(args, "random_mirror", 1)),
+            resize=getattr(args, "resize", 0),
+        )
+        val = None
+        if args.data_val and os.path.exists(args.data_val):
+            val = ImageRecordIter(
+                args.data_val, image_shape, args.batch_size,
+                preprocess_threads=args.data_nthreads,
+                resize=getattr(args, "resize", 0),
+            )
+        return train, val
     if args.data_train.endswith(".npz") and os.path.exists(args.data_train):
         blob = np.load(args.data_train)
         train = NDArrayIter({"data": blob["x"]}, {"softmax_label": blob["y"]},
