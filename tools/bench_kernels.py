@@ -36,12 +36,28 @@ def nhwc(t):
 
 def main():
     torch.manual_seed(0)
+    # 0) 256^2 8-phase kernel: correctness then A/B vs the 128^2 kernel
+    a = torch.randn(512, 512, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(512, 512, dtype=torch.bfloat16, device=DEV)
+    ref = (a.float() @ b.float().T)
+    got = ext.gemm256_nt(a, b).float()
+    err = (got - ref).abs().max().item() / ref.abs().max().item()
+    print(f"gemm256 refcheck 512^3: rel max err {err:.4f}")
+    a = torch.randint(-3, 4, (768, 640), device=DEV).to(torch.bfloat16)
+    b = torch.randint(-3, 4, (520, 640), device=DEV).to(torch.bfloat16)
+    got = ext.gemm256_nt(a, b).double()
+    ref = (a.double() @ b.double().T)
+    exact = (got == ref.to(torch.bfloat16).double()).all().item()
+    print(f"gemm256 integer-exact 768x520x640 (tails): {exact}")
+
     # 1) square dense GEMM peak of the kernel
     for M, N, K in [(4096, 4096, 4096), (8192, 8192, 8192)]:
         a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
         b = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
         s = timeit(lambda: ext.linear_fwd(a, b, None))
-        print(f"dense {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
+        print(f"dense128 {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
+        s = timeit(lambda: ext.gemm256_nt(a, b))
+        print(f"gemm256  {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
 
     # 2) resnet-shaped dense (same GEMM the conv would do, no gather)
     for M, N, K in [(401408, 64, 576), (401408, 128, 256), (100352, 256, 1152),
