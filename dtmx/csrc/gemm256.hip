@@ -34,12 +34,17 @@ struct Dense256P {
   }
 };
 
-// st_16x32 swizzle within a 16 KiB half image ([128 rows][128 B])
+// swizzle within a 16 KiB half image ([128 rows][128 B]). Default: the
+// row-parity XOR of the proven 128^2 kernel (byte ^= (row&7)<<4 — measured
+// 0.17% bank-conflict cycles there vs 2.5% for st_16x32 on this layout);
+// DTMX_G256_ST16 compiles the guide's st_16x32 for A/B.
+template <int ST16>
 __device__ __forceinline__ uint32_t swz256(uint32_t byte_off) {
-  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  if constexpr (ST16) return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  return byte_off ^ (((byte_off >> 7) & 7) << 4);
 }
 
-template <class PA, class PB, typename elem_t, int NOBAR1 = 0>
+template <class PA, class PB, typename elem_t, int NOBAR1 = 0, int ST16 = 0>
 __launch_bounds__(512, 2) __global__
 void gemm256_kernel(PA pa, PB pb, elem_t* __restrict__ c, uint32_t M,
                     uint32_t N, uint32_t ktiles, uint32_t tiles_n) {
@@ -65,7 +70,7 @@ void gemm256_kernel(PA pa, PB pb, elem_t* __restrict__ c, uint32_t M,
 #pragma unroll
     for (uint32_t g = 0; g < 2; ++g) {
       uint32_t off = (g * 8 + wave) * 1024 + lane * 16;  // bytes in half
-      uint32_t lb = swz256(off);
+      uint32_t lb = swz256<ST16>(off);
       uint32_t row = lb >> 7, kb = lb & 127;
       uint32_t k8 = kb >> 4;
       const void* p = part < 2 ? pb.addr(bn + part * 128 + row, k8)
@@ -139,7 +144,7 @@ void gemm256_kernel(PA pa, PB pb, elem_t* __restrict__ c, uint32_t M,
             uint32_t row = wc_local + j * 16 + (lane & 15);
             uint32_t kbyte = kk * 64 + ((lane >> 4) << 4);
             bf[j * 2 + kk] = *(const V8*)((const char*)&smem[cur][b_slot][0] +
-                                          swz256(row * 128 + kbyte));
+                                          swz256<ST16>(row * 128 + kbyte));
           }
       }
       V8 af[2][2];  // i in {2q, 2q+1} x kk
@@ -150,7 +155,7 @@ void gemm256_kernel(PA pa, PB pb, elem_t* __restrict__ c, uint32_t M,
           uint32_t row = (2 * q + ii) * 16 + (lane & 15);
           uint32_t kbyte = kk * 64 + ((lane >> 4) << 4);
           af[ii][kk] = *(const V8*)((const char*)&smem[cur][a_slot][0] +
-                                    swz256(row * 128 + kbyte));
+                                    swz256<ST16>(row * 128 + kbyte));
         }
       stage_stream(7 + P);
       if (q == 0)  // 12-read phase: start draining before the barrier
@@ -228,18 +233,28 @@ at::Tensor gemm256_nt(const at::Tensor& a, const at::Tensor& b) {
     // phase-aligned waves without the pre-MFMA barrier measured +7%
     // (1126 vs 1048 TF @8192^3); DTMX_G256_BAR1=1 restores the two-barrier
     // phase for A/B
-    static const bool bar1 = [] {
-      const char* v = getenv("DTMX_G256_BAR1");
-      return v && v[0] == '1';
+    static const int variant = [] {
+      const char* b = getenv("DTMX_G256_BAR1");
+      const char* w = getenv("DTMX_G256_ST16");
+      return (b && b[0] == '1' ? 0 : 1) | (w && w[0] == '1' ? 2 : 0);
     }();
-    if (!bar1)
-      gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 1>
-          <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles,
-                                tiles_n);
-    else
-      gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 0>
-          <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles,
-                                tiles_n);
+    switch (variant) {
+      case 0:
+        gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 0, 0>
+            <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles, tiles_n);
+        break;
+      case 1:
+        gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 1, 0>
+            <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles, tiles_n);
+        break;
+      case 2:
+        gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 0, 1>
+            <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles, tiles_n);
+        break;
+      default:
+        gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 1, 1>
+            <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles, tiles_n);
+    }
     return y;
   });
   return at::Tensor();

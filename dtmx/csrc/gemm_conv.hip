@@ -320,7 +320,7 @@ struct EpiBF16FwdStats {
 // fused-BN epilogues: threads t ≡ nc (mod NC8) hold partials for channel
 // group nc; after the tree, thread t < NC8 writes its 8 channels of the
 // block's own slab row. `red` is 8 KiB of post-compute LDS.
-template <int NC8>
+template <int NC8, int THREADS = 256>
 __device__ __forceinline__ void epi_colreduce_write(
     float* red, const float (&v)[8], float* slab, uint32_t bm, uint32_t bn,
     uint32_t N, uint32_t t) {
@@ -328,7 +328,7 @@ __device__ __forceinline__ void epi_colreduce_write(
 #pragma unroll
   for (int e = 0; e < 8; ++e) red[t * 8 + e] = v[e];
   __syncthreads();
-  for (uint32_t off = 128; off >= NC8; off >>= 1) {
+  for (uint32_t off = THREADS / 2; off >= NC8; off >>= 1) {
     if (t < off) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) red[t * 8 + e] += red[(t + off) * 8 + e];
@@ -562,6 +562,256 @@ void gemm_tn_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     }
   } else {
     epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
+  }
+}
+
+// ------------------------------------------ 256^2 8-phase dense TN kernel
+// The deep-pipelined big-tile path (cdna_hip_programming.md §5 "256² 8-phase
+// template"; standalone probe: gemm256.hip — 1200 TF @8192³ vs 892 for the
+// 128² structure). Dense providers only (DenseP both sides); carries the
+// full fused-epilogue family (EpiBF16 / EpiBnBwd / EpiBF16FwdStats) so the
+// production 1x1-conv and FC GEMMs can route here. Schedule: per K-tile,
+// phase 0 reads ALL B fragments to registers (+ first A quarter), phases
+// 1-3 one A quarter each; one half-tile of the {B0,B1,A0,A1} stream staged
+// per phase with a 7-half prologue and vmcnt(6) at tile boundaries only.
+template <class PA, class PB, class EPI>
+__launch_bounds__(512, 2) __global__
+void gemm256f_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles, uint32_t tiles_n) {
+  using elem_t = typename PA::elem;
+  using V8 = typename E8<elem_t>::v8;
+  __shared__ __attribute__((aligned(16))) elem_t smem[2][4][8192];
+  const uint32_t t = threadIdx.x;
+  const uint32_t wave = t >> 6, lane = t & 63;
+  const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const uint32_t bm = (bid / tiles_n) * 256, bn = (bid % tiles_n) * 256;
+  const uint32_t am_half = wave >> 2;
+  const uint32_t wc = (wave & 3) * 64;
+
+  // row-parity source swizzle of the 128² kernel (beat st_16x32 by ~2% with
+  // 0.2% vs 2.5% bank-conflict cycles on this layout)
+  auto swz = [](uint32_t b) { return b ^ (((b >> 7) & 7) << 4); };
+
+  // branch-free staging: per-lane source pointers precomputed, stride 0 for
+  // OOB rows (they re-read the zero page); K%64==0 enforced by the caller
+  const elem_t* src0[4][2];
+  size_t sstep[4][2];
+#pragma unroll
+  for (uint32_t part = 0; part < 4; ++part)
+#pragma unroll
+    for (uint32_t g = 0; g < 2; ++g) {
+      uint32_t off = (g * 8 + wave) * 1024 + lane * 16;
+      uint32_t lb = swz(off);
+      uint32_t row = lb >> 7, kb = lb & 127;
+      uint32_t k = (kb >> 4) * 8;
+      if (part < 2) {
+        uint32_t m = bn + part * 128 + row;
+        bool oob = m >= pb.M;
+        src0[part][g] = oob ? pb.zero : pb.base + (size_t)m * pb.ld + k;
+        sstep[part][g] = oob ? 0 : 64;
+      } else {
+        uint32_t m = bm + (part - 2) * 128 + row;
+        bool oob = m >= pa.M;
+        src0[part][g] = oob ? pa.zero : pa.base + (size_t)m * pa.ld + k;
+        sstep[part][g] = oob ? 0 : 64;
+      }
+    }
+  auto stage_part = [&](auto part_c, uint32_t buf, uint32_t kt) {
+    constexpr uint32_t part = decltype(part_c)::value;
+#pragma unroll
+    for (uint32_t g = 0; g < 2; ++g)
+      glds16(src0[part][g] + (size_t)kt * sstep[part][g],
+             &smem[buf][part][(g * 8 + wave) * 512]);
+  };
+  const uint32_t total_halves = ktiles * 4;
+  auto stage_stream = [&](uint32_t h) {
+    if (h >= total_halves) return;
+    uint32_t kt = h >> 2, buf = kt & 1;
+    switch (h & 3) {
+      case 0: stage_part(std::integral_constant<uint32_t, 0>{}, buf, kt); break;
+      case 1: stage_part(std::integral_constant<uint32_t, 1>{}, buf, kt); break;
+      case 2: stage_part(std::integral_constant<uint32_t, 2>{}, buf, kt); break;
+      default: stage_part(std::integral_constant<uint32_t, 3>{}, buf, kt);
+    }
+  };
+  auto wait_vm = [&](uint32_t n) {
+    if (n >= 6)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if (n == 4)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (n == 2)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  };
+
+  uint32_t issued = 0;
+  for (; issued < min(4u, total_halves); ++issued) stage_stream(issued);
+  wait_vm(total_halves > 4 ? 4 : 0);
+  for (; issued < min(7u, total_halves); ++issued) stage_stream(issued);
+  wait_vm(2 * (issued - min(4u, total_halves)));
+  __builtin_amdgcn_s_barrier();
+
+  f32x4 acc[8][4] = {};
+  const uint32_t a_slot = 2 + am_half;
+  const uint32_t b_slot = wc >> 7;
+  const uint32_t wc_local = wc & 127;
+
+  V8 bf[8];
+  uint32_t P = 0;
+  for (uint32_t kt = 0; kt < ktiles; ++kt) {
+    const uint32_t cur = kt & 1;
+#pragma unroll
+    for (uint32_t q = 0; q < 4; ++q, ++P) {
+      if (q == 0) {
+#pragma unroll
+        for (uint32_t j = 0; j < 4; ++j)
+#pragma unroll
+          for (uint32_t kk = 0; kk < 2; ++kk) {
+            uint32_t row = wc_local + j * 16 + (lane & 15);
+            uint32_t kbyte = kk * 64 + ((lane >> 4) << 4);
+            bf[j * 2 + kk] = *(const V8*)((const char*)&smem[cur][b_slot][0] +
+                                          swz(row * 128 + kbyte));
+          }
+      }
+      V8 af[2][2];
+#pragma unroll
+      for (uint32_t ii = 0; ii < 2; ++ii)
+#pragma unroll
+        for (uint32_t kk = 0; kk < 2; ++kk) {
+          uint32_t row = (2 * q + ii) * 16 + (lane & 15);
+          uint32_t kbyte = kk * 64 + ((lane >> 4) << 4);
+          af[ii][kk] = *(const V8*)((const char*)&smem[cur][a_slot][0] +
+                                    swz(row * 128 + kbyte));
+        }
+      stage_stream(7 + P);
+      if (q == 0)
+        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (uint32_t kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (uint32_t ii = 0; ii < 2; ++ii)
+#pragma unroll
+          for (uint32_t j = 0; j < 4; ++j)
+            acc[2 * q + ii][j] = E8<elem_t>::mfma(af[ii][kk], bf[j * 2 + kk],
+                                                  acc[2 * q + ii][j]);
+      __builtin_amdgcn_s_setprio(0);
+      if (q == 3) {
+        uint32_t issued_now = min(total_halves, 7 + P + 1);
+        uint32_t needed = min(total_halves, 4 * (kt + 2));
+        wait_vm(2 * (issued_now - needed));
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // LDS-staged epilogue in two 128-row half passes (the C tile is 128 KiB =
+  // the whole LDS; ct of one half = 64 KiB). Same chunk/tree machinery as
+  // the 128² kernel, scaled to 512 threads / 32 chunks per row.
+  elem_t* ct = &smem[0][0][0];  // [128 rows][256 cols] per pass
+  constexpr uint32_t NC8 = 32;
+  const uint32_t nc = t % NC8;
+  const uint32_t n0 = bn + nc * 8;
+#pragma unroll
+  for (uint32_t h = 0; h < 2; ++h) {
+    __syncthreads();  // no glds in flight after the final boundary wait
+    if (am_half == h) {
+#pragma unroll
+      for (uint32_t i = 0; i < 8; ++i)
+#pragma unroll
+        for (uint32_t j = 0; j < 4; ++j) {
+          uint32_t col = wc + j * 16 + (lane & 15);
+#pragma unroll
+          for (uint32_t r = 0; r < 4; ++r) {
+            uint32_t row = i * 16 + ((lane >> 4) << 2) + r;
+            ct[row * 256 + col] = (elem_t)acc[i][j][r];
+          }
+        }
+    }
+    __syncthreads();
+    const uint32_t bmh = bm + h * 128;
+    constexpr uint32_t CHUNKS = 128 * NC8;
+    if constexpr (EPI::kBnBwd) {
+      float mean8[8], inv8[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        uint32_t n = n0 + e;
+        mean8[e] = n < epi.N ? epi.bnb_mean[n] : 0.f;
+        inv8[e] = n < epi.N ? epi.bnb_invstd[n] : 0.f;
+      }
+      float db[8] = {}, dg[8] = {};
+      for (uint32_t idx = t; idx < CHUNKS; idx += 512) {
+        uint32_t row = idx / NC8;
+        uint32_t m = bmh + row;
+        if (m >= epi.M || n0 >= epi.N) continue;
+        V8 v = *(const V8*)(ct + row * 256 + nc * 8);
+        size_t off = (size_t)m * epi.N + n0;
+        uint32_t rem = epi.N - n0;
+        if (rem >= 8) {
+          V8 a{};
+          if (epi.acc) a = *(const V8*)(epi.acc + off);
+          V8 yv = *(const V8*)(epi.bnb_y + off);
+          V8 xv = *(const V8*)(epi.bnb_x + off);
+          V8 g;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            float f = (float)v[e] + (epi.acc ? (float)a[e] : 0.f);
+            if ((float)yv[e] <= 0.f) f = 0.f;
+            g[e] = (elem_t)f;
+            db[e] += f;
+            dg[e] += f * ((float)xv[e] - mean8[e]) * inv8[e];
+          }
+          *(V8*)(epi.c + off) = g;
+        } else {
+          for (uint32_t e = 0; e < rem; ++e) {
+            float f = (float)v[e] + (epi.acc ? (float)epi.acc[off + e] : 0.f);
+            if ((float)epi.bnb_y[off + e] <= 0.f) f = 0.f;
+            epi.c[off + e] = (elem_t)f;
+            db[e] += f;
+            dg[e] += f * ((float)epi.bnb_x[off + e] - mean8[e]) * inv8[e];
+          }
+        }
+      }
+      float* red = (float*)ct;  // chunk reads done after the next barrier
+      epi_colreduce_write<NC8, 512>(red, db, epi.bnb_pdb, bmh, bn, epi.N, t);
+      epi_colreduce_write<NC8, 512>(red, dg, epi.bnb_pdg, bmh, bn, epi.N, t);
+    } else if constexpr (EPI::kFwdStats) {
+      float s[8] = {}, ss[8] = {};
+      for (uint32_t idx = t; idx < CHUNKS; idx += 512) {
+        uint32_t row = idx / NC8;
+        uint32_t m = bmh + row;
+        if (m >= epi.M || n0 >= epi.N) continue;
+        V8 v = *(const V8*)(ct + row * 256 + nc * 8);
+        size_t off = (size_t)m * epi.N + n0;
+        uint32_t rem = epi.N - n0;
+        if (rem >= 8) {
+          *(V8*)(epi.c + off) = v;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            float f = (float)v[e];
+            s[e] += f;
+            ss[e] += f * f;
+          }
+        } else {
+          for (uint32_t e = 0; e < rem; ++e) {
+            epi.c[off + e] = v[e];
+            float f = (float)v[e];
+            s[e] += f;
+            ss[e] += f * f;
+          }
+        }
+      }
+      float* red = (float*)ct;
+      epi_colreduce_write<NC8, 512>(red, s, epi.bn_psum, bmh, bn, epi.N, t);
+      epi_colreduce_write<NC8, 512>(red, ss, epi.bn_psumsq, bmh, bn, epi.N, t);
+    } else {
+      for (uint32_t idx = t; idx < CHUNKS; idx += 512) {
+        uint32_t row = idx / NC8;
+        epi.store_chunk(bmh + row, bn + (idx % NC8) * 8,
+                        *(const V8*)(ct + row * 256 + (idx % NC8) * 8));
+      }
+    }
   }
 }
 
@@ -999,11 +1249,38 @@ static hipStream_t cur_stream() {
                                 (t).scalar_type() == at::kHalf),             \
               #t " must be a CUDA bf16/fp16 tensor")
 
+// route a dense-by-dense GEMM through the 256² kernel when the shape
+// qualifies (big M/N, pipelined K, 128-row slab granularity preserved)
+template <typename elem_t, class EPI>
+static bool try_gemm256(const DenseP<elem_t>& pa, const DenseP<elem_t>& pb,
+                        const EPI& epi, uint32_t M, uint32_t N, uint32_t K,
+                        uint32_t splitk) {
+  if constexpr (!EPI::kLdsStage) {
+    return false;
+  } else {
+    static const bool off = env_flag("DTMX_DISABLE_GEMM256");
+    if (off || splitk > 1) return false;
+    if (M < 512 || N < 192 || K < 128 || K % 64 != 0) return false;
+    if (pa.K != K || pb.K != K) return false;
+    uint32_t tiles_m = ceil_div(M, 256), tiles_n = ceil_div(N, 256);
+    if (tiles_m * tiles_n < 160) return false;  // underfilled grid
+    dim3 grid(tiles_m * tiles_n);
+    gemm256f_kernel<DenseP<elem_t>, DenseP<elem_t>, EPI>
+        <<<grid, 512, 0, cur_stream()>>>(pa, pb, epi, K / 64, tiles_n);
+    return true;
+  }
+}
+
 // small grids underfill 256 CUs: callers pass want_splitk=true to let the
 // launcher split K into an fp32 atomic buffer (the caller casts back).
 template <class PA, class PB, class EPI>
 static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
                         uint32_t N, uint32_t K, uint32_t splitk = 1) {
+  using elem = typename PA::elem;
+  if constexpr (std::is_same_v<PA, DenseP<elem>> &&
+                std::is_same_v<PB, DenseP<elem>>) {
+    if (try_gemm256(pa, pb, epi, M, N, K, splitk)) return;
+  }
   uint32_t ktiles_total = ceil_div(K, 64);
   splitk = std::min(splitk, ktiles_total);
   uint32_t kt_per = ceil_div(ktiles_total, splitk);
