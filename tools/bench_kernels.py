@@ -55,9 +55,9 @@ def main():
         a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
         b = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
         s = timeit(lambda: ext.linear_fwd(a, b, None))
-        print(f"dense128 {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
+        print(f"linear_fwd(routed) {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
         s = timeit(lambda: ext.gemm256_nt(a, b))
-        print(f"gemm256  {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
+        print(f"gemm256_direct     {M}x{N}x{K}: {tf(2*M*N*K, s):7.1f} TF  {s*1e3:.3f} ms")
 
     # 2) resnet-shaped dense (same GEMM the conv would do, no gather)
     for M, N, K in [(401408, 64, 576), (401408, 128, 256), (100352, 256, 1152),
