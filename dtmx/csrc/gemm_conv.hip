@@ -861,7 +861,12 @@ struct WgradXcolB {
 // tile; WM=1 is the flat 64x(NJ*64) tile for small-M work — conv wgrad's
 // M = K_out is 64 on ResNet's narrow layers, where the 128-row tile wastes
 // HALF its MFMA work on zero-padded rows (measured 222 TF on l1.conv2).
-template <int NJ, int WM, class PA, class PB, class EPI>
+// DEPTH: LDS K-tile ring depth. 2 = classic double buffer (ONE tile in
+// flight across the compute of the next — the ~550-cycle MFMA body of a
+// tile does not cover the ~900-cycle HBM round trip, measured 35% parked
+// cycles on wgrad). 3 = two tiles in flight (costs a third of the LDS;
+// occupancy drops to 1 block/CU at NJ=4 — A/B via DTMX_NT_DEPTH).
+template <int NJ, int WM, int DEPTH, class PA, class PB, class EPI>
 __launch_bounds__(256, 2) __global__
 void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
                     uint32_t tiles_n, uint32_t kt_per_slice) {
@@ -875,7 +880,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   if (kt0 >= ktiles_total) return;
   // 16-B alignment: ds_read_b64_tr_b16 at a misaligned address silently
   // returns the 8-aligned address's data (G17)
-  __shared__ __attribute__((aligned(16))) elem_t smem[2][64 * (BM + BN)];
+  __shared__ __attribute__((aligned(16))) elem_t smem[DEPTH][64 * (BM + BN)];
   constexpr uint32_t B_OFF = 64 * BM;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
@@ -916,8 +921,15 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   f32x4 acc[4][NJ] = {};
 
   constexpr int G = (int)(ITS_A + ITS_B);
-  auto wait_tile = [&](bool one_in_flight) {
-    if (one_in_flight) {
+  auto wait_tile = [&](uint32_t flight) {
+    if (DEPTH == 3 && flight >= 2) {
+      if constexpr (G == 10)
+        asm volatile("s_waitcnt vmcnt(20)" ::: "memory");
+      else if constexpr (G == 8)
+        asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    } else if (flight >= 1) {
       if constexpr (G == 10)
         asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
       else if constexpr (G == 8)
@@ -929,11 +941,11 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
     }
   };
 
-  stage(0, kt0);
-  if (ktiles > 1) stage(1, kt0 + 1);
+  for (uint32_t d = 0; d < (uint32_t)DEPTH && d < ktiles; ++d)
+    stage(d, kt0 + d);
   for (uint32_t kt = 0; kt < ktiles; ++kt) {
-    const uint32_t cur = kt & 1;
-    wait_tile(kt + 1 < ktiles);
+    const uint32_t cur = kt % DEPTH;
+    wait_tile(min((uint32_t)(DEPTH - 1), ktiles - 1 - kt));
     __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -981,6 +993,8 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
             : "v"(a0), "v"(a1), "v"(a2), "v"(a3), "v"(a4), "v"(a5), "v"(a6),
               "v"(a7)
             : "memory");
+        __builtin_amdgcn_sched_barrier(0);  // rule 18: pin MFMA below the
+        // inline-asm lgkmcnt (hipcc may hoist register-only MFMA past it)
         Frag f;
         f.h[0] = t0; f.h[1] = t1; af[0] = f.f;
         f.h[0] = t2; f.h[1] = t3; af[1] = f.f;
@@ -1011,6 +1025,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
             : "=&v"(t0), "=&v"(t1), "=&v"(t2), "=&v"(t3)
             : "v"(b0), "v"(b1), "v"(b2), "v"(b3)
             : "memory");
+        __builtin_amdgcn_sched_barrier(0);
         Frag f;
         f.h[0] = t0; f.h[1] = t1; bfr[0] = f.f;
         f.h[0] = t2; f.h[1] = t3; bfr[1] = f.f;
@@ -1025,6 +1040,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
               : "=&v"(t4), "=&v"(t5), "=&v"(t6), "=&v"(t7)
               : "v"(b4), "v"(b5), "v"(b6), "v"(b7)
               : "memory");
+          __builtin_amdgcn_sched_barrier(0);
           f.h[0] = t4; f.h[1] = t5; bfr[2] = f.f;
           f.h[0] = t6; f.h[1] = t7; bfr[3] = f.f;
         }
@@ -1038,7 +1054,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
       __builtin_amdgcn_s_setprio(0);
     }
     __builtin_amdgcn_s_barrier();
-    if (kt + 2 < ktiles) stage(cur, kt0 + kt + 2);
+    if (kt + DEPTH < ktiles) stage(cur, kt0 + kt + DEPTH);
   }
   epi.template store<NJ>(acc, bm + wr, bn + wc, lane);
 }
@@ -1305,26 +1321,46 @@ static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
   uint32_t ktiles_total = ceil_div(K, 64);
   splitk = std::min(splitk, ktiles_total);
   uint32_t kt_per = ceil_div(ktiles_total, splitk);
+  // measured: depth 3 drops occupancy to 1 block/CU (96 KiB LDS at NJ=4)
+  // and LOSES 20-35% on every wgrad shape — the double buffer's 2-block
+  // overlap already covers the HBM latency better than a deeper ring at
+  // half the resident waves. Default 2; DTMX_NT_DEPTH=3 for A/B.
+  static const bool depth3 = [] {
+    const char* v = getenv("DTMX_NT_DEPTH");
+    return v && v[0] == '3';
+  }();
   if (M <= 64 && N > 64) {
     // flat 64x256 tile: ResNet's K_out=64 wgrads waste half the 128-row
     // tile on zero rows (see gemm_nt_kernel WM doc)
     uint32_t tiles_n = ceil_div(N, 256);
     dim3 grid(tiles_n, ceil_div(ktiles_total, kt_per));
-    gemm_nt_kernel<4, 1, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
-        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    if (depth3)
+      gemm_nt_kernel<4, 1, 3, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    else
+      gemm_nt_kernel<4, 1, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
     return;
   }
   uint32_t tiles_m = ceil_div(M, 128);
   if (N <= 64) {
     uint32_t tiles_n = ceil_div(N, 64);
     dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-    gemm_nt_kernel<2, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
-        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    if (depth3)
+      gemm_nt_kernel<2, 2, 3, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    else
+      gemm_nt_kernel<2, 2, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
   } else {
     uint32_t tiles_n = ceil_div(N, 128);
     dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-    gemm_nt_kernel<4, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
-        pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    if (depth3)
+      gemm_nt_kernel<4, 2, 3, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
+    else
+      gemm_nt_kernel<4, 2, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
   }
 }
 
