@@ -116,3 +116,59 @@ def test_nccl_collectives_and_reinit_inproc():
         assert obj[0]["gen"] == gen
         torch.cuda.synchronize()
         dist.destroy_process_group()
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(600)
+def test_elastic_join_leave_cuda_model_gloo(tmp_path):
+    """Full 2->3->2 join/leave with the MODEL ON THE GPU (three processes
+    sharing cuda:0; gloo carries the collectives since RCCL refuses
+    duplicate devices). Verifies elastic re-forms interleave correctly with
+    live HIP kernel work and the replicated-DP invariant holds on device."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from test_elastic import _wait_epoch
+
+    for attempt in range(2):
+        scratch = tmp_path / f"g{attempt}"
+        scratch.mkdir()
+        try:
+            _run_cuda_gloo(scratch)
+            return
+        except (AssertionError, TimeoutError):
+            if attempt == 1:
+                raise
+
+
+def _run_cuda_gloo(tmp_path):
+    from test_elastic import _wait_epoch
+
+    port = _free_port()
+    sched = Scheduler("127.0.0.1", port, ["127.0.0.1#0", "127.0.0.1#1"],
+                      hostfile=str(tmp_path / "hosts"))
+    outs = {i: str(tmp_path / f"out{i}.json") for i in range(3)}
+    env = {"DTMX_BACKEND": "gloo", "ELASTIC_TEST_DEVICE": "cuda"}
+    w0 = _spawn("127.0.0.1#0", port, outs[0], extra=env)
+    w1 = _spawn("127.0.0.1#1", port, outs[1], extra=env)
+    try:
+        _wait_epoch(sched, 2, timeout=180)
+        sched.publish(["127.0.0.1#0", "127.0.0.1#1", "127.0.0.1#2"])
+        eb = sched.store.get("cluster/env/EPOCH_BEGIN").decode()
+        w2 = _spawn("127.0.0.1#2", port, outs[2],
+                    extra={**env, "NEW_WORKER": "1", "EPOCH_BEGIN": eb})
+        _wait_epoch(sched, 5, timeout=180)
+        sched.publish(["127.0.0.1#0", "127.0.0.1#1"])
+        for name, p in (("w0", w0), ("w1", w1), ("w2", w2)):
+            rc = p.wait(timeout=180)
+            if rc != 0:
+                _, err = p.communicate()
+                raise AssertionError(f"{name} rc={rc}\n{err.decode()[-3000:]}")
+    finally:
+        for p in (w0, w1):
+            if p.poll() is None:
+                p.kill()
+    r0 = json.load(open(outs[0]))
+    r1 = json.load(open(outs[1]))
+    assert r0["device"] == "cuda" and r0["worker_counts"][:2] == [2, 3]
+    assert r0["final_workers"] == 2
+    assert r0["param_sum"] == pytest.approx(r1["param_sum"], rel=0, abs=0)
