@@ -329,12 +329,9 @@ __global__ void bn_apply_kernel(const elem_t* __restrict__ x, elem_t* __restrict
   using V8 = typename E8<elem_t>::v8;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t stride = gridDim.x * blockDim.x;
-  for (; i < total8; i += stride) {
-    V8 v = *(const V8*)(x + (size_t)i * 8);
-    V8 res;
-    if (residual) res = *(const V8*)(residual + (size_t)i * 8);
-    uint32_t q = dcv.div(i);
-    uint32_t c0 = dcv.mod(i, q) * 8;
+  auto body = [&](uint32_t ii, V8 v, V8 res) {
+    uint32_t q = dcv.div(ii);
+    uint32_t c0 = dcv.mod(ii, q) * 8;
     V8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
@@ -343,7 +340,23 @@ __global__ void bn_apply_kernel(const elem_t* __restrict__ x, elem_t* __restrict
       if (relu) r = fmaxf(r, 0.f);
       o[e] = (elem_t)r;
     }
-    *(V8*)(y + (size_t)i * 8) = o;
+    *(V8*)(y + (size_t)ii * 8) = o;
+  };
+  V8 zed = {};
+  // 2x unrolled so 2-4 16-B loads are in flight per wave (single-chunk loop
+  // compiles to load -> wait -> use: HBM-latency-bound)
+  for (; i + stride < total8; i += 2 * stride) {
+    V8 v0 = *(const V8*)(x + (size_t)i * 8);
+    V8 v1 = *(const V8*)(x + (size_t)(i + stride) * 8);
+    V8 r0 = residual ? *(const V8*)(residual + (size_t)i * 8) : zed;
+    V8 r1 = residual ? *(const V8*)(residual + (size_t)(i + stride) * 8) : zed;
+    body(i, v0, r0);
+    body(i + stride, v1, r1);
+  }
+  for (; i < total8; i += stride) {
+    V8 v = *(const V8*)(x + (size_t)i * 8);
+    V8 res = residual ? *(const V8*)(residual + (size_t)i * 8) : zed;
+    body(i, v, res);
   }
 }
 
@@ -441,14 +454,10 @@ __global__ void bn_bwd_dx_kernel(const elem_t* __restrict__ x,
   using V8 = typename E8<elem_t>::v8;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   const uint32_t stride = gridDim.x * blockDim.x;
-  for (; i < total8; i += stride) {
-    size_t off = (size_t)i * 8;
-    V8 xv = *(const V8*)(x + off);
-    V8 gv = *(const V8*)(dy + off);
-    V8 yv;
-    if (relu) yv = *(const V8*)(y + off);
-    uint32_t q = dcv.div(i);
-    uint32_t c0 = dcv.mod(i, q) * 8;
+  auto body = [&](uint32_t ii, V8 xv, V8 gv, V8 yv) {
+    uint32_t q = dcv.div(ii);
+    uint32_t c0 = dcv.mod(ii, q) * 8;
+    size_t off = (size_t)ii * 8;
     V8 o, om;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
@@ -463,6 +472,22 @@ __global__ void bn_bwd_dx_kernel(const elem_t* __restrict__ x,
     }
     *(V8*)(dx + off) = o;
     if (dres) *(V8*)(dres + off) = om;
+  };
+  V8 zed = {};
+  // 2x unrolled: 4-6 loads in flight per wave instead of load->wait->use
+  for (; i + stride < total8; i += 2 * stride) {
+    size_t o0 = (size_t)i * 8, o1 = (size_t)(i + stride) * 8;
+    V8 x0 = *(const V8*)(x + o0), x1 = *(const V8*)(x + o1);
+    V8 g0 = *(const V8*)(dy + o0), g1 = *(const V8*)(dy + o1);
+    V8 y0 = relu ? *(const V8*)(y + o0) : zed;
+    V8 y1 = relu ? *(const V8*)(y + o1) : zed;
+    body(i, x0, g0, y0);
+    body(i + stride, x1, g1, y1);
+  }
+  for (; i < total8; i += stride) {
+    size_t o0 = (size_t)i * 8;
+    body(i, *(const V8*)(x + o0), *(const V8*)(dy + o0),
+         relu ? *(const V8*)(y + o0) : zed);
   }
 }
 

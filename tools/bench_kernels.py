@@ -91,5 +91,41 @@ def main():
               f"wgrad {tf(flops,sw):6.1f} TF ({sw*1e3:.3f})")
 
 
+
+
+def bn_micro():
+    """Per-kernel GB/s of the BN elementwise/reduction passes (roofline is
+    ~6.3 TB/s; anything far below is a tuning target, not 'memory-bound')."""
+    torch.manual_seed(0)
+    N, C, H = 1024, 256, 56
+    x = torch.randn(N, C, H, H, dtype=torch.bfloat16, device=DEV).contiguous(
+        memory_format=torch.channels_last)
+    dy = torch.randn_like(x)
+    y = torch.relu(x)
+    gamma = torch.randn(C, dtype=torch.bfloat16, device=DEV)
+    beta = torch.randn(C, dtype=torch.bfloat16, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    nbytes = x.numel() * 2
+
+    s = timeit(lambda: ext.bn_fwd_train(x, gamma, beta, rm, rv, 0.9, 1e-5,
+                                        True, None, None, None), iters=10)
+    print(f"bn_fwd_train (stats+apply, ~3T): {3 * nbytes / s / 1e12:.2f} TB/s  {s*1e3:.3f} ms")
+    out = ext.bn_fwd_train(x, gamma, beta, rm, rv, 0.9, 1e-5, True, None,
+                           None, None)
+    yv, mean, invstd = out
+    s = timeit(lambda: ext.bn_bwd(x, dy, gamma, mean, invstd, True, yv, False),
+               iters=10)
+    print(f"bn_bwd (stats+dx, ~7T):          {7 * nbytes / s / 1e12:.2f} TB/s  {s*1e3:.3f} ms")
+    tdb = mean.clone(); tdg = invstd.clone()
+    s = timeit(lambda: ext.bn_bwd_dx_presummed(x, dy, yv, mean, invstd, gamma,
+                                               tdb, tdg, N * H * H, False,
+                                               False), iters=10)
+    print(f"bn_bwd_dx (2R+1W):               {3 * nbytes / s / 1e12:.2f} TB/s  {s*1e3:.3f} ms")
+    s = timeit(lambda: ext.bn_local_sums(x), iters=10)
+    print(f"bn_stats (1R):                   {1 * nbytes / s / 1e12:.2f} TB/s  {s*1e3:.3f} ms")
+
+
 if __name__ == "__main__":
     main()
+    bn_micro()
