@@ -69,6 +69,30 @@ class Trainer:
         if self._bucketer is not None:
             self._bucketer.finish()
 
+    def save_states(self, fname: str):
+        """Optimizer-state checkpoint (reference gluon trainer.save_states)."""
+        import pickle
+
+        with open(fname, "wb") as f:
+            pickle.dump({k: _cpu_state(v) for k, v in
+                         self._updater.get_states().items()}, f)
+
+    def load_states(self, fname: str):
+        import pickle
+
+        with open(fname, "rb") as f:
+            self._updater.set_states(pickle.load(f))
+
+
+def _cpu_state(s):
+    import torch as _t
+
+    if isinstance(s, _t.Tensor):
+        return s.detach().cpu()
+    if isinstance(s, tuple):
+        return tuple(_cpu_state(x) for x in s)
+    return s
+
 
 class L2Loss(nn.Module):
     def forward(self, pred, label):

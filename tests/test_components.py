@@ -345,3 +345,29 @@ def test_kill_dtmx_pidfiles(tmp_path):
     assert n == 1
     assert p.wait(timeout=15) != 0  # SIGTERM'd
     assert not list(rundir.glob("*.pid"))  # ledger cleaned
+
+
+def test_gluon_trainer_state_roundtrip(tmp_path):
+    import dtmx.gluon as gluon
+    from dtmx.models import get_symbol
+
+    net = get_symbol("mlp", num_classes=10, input_dim=16)
+    tr = gluon.Trainer(net.parameters(), "sgd",
+                       {"learning_rate": 0.1, "momentum": 0.9})
+    x = torch.randn(4, 16)
+    out = net(x)
+    out.sum().backward()
+    tr.step(batch_size=4)
+    f = str(tmp_path / "t.states")
+    tr.save_states(f)
+    tr2 = gluon.Trainer(net.parameters(), "sgd",
+                        {"learning_rate": 0.1, "momentum": 0.9})
+    tr2.load_states(f)
+    s1 = tr._updater.get_states()
+    s2 = tr2._updater.get_states()
+    assert set(s1) == set(s2)
+    for k in s1:
+        a = s1[k][0] if isinstance(s1[k], tuple) else s1[k]
+        b = s2[k][0] if isinstance(s2[k], tuple) else s2[k]
+        if a is not None:
+            torch.testing.assert_close(a.cpu(), b.cpu())
