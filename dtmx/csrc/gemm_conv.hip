@@ -36,6 +36,7 @@ namespace dtmx {
 template <typename elem_t>
 struct DenseP {
   using elem = elem_t;
+  static constexpr bool kDense = true;
   const elem_t* base;
   const elem_t* zero;
   uint32_t M, K;  // rows, k extent (elements, multiple of 8)
@@ -57,6 +58,7 @@ struct DenseP {
 template <typename elem_t>
 struct ConvFwdA {
   using elem = elem_t;
+  static constexpr bool kDense = false;
   const elem_t* x;
   const elem_t* zero;
   uint32_t M, Ktot;              // M = N*P*Q, Ktot = R*S*C
@@ -95,6 +97,7 @@ struct ConvFwdA {
 template <typename elem_t>
 struct ConvDgradA {
   using elem = elem_t;
+  static constexpr bool kDense = false;
   const elem_t* dy;
   const elem_t* zero;
   uint32_t M, Ktot;              // M = N*H*W, Ktot = R*S*Kout
@@ -591,10 +594,16 @@ void gemm256f_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles, uint32_t tiles_n) {
   // 0.2% vs 2.5% bank-conflict cycles on this layout)
   auto swz = [](uint32_t b) { return b ^ (((b >> 7) & 7) << 4); };
 
-  // branch-free staging: per-lane source pointers precomputed, stride 0 for
-  // OOB rows (they re-read the zero page); K%64==0 enforced by the caller
+  // staging sources. Dense sides get branch-free precomputed pointers with
+  // a per-tile stride (0 for OOB rows -> zero page; K%64==0 enforced by the
+  // caller). Gather sides (conv im2col/dgrad A) hoist the per-row decode
+  // into a Row context once (the same trick as the 128^2 kernel) and call
+  // addr(ctx, k8) per staged piece.
   const elem_t* src0[4][2];
   size_t sstep[4][2];
+  typename PA::Row actx[2][2];
+  typename PB::Row bctx[2][2];
+  uint32_t k8l[4][2];
 #pragma unroll
   for (uint32_t part = 0; part < 4; ++part)
 #pragma unroll
@@ -603,24 +612,44 @@ void gemm256f_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles, uint32_t tiles_n) {
       uint32_t lb = swz(off);
       uint32_t row = lb >> 7, kb = lb & 127;
       uint32_t k = (kb >> 4) * 8;
+      k8l[part][g] = kb >> 4;
       if (part < 2) {
         uint32_t m = bn + part * 128 + row;
-        bool oob = m >= pb.M;
-        src0[part][g] = oob ? pb.zero : pb.base + (size_t)m * pb.ld + k;
-        sstep[part][g] = oob ? 0 : 64;
+        if constexpr (PB::kDense) {
+          bool oob = m >= pb.M;
+          src0[part][g] = oob ? pb.zero : pb.base + (size_t)m * pb.ld + k;
+          sstep[part][g] = oob ? 0 : 64;
+        } else {
+          bctx[part][g] = pb.row(m);
+        }
       } else {
         uint32_t m = bm + (part - 2) * 128 + row;
-        bool oob = m >= pa.M;
-        src0[part][g] = oob ? pa.zero : pa.base + (size_t)m * pa.ld + k;
-        sstep[part][g] = oob ? 0 : 64;
+        if constexpr (PA::kDense) {
+          bool oob = m >= pa.M;
+          src0[part][g] = oob ? pa.zero : pa.base + (size_t)m * pa.ld + k;
+          sstep[part][g] = oob ? 0 : 64;
+        } else {
+          actx[part - 2][g] = pa.row(m);
+        }
       }
     }
   auto stage_part = [&](auto part_c, uint32_t buf, uint32_t kt) {
     constexpr uint32_t part = decltype(part_c)::value;
 #pragma unroll
-    for (uint32_t g = 0; g < 2; ++g)
-      glds16(src0[part][g] + (size_t)kt * sstep[part][g],
-             &smem[buf][part][(g * 8 + wave) * 512]);
+    for (uint32_t g = 0; g < 2; ++g) {
+      void* dst = &smem[buf][part][(g * 8 + wave) * 512];
+      if constexpr (part < 2) {
+        if constexpr (PB::kDense)
+          glds16(src0[part][g] + (size_t)kt * sstep[part][g], dst);
+        else
+          glds16(pb.addr(bctx[part][g], kt * 8 + k8l[part][g]), dst);
+      } else {
+        if constexpr (PA::kDense)
+          glds16(src0[part][g] + (size_t)kt * sstep[part][g], dst);
+        else
+          glds16(pa.addr(actx[part - 2][g], kt * 8 + k8l[part][g]), dst);
+      }
+    }
   };
   const uint32_t total_halves = ktiles * 4;
   auto stage_stream = [&](uint32_t h) {
@@ -1265,24 +1294,24 @@ static hipStream_t cur_stream() {
                                 (t).scalar_type() == at::kHalf),             \
               #t " must be a CUDA bf16/fp16 tensor")
 
-// route a dense-by-dense GEMM through the 256² kernel when the shape
-// qualifies (big M/N, pipelined K, 128-row slab granularity preserved)
-template <typename elem_t, class EPI>
-static bool try_gemm256(const DenseP<elem_t>& pa, const DenseP<elem_t>& pb,
-                        const EPI& epi, uint32_t M, uint32_t N, uint32_t K,
-                        uint32_t splitk) {
+// route a GEMM through the 256² kernel when the shape qualifies (big M/N,
+// pipelined K%64==0, 128-row slab granularity preserved). Dense AND gather
+// providers (conv fwd/dgrad A) are supported; OOB k pieces read the zero
+// page via the provider's own bounds checks on the gather side.
+template <class PA, class PB, class EPI>
+static bool try_gemm256(const PA& pa, const PB& pb, const EPI& epi,
+                        uint32_t M, uint32_t N, uint32_t K, uint32_t splitk) {
   if constexpr (!EPI::kLdsStage) {
     return false;
   } else {
     static const bool off = env_flag("DTMX_DISABLE_GEMM256");
     if (off || splitk > 1) return false;
     if (M < 512 || N < 192 || K < 128 || K % 64 != 0) return false;
-    if (pa.K != K || pb.K != K) return false;
     uint32_t tiles_m = ceil_div(M, 256), tiles_n = ceil_div(N, 256);
     if (tiles_m * tiles_n < 160) return false;  // underfilled grid
     dim3 grid(tiles_m * tiles_n);
-    gemm256f_kernel<DenseP<elem_t>, DenseP<elem_t>, EPI>
-        <<<grid, 512, 0, cur_stream()>>>(pa, pb, epi, K / 64, tiles_n);
+    gemm256f_kernel<PA, PB, EPI><<<grid, 512, 0, cur_stream()>>>(
+        pa, pb, epi, K / 64, tiles_n);
     return true;
   }
 }
@@ -1292,11 +1321,7 @@ static bool try_gemm256(const DenseP<elem_t>& pa, const DenseP<elem_t>& pb,
 template <class PA, class PB, class EPI>
 static void launch_gemm(const PA& pa, const PB& pb, const EPI& epi, uint32_t M,
                         uint32_t N, uint32_t K, uint32_t splitk = 1) {
-  using elem = typename PA::elem;
-  if constexpr (std::is_same_v<PA, DenseP<elem>> &&
-                std::is_same_v<PB, DenseP<elem>>) {
-    if (try_gemm256(pa, pb, epi, M, N, K, splitk)) return;
-  }
+  if (try_gemm256(pa, pb, epi, M, N, K, splitk)) return;
   uint32_t ktiles_total = ceil_div(K, 64);
   splitk = std::min(splitk, ktiles_total);
   uint32_t kt_per = ceil_div(ktiles_total, splitk);

@@ -749,3 +749,26 @@ def test_gemm256_routed_conv_and_bnfuse():
     assert (g.float() - gref).abs().max().item() / (gref.abs().max().item() + 1e-6) < 0.02
     torch.testing.assert_close(pdb.sum(0), gref.sum(dim=(0, 2, 3)), rtol=2e-3,
                                atol=3.0)
+
+
+@pytest.mark.gpu
+def test_gemm256_routed_gather_conv_matches_torch():
+    """3x3 conv fwd/dgrad on shapes big enough to route the GATHER providers
+    through the 256^2 kernel (batch 512 l3-shape) must match torch."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from dtmx.ops.hip import require_ext
+
+    ext = require_ext()
+    torch.manual_seed(3)
+    N, C, Ko, H = 512, 256, 256, 14
+    x = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16) * 0.5)
+    w = _cl(torch.randn(Ko, C, 3, 3, device="cuda").to(torch.bfloat16) * 0.05)
+    y = ext.conv_fwd(x, w, 1, 1)
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), padding=1)
+    assert (y.float() - ref).abs().max().item() / ref.abs().max().item() < 0.02
+    dy = _cl(torch.randn(N, Ko, H, H, device="cuda").to(torch.bfloat16))
+    dx = ext.conv_dgrad(dy, w, 1, 1, H, H, None)
+    dref = torch.nn.grad.conv2d_input((N, C, H, H), w.float(), dy.float(),
+                                      padding=1)
+    assert (dx.float() - dref).abs().max().item() / dref.abs().max().item() < 0.02
