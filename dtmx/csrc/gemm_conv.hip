@@ -1828,8 +1828,15 @@ at::Tensor conv_wgrad(const at::Tensor& x, const at::Tensor& dy, long R, long S,
               ? ceil_div(RSC, 256)  // flat 64x256 tile (launch_gemm_nt)
               : ceil_div(Ko, 128) * ceil_div(RSC, RSC <= 64 ? 64 : 128);
       uint32_t ktiles = ceil_div(M, 64);
+      static const uint32_t target_blocks = [] {
+        const char* v = getenv("DTMX_WGRAD_TARGET_BLOCKS");
+        // 512 = exactly 2 blocks/CU: the sweep 256..4096 measured 512 best
+        // on every wgrad shape (+25% over the old 1024 — no partial block
+        // wave, fewest split-K atomic passes that still fill the chip)
+        return v ? (uint32_t)atoi(v) : 512u;
+      }();
       uint32_t splitk = std::max<uint32_t>(
-          1, std::min<uint32_t>(ktiles, 1024 / std::max(1u, tiles_mn)));
+          1, std::min<uint32_t>(ktiles, target_blocks / std::max(1u, tiles_mn)));
       static const bool no_ws = env_flag("DTMX_DISABLE_WGRAD_WS");
       auto dw32 = no_ws ? at::zeros({(long)Ko, (long)RSC},
                                     x.options().dtype(at::kFloat))
