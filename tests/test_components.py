@@ -371,3 +371,33 @@ def test_gluon_trainer_state_roundtrip(tmp_path):
         b = s2[k][0] if isinstance(s2[k], tuple) else s2[k]
         if a is not None:
             torch.testing.assert_close(a.cpu(), b.cpu())
+
+
+def test_bench_json_contract(tmp_path):
+    """bench.py emits exactly one driver-consumable JSON line with the
+    contract fields (BASELINE.json metric/config names)."""
+    import json as _json
+    import subprocess
+    import sys as _sys
+
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [_sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "1",
+         "--warmup", "0", "--batch-size", "4"],
+        capture_output=True, timeout=600, env=env)
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    lines = [ln for ln in r.stdout.decode().splitlines() if ln.startswith("{")]
+    assert len(lines) == 1
+    d = _json.loads(lines[0])
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in d, k
+    assert d["metric"].startswith("images/sec ResNet-50")
+    assert d["data"] == "synthetic"
+    assert d["scaling"] == "weak"
+    assert d["config"]["model"] == "resnet-50"
+    assert d["config"]["global_batch"] == 4
+    assert d["config"]["parallelism"] == "dp1"
