@@ -401,3 +401,31 @@ def test_bench_json_contract(tmp_path):
     assert d["config"]["model"] == "resnet-50"
     assert d["config"]["global_batch"] == 4
     assert d["config"]["parallelism"] == "dp1"
+
+
+def test_random_api():
+    import dtmx.random as rnd
+
+    rnd.seed(42)
+    a = rnd.uniform(0, 1, (4, 4))
+    rnd.seed(42)
+    b = rnd.uniform(0, 1, (4, 4))
+    torch.testing.assert_close(a, b)
+    n = rnd.normal(0, 2, (1000,))
+    assert abs(n.std().item() - 2.0) < 0.3
+    r = rnd.randint(0, 10, (100,))
+    assert r.min() >= 0 and r.max() < 10
+
+
+def test_rnn_layer_shapes_and_training():
+    from dtmx.rnn import RNNLayer
+
+    for mode in ("lstm", "gru", "rnn_tanh"):
+        layer = RNNLayer(hidden_size=8, num_layers=2, mode=mode,
+                         input_size=6)
+        x = torch.randn(5, 3, 6)  # (T, N, C)
+        st = layer.begin_state(3)
+        out, st2 = layer(x, st if mode == "lstm" else st[0])
+        assert out.shape == (5, 3, 8)
+        out.sum().backward()
+        assert next(layer.parameters()).grad is not None
