@@ -429,3 +429,35 @@ def test_rnn_layer_shapes_and_training():
         assert out.shape == (5, 3, 8)
         out.sum().backward()
         assert next(layer.parameters()).grad is not None
+
+
+def test_image_api_roundtrip():
+    import numpy as np
+
+    import dtmx.image as img
+    from dtmx.ops.hip import require_ext
+
+    ext = require_ext()
+    ys, xs = np.mgrid[0:20, 0:24]
+    im = np.stack([ys * 5 % 256, xs * 7 % 256, (ys + xs) % 256],
+                  axis=-1).astype(np.uint8)
+    enc = ext.encode_jpeg(im.tobytes(), 20, 24, 3, 95)
+    dec = img.imdecode(bytes(enc))
+    assert tuple(dec.shape) == (20, 24, 3)
+    small = img.resize_short(dec, 10)
+    assert min(small.shape[0], small.shape[1]) == 10
+    crop, (x0, y0, tw, th) = img.center_crop(dec, (10, 12))
+    assert tuple(crop.shape) == (10, 12, 3) and (x0, y0) == (6, 5)
+    rc, _ = img.random_crop(dec, (8, 8))
+    assert tuple(rc.shape) == (8, 8, 3)
+    flipped = img.horizontal_flip(dec)
+    assert torch.equal(flipped[:, 0], dec[:, -1])
+
+
+def test_print_summary():
+    from dtmx.models import get_symbol
+    from dtmx.visualization import print_summary
+
+    net = get_symbol("mlp", num_classes=10, input_dim=16)
+    text = print_summary(net, shape=(2, 16))
+    assert "Total params" in text and "Linear" in text
