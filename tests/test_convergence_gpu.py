@@ -94,3 +94,22 @@ def test_resnet18_converges_bf16():
             optimizer_params=(("learning_rate", 0.05), ("momentum", 0.9)))
     acc = _train_acc(mod, it)
     assert acc >= 0.9, f"resnet-18 train accuracy {acc:.3f} < 0.9"
+
+
+def test_resnet50_converges_bf16():
+    """The FLAGSHIP model itself (full fused stack incl. gemm256 routing,
+    BN epilogue fusions, fused SGD) fits a learnable task — measured 1.0
+    accuracy at 320 steps; asserted at 160 for CI budget."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    X, Y = _pattern_images(n=2048, hw=32, classes=8, seed=9)
+    net = get_symbol("resnet", num_layers=50, num_classes=8,
+                     image_shape="3,32,32")
+    mod = dtmx.Module(net, context=dtmx.gpu(0))
+    mod.bind(data_shapes=[("data", (128, 3, 32, 32))],
+             label_shapes=[("softmax_label", (128,))], dtype=torch.bfloat16)
+    it = NDArrayIter({"data": X}, {"softmax_label": Y}, 128)
+    mod.fit(it, num_epoch=10, kvstore="local",
+            optimizer_params=(("learning_rate", 0.05), ("momentum", 0.9)))
+    acc = _train_acc(mod, it)
+    assert acc >= 0.95, f"resnet-50 train accuracy {acc:.3f} < 0.95"
