@@ -126,6 +126,31 @@ def bn_micro():
     print(f"bn_stats (1R):                   {1 * nbytes / s / 1e12:.2f} TB/s  {s*1e3:.3f} ms")
 
 
+
+
+def conv_routed():
+    """Big-batch conv shapes that route the GATHER providers through
+    gemm256f — per-shape A/B vs DTMX_DISABLE_GEMM256=1."""
+    import os
+    for (C, Ko, H, s, p, bs) in [(256, 256, 14, 1, 1, 512),
+                                 (512, 512, 7, 1, 1, 1024),
+                                 (128, 128, 28, 1, 1, 512)]:
+        x = nhwc(torch.randn(bs, C, H, H, dtype=torch.bfloat16))
+        w = nhwc(torch.randn(Ko, C, 3, 3, dtype=torch.bfloat16) * 0.05)
+        dy = nhwc(torch.randn(bs, Ko, H, H, dtype=torch.bfloat16))
+        P = H
+        fl = 2 * bs * P * P * Ko * C * 9
+        sf = timeit(lambda: ext.conv_fwd(x, w, s, p), iters=10)
+        sd = timeit(lambda: ext.conv_dgrad(dy, w, s, p, H, H, None), iters=10)
+        print(f"conv3x3 bs{bs} {C}->{Ko} {H}^2: fwd {tf(fl, sf):6.1f} TF  "
+              f"dgrad {tf(fl, sd):6.1f} TF")
+
+
+
+
 if __name__ == "__main__":
-    main()
-    bn_micro()
+    if os.environ.get("DTMX_BK_ONLY_CONV") == "1":
+        conv_routed()
+    else:
+        main()
+        bn_micro()
