@@ -78,3 +78,55 @@ def test_score_and_predict():
     assert out.shape == (64, 10)
     res = mod.score(it, "acc")
     assert 0.0 <= dict(res)["accuracy"] <= 1.0
+
+
+def test_checkpoint_resume_equivalence(tmp_path):
+    """Train 3 epochs -> checkpoint (+optimizer states) -> 2 more epochs
+    must equal fresh-process load(checkpoint) -> 2 epochs (reference
+    --load-epoch resume semantics, fit.py:74-84 / module.py:131-190)."""
+    import numpy as np
+
+    import dtmx
+    from dtmx.io import NDArrayIter
+    from dtmx.models import get_symbol
+
+    rng = np.random.RandomState(0)
+    X = rng.randn(64, 16).astype(np.float32)
+    Y = rng.randint(0, 10, 64).astype(np.float32)
+
+    def make():
+        torch.manual_seed(0)
+        net = get_symbol("mlp", num_classes=10, input_dim=16)
+        mod = dtmx.Module(net, context=dtmx.cpu())
+        mod.bind(data_shapes=[("data", (8, 16))],
+                 label_shapes=[("softmax_label", (8,))])
+        return mod
+
+    it = NDArrayIter({"data": X}, {"softmax_label": Y}, 8)
+    opt = (("learning_rate", 0.1), ("momentum", 0.9))
+
+    # continuous run: 3 + 2 epochs with a checkpoint at 3
+    mod = make()
+    mod.fit(it, num_epoch=3, kvstore="local", optimizer_params=opt)
+    prefix = str(tmp_path / "ck")
+    mod.save_checkpoint(prefix, 3, save_optimizer_states=True)
+    mod.fit(it, num_epoch=5, begin_epoch=3, kvstore="local",
+            optimizer_params=opt, force_init=False)
+    arg_cont, _ = mod.get_params()
+
+    # resumed run: load the checkpoint into a fresh module
+    mod2 = dtmx.Module.load(prefix, 3, get_symbol("mlp", num_classes=10,
+                                                  input_dim=16))
+    mod2.bind(data_shapes=[("data", (8, 16))],
+              label_shapes=[("softmax_label", (8,))])
+    mod2.init_params()
+    mod2.init_optimizer(kvstore="local", optimizer_params=opt)
+    mod2.load_optimizer_states(prefix + "-0003.states")
+    it2 = NDArrayIter({"data": X}, {"softmax_label": Y}, 8)
+    mod2.fit(it2, num_epoch=5, begin_epoch=3, kvstore="local",
+             optimizer_params=opt, force_init=False)
+    arg_res, _ = mod2.get_params()
+
+    for k in arg_cont:
+        torch.testing.assert_close(arg_res[k], arg_cont[k], rtol=1e-5,
+                                    atol=1e-6)
