@@ -170,3 +170,51 @@ def test_ssh_launch_path(tmp_path, monkeypatch):
     wid, cwd = marker.read_text().split("\n")
     assert wid == "fakehost#0"
     assert cwd == os.getcwd()  # launcher cd's to the working dir
+
+
+def test_prepare_data_admission_gate(tmp_path):
+    """A joiner whose prepare-data hook fails must NOT enter the roster
+    (reference prepare-data success-marker gate, README.md:100-110)."""
+    import importlib
+    import threading
+
+    sys.path.insert(0, os.path.join(ROOT, "tools"))
+    import launch as launch_mod
+    importlib.reload(launch_mod)
+
+    script = tmp_path / "prep.sh"
+    script.write_text("#!/bin/bash\n"
+                      '[ "$1" = "127.0.0.1#2" ] && exit 1\n'  # reject #2
+                      'exit 0\n')
+    script.chmod(0o755)
+    hostfile = tmp_path / "hosts"
+    hostfile.write_text("127.0.0.1\n127.0.0.1\n")
+
+    class A:
+        hostfile = str(tmp_path / "hosts")
+        num_workers = 2
+        scheduler_host = "127.0.0.1"
+        scheduler_port = _free_port()
+        elastic_training_enabled = True
+        poll_seconds = 0.2
+        sync_dst_dir = None
+        prepare_data_script = str(script)
+
+    lau = launch_mod.Launcher(A(), [sys.executable, "-c", "pass"])
+    stop = threading.Event()
+    t = threading.Thread(target=lau.watch, args=(stop,), daemon=True)
+    t.start()
+    try:
+        # add two joiners; #2 is rejected by the hook, #3 admitted
+        hostfile.write_text("127.0.0.1\n" * 4)
+        deadline = time.time() + 30
+        while time.time() < deadline and "127.0.0.1#3" not in lau.procs:
+            time.sleep(0.1)  # (the EPOCH_BEGIN hint fetch adds ~2s pre-launch)
+        assert "127.0.0.1#3" in lau.scheduler.members
+        assert "127.0.0.1#2" not in lau.scheduler.members  # gated out
+        assert "127.0.0.1#3" in lau.procs and "127.0.0.1#2" not in lau.procs
+    finally:
+        stop.set()
+        for p in lau.procs.values():
+            if p.poll() is None:
+                p.kill()

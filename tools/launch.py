@@ -95,6 +95,23 @@ class Launcher:
                 self.scheduler.store.set_timeout(timedelta(seconds=300))
         return env
 
+    def prepare_data(self, wid: str) -> bool:
+        """Admission gate for joiners (reference prepare-data.py workflow:
+        the EC2 manager admits a host only after its data-prep success
+        marker appears, README.md:100-110). Runs the configured script on
+        the joiner's host; True = admit."""
+        script = getattr(self.args, "prepare_data_script", None)
+        if not script:
+            return True
+        host = wid.split("#")[0]
+        if is_local(host):
+            cmd = [script, wid]
+        else:
+            cmd = ["ssh", "-o", "StrictHostKeyChecking=no", host,
+                   f"{shlex.quote(script)} {shlex.quote(wid)}"]
+        r = subprocess.run(cmd)
+        return r.returncode == 0
+
     def launch_worker(self, wid: str, new_worker: bool = False):
         host = wid.split("#")[0]
         env = self.worker_env(wid, new_worker)
@@ -143,7 +160,15 @@ class Launcher:
             if members == last:
                 continue
             before = set(self.scheduler.members)
-            self.scheduler.publish(members)
+            # admission gate BEFORE publish: a joiner that enters the roster
+            # but never starts would strand the survivors' re-form (the
+            # reference's marker gates hostfile entry the same way)
+            admitted = [m for m in members
+                        if m in before or self.prepare_data(m)]
+            for m in set(members) - set(admitted):
+                logging.error("prepare-data failed for %s; joiner NOT "
+                              "admitted to the roster", m)
+            self.scheduler.publish(admitted)
             added = set(self.scheduler.members) - before
             for wid in sorted(added):
                 self.launch_worker(wid, new_worker=True)
@@ -192,6 +217,11 @@ def main():
     ap.add_argument("--poll-seconds", type=float, default=1.0)
     ap.add_argument("--sync-dst-dir", type=str, default=None,
                     help="rsync working dir to remote hosts before launch")
+    ap.add_argument("--prepare-data-script", type=str, default=None,
+                    help="per-joiner data-preparation hook: run (locally or "
+                         "over ssh) before a NEW worker launches; a nonzero "
+                         "exit blocks admission (reference prepare-data.py "
+                         "+ the prepare_data_success_<ip> EFS marker gate)")
     ap.add_argument("command", nargs=argparse.REMAINDER)
     args = ap.parse_args()
     command = [c for c in args.command if c != "--"]
