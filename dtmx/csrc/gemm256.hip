@@ -202,6 +202,158 @@ void gemm256_kernel(PA pa, PB pb, elem_t* __restrict__ c, uint32_t M,
     }
 }
 
+// ---- 32x32x16 MFMA variant (probe): same schedule, bigger matrix op
+// (2382 vs 2075 TF ubench ceiling). Per-wave 128x64 = 4 M-tiles x 2
+// N-tiles of 32^2; phase q computes M-tile q (8 MFMA/phase).
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+template <typename elem_t>
+__device__ __forceinline__ f32x16 mfma32(typename E8<elem_t>::v8 a,
+                                         typename E8<elem_t>::v8 b, f32x16 c) {
+  if constexpr (std::is_same_v<elem_t, __bf16>)
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  else
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, c, 0, 0, 0);
+}
+
+template <class PA, class PB, typename elem_t>
+__launch_bounds__(512, 2) __global__
+void gemm256_m32_kernel(PA pa, PB pb, elem_t* __restrict__ c, uint32_t M,
+                        uint32_t N, uint32_t ktiles, uint32_t tiles_n) {
+  using V8 = typename E8<elem_t>::v8;
+  __shared__ __attribute__((aligned(16))) elem_t smem[2][4][8192];
+  const uint32_t t = threadIdx.x;
+  const uint32_t wave = t >> 6, lane = t & 63;
+  const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const uint32_t bm = (bid / tiles_n) * 256, bn = (bid % tiles_n) * 256;
+  const uint32_t am_half = wave >> 2;
+  const uint32_t wc = (wave & 3) * 64;
+
+  const elem_t* src0[4][2];
+  size_t sstep[4][2];
+#pragma unroll
+  for (uint32_t part = 0; part < 4; ++part)
+#pragma unroll
+    for (uint32_t g = 0; g < 2; ++g) {
+      uint32_t off = (g * 8 + wave) * 1024 + lane * 16;
+      uint32_t lb = swz256<0>(off);
+      uint32_t row = lb >> 7, kb = lb & 127;
+      uint32_t k = (kb >> 4) * 8;
+      if (part < 2) {
+        uint32_t m = bn + part * 128 + row;
+        bool oob = m >= pb.M;
+        src0[part][g] = oob ? pb.zero : pb.base + (size_t)m * pb.ld + k;
+        sstep[part][g] = oob ? 0 : 64;
+      } else {
+        uint32_t m = bm + (part - 2) * 128 + row;
+        bool oob = m >= pa.M;
+        src0[part][g] = oob ? pa.zero : pa.base + (size_t)m * pa.ld + k;
+        sstep[part][g] = oob ? 0 : 64;
+      }
+    }
+  auto stage_part = [&](auto part_c, uint32_t buf, uint32_t kt) {
+    constexpr uint32_t part = decltype(part_c)::value;
+#pragma unroll
+    for (uint32_t g = 0; g < 2; ++g)
+      glds16(src0[part][g] + (size_t)kt * sstep[part][g],
+             &smem[buf][part][(g * 8 + wave) * 512]);
+  };
+  const uint32_t total_halves = ktiles * 4;
+  auto stage_stream = [&](uint32_t h) {
+    if (h >= total_halves) return;
+    uint32_t kt = h >> 2, buf = kt & 1;
+    switch (h & 3) {
+      case 0: stage_part(std::integral_constant<uint32_t, 0>{}, buf, kt); break;
+      case 1: stage_part(std::integral_constant<uint32_t, 1>{}, buf, kt); break;
+      case 2: stage_part(std::integral_constant<uint32_t, 2>{}, buf, kt); break;
+      default: stage_part(std::integral_constant<uint32_t, 3>{}, buf, kt);
+    }
+  };
+  auto wait_vm = [&](uint32_t n) {
+    if (n >= 6)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if (n == 4)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (n == 2)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  };
+
+  uint32_t issued = 0;
+  for (; issued < min(4u, total_halves); ++issued) stage_stream(issued);
+  wait_vm(total_halves > 4 ? 4 : 0);
+  for (; issued < min(7u, total_halves); ++issued) stage_stream(issued);
+  wait_vm(2 * (issued - min(4u, total_halves)));
+  __builtin_amdgcn_s_barrier();
+
+  f32x16 acc4[4][2] = {};
+  const uint32_t a_slot = 2 + am_half;
+  const uint32_t b_slot = wc >> 7;
+  const uint32_t wc_local = wc & 127;
+
+  V8 bf[8];  // 2 j x 4 kk
+  uint32_t P = 0;
+  for (uint32_t kt = 0; kt < ktiles; ++kt) {
+    const uint32_t cur = kt & 1;
+#pragma unroll
+    for (uint32_t q = 0; q < 4; ++q, ++P) {
+      if (q == 0) {
+#pragma unroll
+        for (uint32_t j = 0; j < 2; ++j)
+#pragma unroll
+          for (uint32_t kk = 0; kk < 4; ++kk) {
+            uint32_t row = wc_local + j * 32 + (lane & 31);
+            uint32_t kbyte = kk * 32 + ((lane >> 5) << 4);
+            bf[j * 4 + kk] = *(const V8*)((const char*)&smem[cur][b_slot][0] +
+                                          swz256<0>(row * 128 + kbyte));
+          }
+      }
+      V8 af[4];
+#pragma unroll
+      for (uint32_t kk = 0; kk < 4; ++kk) {
+        uint32_t row = q * 32 + (lane & 31);
+        uint32_t kbyte = kk * 32 + ((lane >> 5) << 4);
+        af[kk] = *(const V8*)((const char*)&smem[cur][a_slot][0] +
+                              swz256<0>(row * 128 + kbyte));
+      }
+      stage_stream(7 + P);
+      if (q == 0)
+        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (uint32_t kk = 0; kk < 4; ++kk)
+#pragma unroll
+        for (uint32_t j = 0; j < 2; ++j)
+          acc4[q][j] = mfma32<elem_t>(af[kk], bf[j * 4 + kk], acc4[q][j]);
+      __builtin_amdgcn_s_setprio(0);
+      if (q == 3) {
+        uint32_t issued_now = min(total_halves, 7 + P + 1);
+        uint32_t needed = min(total_halves, 4 * (kt + 2));
+        wait_vm(2 * (issued_now - needed));
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue: 32x32 C map col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
+  const uint32_t m0 = bm + am_half * 128, n0 = bn + wc;
+#pragma unroll
+  for (uint32_t i = 0; i < 4; ++i)
+#pragma unroll
+    for (uint32_t j = 0; j < 2; ++j) {
+      uint32_t n = n0 + j * 32 + (lane & 31);
+      if (n >= N) continue;
+#pragma unroll
+      for (uint32_t r = 0; r < 16; ++r) {
+        uint32_t m = m0 + i * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        if (m >= M) continue;
+        c[(size_t)m * N + n] = (elem_t)acc4[i][j][r];
+      }
+    }
+}
+
 static const __bf16* zero_page256(const at::Tensor& like) {
   static at::Tensor z;
   if (!z.defined() || z.device() != like.device())
@@ -236,9 +388,15 @@ at::Tensor gemm256_nt(const at::Tensor& a, const at::Tensor& b) {
     static const int variant = [] {
       const char* b = getenv("DTMX_G256_BAR1");
       const char* w = getenv("DTMX_G256_ST16");
+      const char* m = getenv("DTMX_G256_M32");
+      if (m && m[0] == '1') return 4;
       return (b && b[0] == '1' ? 0 : 1) | (w && w[0] == '1' ? 2 : 0);
     }();
     switch (variant) {
+      case 4:  // 32x32x16 MFMA probe (same stream schedule, f32x16 acc)
+        gemm256_m32_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t>
+            <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles, tiles_n);
+        break;
       case 0:
         gemm256_kernel<Dense256P<elem_t>, Dense256P<elem_t>, elem_t, 0, 0>
             <<<grid, 512, 0, s>>>(pa, pb, (elem_t*)y.data_ptr(), M, N, ktiles, tiles_n);
