@@ -895,16 +895,34 @@ struct WgradXcolB {
 // tile does not cover the ~900-cycle HBM round trip, measured 35% parked
 // cycles on wgrad). 3 = two tiles in flight (costs a third of the LDS;
 // occupancy drops to 1 block/CU at NJ=4 — A/B via DTMX_NT_DEPTH).
-template <int NJ, int WM, int DEPTH, class PA, class PB, class EPI>
+// GROUP: XCD-grouped 1-D launch (DTMX_NT_GROUP): all output tiles of one
+// split-K slice are mapped onto ONE XCD (hardware round-robins 1-D block
+// slots across the 8 XCDs, so slot%8 selects the die). The tiles of a
+// slice stream the SAME 16-32 KiB dy K-tile per step — co-locating them
+// turns the tiles_n-fold dy re-read (18x on a 256-channel 3x3 wgrad, the
+// dominant HBM traffic of the whole wgrad family) into XCD-L2 hits.
+template <int NJ, int WM, int DEPTH, class PA, class PB, class EPI,
+          int GROUP = 0>
 __launch_bounds__(256, 2) __global__
 void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
-                    uint32_t tiles_n, uint32_t kt_per_slice) {
+                    uint32_t tiles_n, uint32_t kt_per_slice,
+                    uint32_t tiles_total = 0) {
   using elem_t = typename PA::elem;
   using V8 = typename E8<elem_t>::v8;
   constexpr uint32_t WN = 4 / WM;
   constexpr uint32_t BM = WM * 64;
   constexpr uint32_t BN = WN * NJ * 16;
-  const uint32_t kt0 = blockIdx.y * kt_per_slice;
+  uint32_t slice;
+  uint32_t bid_raw;
+  if constexpr (GROUP) {
+    const uint32_t xcd = blockIdx.x & 7, j = blockIdx.x >> 3;
+    bid_raw = j % tiles_total;
+    slice = xcd + 8 * (j / tiles_total);
+  } else {
+    bid_raw = blockIdx.x;
+    slice = blockIdx.y;
+  }
+  const uint32_t kt0 = slice * kt_per_slice;
   const uint32_t ktiles = min(kt_per_slice, ktiles_total - kt0);
   if (kt0 >= ktiles_total) return;
   // 16-B alignment: ds_read_b64_tr_b16 at a misaligned address silently
@@ -913,7 +931,7 @@ void gemm_nt_kernel(PA pa, PB pb, EPI epi, uint32_t ktiles_total,
   constexpr uint32_t B_OFF = 64 * BM;
   const uint32_t t = threadIdx.x;
   const uint32_t wave = t >> 6, lane = t & 63;
-  const uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const uint32_t bid = GROUP ? bid_raw : xcd_swizzle(bid_raw, gridDim.x);
   const uint32_t bm = (bid / tiles_n) * BM, bn = (bid % tiles_n) * BN;
 
   // staging: A rows are 256 B (128 m), chunk = 16 kd-rows per 4 KiB;
@@ -1354,12 +1372,42 @@ static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
     const char* v = getenv("DTMX_NT_DEPTH");
     return v && v[0] == '3';
   }();
+  // XCD-grouped slice scheduling (see gemm_nt_kernel GROUP doc): 1-D grid,
+  // all tiles of a split-K slice on one die so shared dy K-tiles hit L2.
+  // Measured per-shape (tools/bench_wgrad.py, bs1024): +11-12% at T=9 tiles
+  // (128-channel 3x3 wgrad class, slices/XCD fill 63 of 64 slots), -3% at
+  // T=144, -33%/-43% at T=36/T=3 (slot spill -> straggler slices, and an
+  // unexplained loss on the flat WM1 tile) — so auto mode gates it to the
+  // tile-count class that wins. DTMX_NT_GROUP=0/1 forces off/on everywhere.
+  static const int group_env = [] {
+    const char* v = getenv("DTMX_NT_GROUP");
+    return v ? (v[0] == '1' ? 1 : 0) : -1;
+  }();
+  uint32_t tiles_mn_probe =
+      (M <= 64 && N > 64) ? ceil_div(N, 256)
+                          : ceil_div(M, 128) * ceil_div(N, N <= 64 ? 64u : 128u);
+  const bool group = group_env == 1 ||
+                     (group_env == -1 && M > 64 && tiles_mn_probe > 4 &&
+                      tiles_mn_probe <= 16);
+  uint32_t nslices = ceil_div(ktiles_total, kt_per);
+  if (group) {
+    // slice->XCD is slice%8: a slice count that is not a multiple of 8
+    // leaves some dies with one more slice than others (measured -40% on
+    // the 3-slice 512-channel wgrad) — round UP to 8k slices instead
+    uint32_t ns8 = std::min(ktiles_total, ceil_div(nslices, 8u) * 8u);
+    kt_per = ceil_div(ktiles_total, ns8);
+    nslices = ceil_div(ktiles_total, kt_per);
+  }
   if (M <= 64 && N > 64) {
     // flat 64x256 tile: ResNet's K_out=64 wgrads waste half the 128-row
     // tile on zero rows (see gemm_nt_kernel WM doc)
     uint32_t tiles_n = ceil_div(N, 256);
-    dim3 grid(tiles_n, ceil_div(ktiles_total, kt_per));
-    if (depth3)
+    dim3 grid(tiles_n, nslices);
+    if (group && !depth3)
+      gemm_nt_kernel<4, 1, 2, PA, PB, EPI, 1>
+          <<<8 * tiles_n * ceil_div(nslices, 8u), 256, 0, cur_stream()>>>(
+              pa, pb, epi, ktiles_total, tiles_n, kt_per, tiles_n);
+    else if (depth3)
       gemm_nt_kernel<4, 1, 3, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
           pa, pb, epi, ktiles_total, tiles_n, kt_per);
     else
@@ -1370,8 +1418,13 @@ static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
   uint32_t tiles_m = ceil_div(M, 128);
   if (N <= 64) {
     uint32_t tiles_n = ceil_div(N, 64);
-    dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-    if (depth3)
+    dim3 grid(tiles_m * tiles_n, nslices);
+    if (group && !depth3)
+      gemm_nt_kernel<2, 2, 2, PA, PB, EPI, 1>
+          <<<8 * tiles_m * tiles_n * ceil_div(nslices, 8u), 256, 0,
+             cur_stream()>>>(pa, pb, epi, ktiles_total, tiles_n, kt_per,
+                             tiles_m * tiles_n);
+    else if (depth3)
       gemm_nt_kernel<2, 2, 3, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
           pa, pb, epi, ktiles_total, tiles_n, kt_per);
     else
@@ -1379,8 +1432,13 @@ static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
           pa, pb, epi, ktiles_total, tiles_n, kt_per);
   } else {
     uint32_t tiles_n = ceil_div(N, 128);
-    dim3 grid(tiles_m * tiles_n, ceil_div(ktiles_total, kt_per));
-    if (depth3)
+    dim3 grid(tiles_m * tiles_n, nslices);
+    if (group && !depth3)
+      gemm_nt_kernel<4, 2, 2, PA, PB, EPI, 1>
+          <<<8 * tiles_m * tiles_n * ceil_div(nslices, 8u), 256, 0,
+             cur_stream()>>>(pa, pb, epi, ktiles_total, tiles_n, kt_per,
+                             tiles_m * tiles_n);
+    else if (depth3)
       gemm_nt_kernel<4, 2, 3, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
           pa, pb, epi, ktiles_total, tiles_n, kt_per);
     else
