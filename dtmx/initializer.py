@@ -101,6 +101,89 @@ class MSRAPrelu(Xavier):
         super().__init__("gaussian", factor_type, 2.0 / (1 + slope ** 2))
 
 
+class Constant(Initializer):
+    def __init__(self, value: float = 0.0):
+        self.value = float(value)
+
+    def _init_weight(self, name, arr):
+        arr.fill_(self.value)
+
+
+class Orthogonal(Initializer):
+    """Orthogonal init via QR of a gaussian (reference initializer.py
+    Orthogonal; rand_type='uniform' uses a uniform base instead)."""
+
+    def __init__(self, scale: float = 1.414, rand_type: str = "uniform"):
+        self.scale = scale
+        self.rand_type = rand_type
+
+    def _init_weight(self, name, arr):
+        rows = arr.shape[0]
+        cols = arr.numel() // rows
+        hi, lo = max(rows, cols), min(rows, cols)
+        if self.rand_type == "uniform":
+            base = torch.rand(hi, lo, dtype=torch.float32) * 2 - 1
+        else:
+            base = torch.randn(hi, lo, dtype=torch.float32)
+        q, r = torch.linalg.qr(base)  # q: hi x lo, orthonormal columns
+        q = q * torch.sign(torch.diagonal(r, 0)).reshape(1, lo)
+        mat = q if rows >= cols else q.T
+        arr.copy_((self.scale * mat).reshape(arr.shape).to(arr.dtype))
+
+
+class Bilinear(Initializer):
+    """Bilinear upsampling kernel for deconvolution weights (reference
+    initializer.py Bilinear)."""
+
+    def _init_weight(self, name, arr):
+        w = torch.zeros(arr.numel(), dtype=torch.float32)
+        shape = arr.shape
+        f = math.ceil(shape[3] / 2.0)
+        c = (2 * f - 1 - f % 2) / (2.0 * f)
+        for i in range(arr.numel()):
+            x = i % shape[3]
+            y = (i // shape[3]) % shape[2]
+            w[i] = (1 - abs(x / f - c)) * (1 - abs(y / f - c))
+        arr.copy_(w.reshape(shape).to(arr.dtype))
+
+
+class LSTMBias(Initializer):
+    """Zero bias with the forget gate set to 1 (reference initializer.py
+    LSTMBias _init_bias; gate order i,f,g,o — forget = second quarter).
+    Overrides __call__: the base class would zero *bias-named* params before
+    this initializer ever saw them."""
+
+    def __init__(self, forget_bias: float = 1.0):
+        self.forget_bias = forget_bias
+
+    def __call__(self, name, arr):
+        with torch.no_grad():
+            self._init_weight(name, arr)
+
+    def _init_weight(self, name, arr):
+        arr.zero_()
+        n = arr.shape[0] // 4
+        arr[n:2 * n] = self.forget_bias
+
+
+class Mixed:
+    """Dispatch by name-pattern to member initializers (reference
+    initializer.py Mixed): patterns are regexes tried in order."""
+
+    def __init__(self, patterns, initializers):
+        import re
+        assert len(patterns) == len(initializers)
+        self.map = [(re.compile(p), i) for p, i in zip(patterns, initializers)]
+
+    def __call__(self, name: str, arr: torch.Tensor) -> None:
+        for pat, init in self.map:
+            if pat.match(name):
+                init(name, arr)
+                return
+        raise ValueError(f"Mixed: no pattern matches parameter {name} — "
+                         "add a '.*' catch-all")
+
+
 _REGISTRY = {
     "default": lambda: Xavier(rnd_type="gaussian", factor_type="in", magnitude=2),
     "xavier": Xavier,
@@ -109,6 +192,10 @@ _REGISTRY = {
     "normal": Normal,
     "zero": Zero,
     "one": One,
+    "constant": Constant,
+    "orthogonal": Orthogonal,
+    "bilinear": Bilinear,
+    "lstmbias": LSTMBias,
 }
 
 

@@ -70,6 +70,26 @@ class PolyScheduler(LRScheduler):
         return self.base_lr
 
 
+class CosineScheduler(LRScheduler):
+    """Cosine decay from base_lr to final_lr over max_update (reference
+    lr_scheduler.py CosineScheduler)."""
+
+    def __init__(self, max_update: int, base_lr: float = 0.01,
+                 final_lr: float = 0.0):
+        super().__init__(base_lr)
+        if max_update < 1:
+            raise ValueError("max_update must be >= 1")
+        self.max_update = max_update
+        self.final_lr = final_lr
+        self.base_lr_orig = base_lr
+
+    def __call__(self, num_update: int) -> float:
+        if num_update <= self.max_update:
+            self.base_lr = self.final_lr + (self.base_lr_orig - self.final_lr) * (
+                1 + math.cos(math.pi * num_update / self.max_update)) / 2
+        return self.base_lr
+
+
 class WarmupScheduler(LRScheduler):
     """Linear warmup to base_lr over warmup_steps, then delegate.
 

@@ -132,6 +132,63 @@ def test_composite_metric():
     assert "accuracy" in names and "cross-entropy" in names
 
 
+def test_metric_tail_vs_closed_forms():
+    """F1/MCC/MAE/MSE/RMSE/Perplexity/Pearson vs hand-computed values
+    (reference metric.py:605-1263 classes)."""
+    import math
+
+    labels = torch.tensor([1.0, 0.0, 1.0, 0.0])
+    preds = torch.tensor([[0.2, 0.8], [0.3, 0.7], [0.9, 0.1], [0.6, 0.4]])
+    # pred labels: 1, 1, 0, 0 -> tp=1 fp=1 tn=1 fn=1
+    f1 = metric.create("f1")
+    f1.update([labels], [preds])
+    assert abs(dict(f1.get_name_value())["f1"] - 0.5) < 1e-9
+    mcc = metric.create("mcc")
+    mcc.update([labels], [preds])
+    assert abs(dict(mcc.get_name_value())["mcc"] - 0.0) < 1e-9
+
+    y = torch.tensor([[1.0, 2.0, 3.0]])
+    p = torch.tensor([[2.0, 2.0, 5.0]])
+    mae = metric.create("mae")
+    mae.update([y], [p])
+    assert abs(mae.get()[1] - 1.0) < 1e-6
+    mse = metric.create("mse")
+    mse.update([y], [p])
+    assert abs(mse.get()[1] - 5.0 / 3) < 1e-6
+    rmse = metric.create("rmse")
+    rmse.update([y], [p])
+    assert abs(rmse.get()[1] - math.sqrt(5.0 / 3)) < 1e-6
+
+    probs = torch.tensor([[0.5, 0.5], [0.25, 0.75]])
+    lab = torch.tensor([0.0, 1.0])
+    ppl = metric.create("perplexity")
+    ppl.update([lab], [probs])
+    expect = math.exp(-(math.log(0.5) + math.log(0.75)) / 2)
+    assert abs(ppl.get()[1] - expect) < 1e-6
+    # ignore_label drops the masked row
+    ppl2 = metric.Perplexity(ignore_label=0)
+    ppl2.update([lab], [probs])
+    assert abs(ppl2.get()[1] - math.exp(-math.log(0.75))) < 1e-6
+
+    pr = metric.create("pearsonr")
+    pr.update([torch.tensor([1.0, 2.0, 3.0])], [torch.tensor([2.0, 4.0, 6.0])])
+    assert abs(pr.get()[1] - 1.0) < 1e-6
+
+    nll = metric.create("nll_loss")
+    nll.update([lab], [probs])
+    assert abs(nll.get()[1] + (math.log(0.5) + math.log(0.75)) / 2) < 1e-6
+
+
+def test_custom_metric_and_np():
+    def feval(label, pred):
+        return float((label == pred.argmax(-1)).mean())
+
+    m = metric.np(feval, name="myacc")
+    m.update([torch.tensor([1, 0])], [torch.tensor([[0.1, 0.9], [0.2, 0.8]])])
+    assert abs(m.get()[1] - 0.5) < 1e-9
+    assert "myacc" in m.name
+
+
 # ---------------------------------------------------------- lr scheduler
 
 def test_multifactor():
@@ -151,6 +208,14 @@ def test_warmup_rearm():
     assert s(105) == pytest.approx(0.5)
 
 
+def test_cosine_scheduler():
+    s = lr_scheduler.CosineScheduler(max_update=100, base_lr=1.0, final_lr=0.1)
+    assert s(0) == pytest.approx(1.0)
+    assert s(50) == pytest.approx(0.55)   # midpoint = (1.0+0.1)/2
+    assert s(100) == pytest.approx(0.1)
+    assert s(150) == pytest.approx(0.1)   # clamped after max_update
+
+
 # ----------------------------------------------------------- initializer
 
 def test_xavier_scale():
@@ -168,6 +233,36 @@ def test_name_based_dispatch():
     g = torch.randn(4)
     ini("bn0_gamma", g)
     assert torch.equal(g, torch.ones(4))
+
+
+def test_initializer_tail():
+    """Constant/Orthogonal/Bilinear/LSTMBias/Mixed (reference
+    initializer.py:287-702)."""
+    w = torch.empty(3, 5)
+    initializer.Constant(0.5)("w_weight", w)
+    assert torch.equal(w, torch.full((3, 5), 0.5))
+
+    q = torch.empty(16, 32)
+    initializer.Orthogonal(scale=1.0, rand_type="gaussian")("w_weight", q)
+    eye = q.float() @ q.float().T
+    assert torch.allclose(eye, torch.eye(16), atol=1e-4)
+
+    d = torch.empty(2, 1, 4, 4)
+    initializer.Bilinear()("deconv_weight", d)
+    # bilinear kernel: symmetric, peak at the center block
+    assert d[0, 0, 1, 1] == d[0, 0, 2, 2] == d.max()
+    assert torch.allclose(d[0, 0], d[0, 0].flip(0).flip(1))
+
+    b = torch.empty(16)
+    initializer.LSTMBias(forget_bias=1.0)("lstm_i2h_bias", b)
+    assert b[4:8].eq(1).all() and b[:4].eq(0).all() and b[8:].eq(0).all()
+
+    mixed = initializer.Mixed([".*bias", ".*"],
+                              [initializer.Zero(), initializer.One()])
+    t1, t2 = torch.randn(3), torch.randn(3)
+    mixed("fc_bias", t1)
+    mixed("fc_weight", t2)
+    assert t1.eq(0).all() and t2.eq(1).all()
 
 
 # ----------------------------------------------------- optimizer states io
