@@ -62,6 +62,52 @@ def do_checkpoint(prefix: str, period: int = 1):
     return _callback
 
 
+def module_checkpoint(mod, prefix: str, period: int = 1,
+                      save_optimizer_states: bool = False):
+    """Epoch-end callback saving from a Module handle (reference
+    callback.py:27-54 module_checkpoint): unlike do_checkpoint it asks the
+    module itself, so optimizer state can ride along."""
+    period = int(max(1, period))
+
+    def _callback(iter_no, sym=None, arg=None, aux=None):
+        if (iter_no + 1) % period == 0:
+            mod.save_checkpoint(prefix, iter_no + 1, save_optimizer_states)
+
+    return _callback
+
+
+def log_train_metric(period: int, auto_reset: bool = False):
+    """Batch-end callback logging the training metric every `period`
+    batches (reference callback.py:93-117)."""
+
+    def _callback(param: BatchEndParam):
+        if param.nbatch % period == 0 and param.eval_metric is not None:
+            name_value = param.eval_metric.get_name_value()
+            for name, value in name_value:
+                logging.info("Iter[%d] Batch[%d] Train-%s=%f", param.epoch,
+                             param.nbatch, name, value)
+            if auto_reset:
+                param.eval_metric.reset()
+
+    return _callback
+
+
+class ProgressBar:
+    """Text progress bar over total batches (reference callback.py:184)."""
+
+    def __init__(self, total: int, length: int = 80):
+        self.bar_len = length
+        self.total = total
+
+    def __call__(self, param: BatchEndParam):
+        import sys
+        count = param.nbatch
+        filled_len = int(round(self.bar_len * count / float(self.total)))
+        percents = int(round(100.0 * count / float(self.total)))
+        prog_bar = "=" * filled_len + "-" * (self.bar_len - filled_len)
+        sys.stdout.write("[%s] %s%s\r" % (prog_bar, percents, "%"))
+
+
 class LogValidationMetricsCallback:
     def __call__(self, param):
         if param.eval_metric is None:

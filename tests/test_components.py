@@ -333,6 +333,38 @@ def test_speedometer_and_checkpoint_callbacks(tmp_path, caplog):
     assert set(arg) == {n for n, _ in net.named_parameters()}
 
 
+def test_callback_tail(tmp_path, caplog):
+    """module_checkpoint / log_train_metric / ProgressBar (reference
+    callback.py:27-117,184-212)."""
+    import logging
+
+    from dtmx import callback
+
+    class FakeMod:
+        saved = None
+
+        def save_checkpoint(self, prefix, epoch, save_optimizer_states=False):
+            FakeMod.saved = (prefix, epoch, save_optimizer_states)
+
+    cb = callback.module_checkpoint(FakeMod(), str(tmp_path / "m"), period=2,
+                                    save_optimizer_states=True)
+    cb(0)
+    assert FakeMod.saved is None  # epoch 1: not on period
+    cb(1)
+    assert FakeMod.saved == (str(tmp_path / "m"), 2, True)
+
+    m = metric.create("acc")
+    m.update([torch.tensor([1.0])], [torch.tensor([[0.1, 0.9]])])
+    with caplog.at_level(logging.INFO):
+        callback.log_train_metric(5, auto_reset=True)(
+            callback.BatchEndParam(epoch=0, nbatch=5, eval_metric=m))
+    assert any("Train-accuracy" in r.getMessage() for r in caplog.records)
+    assert m.num_inst == 0  # auto_reset
+
+    callback.ProgressBar(total=10, length=10)(
+        callback.BatchEndParam(epoch=0, nbatch=5, eval_metric=None))
+
+
 def test_monitor_collects_stats():
     import dtmx
 
