@@ -1401,6 +1401,19 @@ static void launch_gemm_nt(const PA& pa, const PB& pb, const EPI& epi,
   if (M <= 64 && N > 64) {
     // flat 64x256 tile: ResNet's K_out=64 wgrads waste half the 128-row
     // tile on zero rows (see gemm_nt_kernel WM doc)
+    // DTMX_NT_WM1NJ2: 64x128 flat tile (48 KiB LDS -> 3 blocks/CU, 12
+    // waves) — occupancy-vs-re-read probe for this latency-bound class
+    static const bool wm1nj2 = [] {
+      const char* v = getenv("DTMX_NT_WM1NJ2");
+      return v && v[0] == '1';
+    }();
+    if (wm1nj2) {
+      uint32_t tiles_n = ceil_div(N, 128);
+      dim3 grid(tiles_n, nslices);
+      gemm_nt_kernel<2, 1, 2, PA, PB, EPI><<<grid, 256, 0, cur_stream()>>>(
+          pa, pb, epi, ktiles_total, tiles_n, kt_per);
+      return;
+    }
     uint32_t tiles_n = ceil_div(N, 256);
     dim3 grid(tiles_n, nslices);
     if (group && !depth3)
