@@ -397,7 +397,9 @@ class ImageRecordIter(DataIter):
                  preprocess_threads: int = 4, prefetch_buffer: int = 4,
                  seed: int = 0, label_name: str = "softmax_label",
                  rand_crop: bool = False, rand_mirror: bool = False,
-                 resize: int = 0, **kwargs):
+                 resize: int = 0, mean_r: float = 0.0, mean_g: float = 0.0,
+                 mean_b: float = 0.0, std_r: float = 1.0, std_g: float = 1.0,
+                 std_b: float = 1.0, **kwargs):
         super().__init__(batch_size)
         from dtmx.ops.hip import require_ext
 
@@ -412,6 +414,14 @@ class ImageRecordIter(DataIter):
             rand_crop, rand_mirror, resize,
         )
         self.label_name = label_name
+        # channel-wise (x - mean) / std (reference ImageRecordIter
+        # mean_r/g/b + std_r/g/b params, image_aug_default.cc)
+        if (mean_r, mean_g, mean_b) != (0.0, 0.0, 0.0) or \
+                (std_r, std_g, std_b) != (1.0, 1.0, 1.0):
+            self._mean = torch.tensor([mean_r, mean_g, mean_b][:c]).reshape(1, c, 1, 1)
+            self._std = torch.tensor([std_r, std_g, std_b][:c]).reshape(1, c, 1, 1)
+        else:
+            self._mean = self._std = None
 
     @property
     def provide_data(self):
@@ -430,6 +440,8 @@ class ImageRecordIter(DataIter):
             raise StopIteration
         data, label = out
         data = data.permute(0, 3, 1, 2)  # HWC records -> logical NCHW (NHWC memory)
+        if self._mean is not None:
+            data = (data - self._mean) / self._std
         return DataBatch(data=[data], label=[label], pad=0,
                          provide_data=self.provide_data,
                          provide_label=self.provide_label)

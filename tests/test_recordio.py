@@ -53,6 +53,22 @@ def test_image_record_iter(tmp_path):
     assert len(list(it)) == 4
 
 
+def test_mean_std_normalization(tmp_path):
+    """mean_r/g/b + std_r/g/b channel normalization (reference
+    ImageRecordIter params, image_aug_default.cc mean/std step)."""
+    from dtmx.io import ImageRecordIter
+
+    path, x, y = _write(tmp_path)
+    plain = next(iter(ImageRecordIter(path, (3, 8, 8), batch_size=16)))
+    norm = next(iter(ImageRecordIter(path, (3, 8, 8), batch_size=16,
+                                     mean_r=0.5, mean_g=0.25, mean_b=0.0,
+                                     std_r=2.0, std_g=1.0, std_b=4.0)))
+    mean = torch.tensor([0.5, 0.25, 0.0]).reshape(1, 3, 1, 1)
+    std = torch.tensor([2.0, 1.0, 4.0]).reshape(1, 3, 1, 1)
+    assert torch.allclose(norm.data[0], (plain.data[0] - mean) / std,
+                          atol=1e-6)
+
+
 def test_sharding(tmp_path):
     from dtmx.io import ImageRecordIter
 
