@@ -46,5 +46,22 @@ def main(bs=1024):
               f"{min_gb/s/1e3:5.2f} TB/s floor", flush=True)
 
 
+def triad():
+    """bf16 RW-mix ceilings for the BN elementwise kernels: what does THIS
+    box sustain on d=a+b (2R+1W, the bn_bwd_dx mix) and b=a*s (1R+1W, the
+    bn_apply mix)? Quantifies how close 4.9-5.0 TB/s is to attainable."""
+    n = 1024 * 256 * 56 * 56  # the bn micro shape's element count
+    a = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn_like(a)
+    d = torch.empty_like(a)
+    s = timeit(lambda: torch.add(a, b, out=d), iters=20)
+    print(f"triad d=a+b  (2R+1W): {3 * n * 2 / s / 1e12:5.2f} TB/s")
+    s = timeit(lambda: torch.mul(a, 1.5, out=d), iters=20)
+    print(f"scale b=a*s  (1R+1W): {2 * n * 2 / s / 1e12:5.2f} TB/s")
+
+
 if __name__ == "__main__":
-    main(int(sys.argv[1]) if len(sys.argv) > 1 else 1024)
+    if len(sys.argv) > 1 and sys.argv[1] == "triad":
+        triad()
+    else:
+        main(int(sys.argv[1]) if len(sys.argv) > 1 else 1024)
