@@ -562,6 +562,128 @@ __global__ void bn_bwd_dx_coef_kernel(const elem_t* __restrict__ x,
   }
 }
 
+// ---- fixed-column variants: thread owns ONE channel-vector column --------
+// The strided-chunk kernels re-load the per-channel tables (and run a
+// FastDiv) for every 16-B chunk; torch's pure d=a+b triad reaches 6.06 TB/s
+// on this box where bn_bwd_dx measured 4.87 — the table traffic is the gap.
+// Owning a fixed cv per thread hoists the tables into registers before the
+// row loop; coalescing stays intact (cpb consecutive lanes cover cpb*16 B
+// of one row). Geometry = bn_stats_kernel's (cv, strided rows).
+template <typename elem_t>
+__global__ void bn_apply_col_kernel(const elem_t* __restrict__ x,
+                                    elem_t* __restrict__ y,
+                                    const float* __restrict__ scale,
+                                    const float* __restrict__ shift,
+                                    const elem_t* __restrict__ residual,
+                                    uint32_t rows, uint32_t cvecs, uint32_t cpb,
+                                    uint32_t rows_per_block, int relu) {
+  using V8 = typename E8<elem_t>::v8;
+  const uint32_t t = threadIdx.x;
+  const uint32_t cv = blockIdx.x * cpb + t % cpb;
+  if (cv >= cvecs) return;
+  const uint32_t rstep = blockDim.x / cpb;
+  const uint32_t r0 = blockIdx.y * rows_per_block + t / cpb;
+  const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
+  float sc[8], sh[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    sc[e] = scale[cv * 8 + e];
+    sh[e] = shift[cv * 8 + e];
+  }
+  auto body = [&](size_t off, V8 v, V8 res) {
+    V8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float r = (float)v[e] * sc[e] + sh[e];
+      if (residual) r += (float)res[e];
+      if (relu) r = fmaxf(r, 0.f);
+      o[e] = (elem_t)r;
+    }
+    *(V8*)(y + off) = o;
+  };
+  V8 zed = {};
+  uint32_t r = r0;
+  for (; r + rstep < r1; r += 2 * rstep) {
+    size_t o0 = ((size_t)r * cvecs + cv) * 8;
+    size_t o1 = ((size_t)(r + rstep) * cvecs + cv) * 8;
+    V8 v0 = *(const V8*)(x + o0), v1 = *(const V8*)(x + o1);
+    V8 r0v = residual ? *(const V8*)(residual + o0) : zed;
+    V8 r1v = residual ? *(const V8*)(residual + o1) : zed;
+    body(o0, v0, r0v);
+    body(o1, v1, r1v);
+  }
+  for (; r < r1; r += rstep) {
+    size_t o0 = ((size_t)r * cvecs + cv) * 8;
+    body(o0, *(const V8*)(x + o0),
+         residual ? *(const V8*)(residual + o0) : zed);
+  }
+}
+
+template <typename elem_t>
+__global__ void bn_bwd_dx_col_kernel(const elem_t* __restrict__ x,
+                                     const elem_t* __restrict__ dy,
+                                     const elem_t* __restrict__ y,
+                                     const float* __restrict__ ka,
+                                     const float* __restrict__ kb,
+                                     const float* __restrict__ kc,
+                                     elem_t* __restrict__ dx,
+                                     elem_t* __restrict__ dres,
+                                     uint32_t rows, uint32_t cvecs,
+                                     uint32_t cpb, uint32_t rows_per_block,
+                                     int relu) {
+  using V8 = typename E8<elem_t>::v8;
+  const uint32_t t = threadIdx.x;
+  const uint32_t cv = blockIdx.x * cpb + t % cpb;
+  if (cv >= cvecs) return;
+  const uint32_t rstep = blockDim.x / cpb;
+  const uint32_t r0 = blockIdx.y * rows_per_block + t / cpb;
+  const uint32_t r1 = min((blockIdx.y + 1) * rows_per_block, rows);
+  float a8[8], b8[8], c8[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    a8[e] = ka[cv * 8 + e];
+    b8[e] = kb[cv * 8 + e];
+    c8[e] = kc[cv * 8 + e];
+  }
+  auto body = [&](size_t off, V8 xv, V8 gv, V8 yv) {
+    V8 o, om;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float g = (float)gv[e];
+      if (relu && (float)yv[e] <= 0.f) g = 0.f;
+      if (dres) om[e] = (elem_t)g;
+      o[e] = (elem_t)(a8[e] * g + c8[e] * (float)xv[e] + b8[e]);
+    }
+    *(V8*)(dx + off) = o;
+    if (dres) *(V8*)(dres + off) = om;
+  };
+  V8 zed = {};
+  uint32_t r = r0;
+  for (; r + rstep < r1; r += 2 * rstep) {
+    size_t o0 = ((size_t)r * cvecs + cv) * 8;
+    size_t o1 = ((size_t)(r + rstep) * cvecs + cv) * 8;
+    V8 x0 = *(const V8*)(x + o0), x1 = *(const V8*)(x + o1);
+    V8 g0 = *(const V8*)(dy + o0), g1 = *(const V8*)(dy + o1);
+    V8 y0 = relu ? *(const V8*)(y + o0) : zed;
+    V8 y1 = relu ? *(const V8*)(y + o1) : zed;
+    body(o0, x0, g0, y0);
+    body(o1, x1, g1, y1);
+  }
+  for (; r < r1; r += rstep) {
+    size_t o0 = ((size_t)r * cvecs + cv) * 8;
+    body(o0, *(const V8*)(x + o0), *(const V8*)(dy + o0),
+         relu ? *(const V8*)(y + o0) : zed);
+  }
+}
+
+static bool bn_col_on() {
+  static const bool on = [] {
+    const char* e = getenv("DTMX_BN_COL");
+    return !e || e[0] != '0';  // default ON (A/B'd on-box)
+  }();
+  return on;
+}
+
 static bool bn_dx_coef_on() {
   static const bool on = [] {
     const char* e = getenv("DTMX_BN_DX_COEF");
@@ -590,6 +712,53 @@ static void bn_grid(uint32_t rows, uint32_t cvecs, uint32_t cpb, dim3& grid,
   rows_per_block = (rows + rb - 1) / rb;
   rb = (rows + rows_per_block - 1) / rows_per_block;
   grid = dim3((cvecs + cpb - 1) / cpb, rb);
+}
+
+// dispatch helper: fixed-column kernel (tables hoisted to registers) by
+// default, strided-chunk kernel under DTMX_BN_COL=0
+template <typename elem_t>
+static void launch_bn_apply(const elem_t* x, elem_t* y, const float* scale,
+                            const float* shift, const elem_t* residual,
+                            uint32_t rows, uint32_t cvecs, int relu,
+                            hipStream_t s) {
+  if (bn_col_on()) {
+    uint32_t cpb = bn_cpb(cvecs);
+    dim3 grid;
+    uint32_t rpb;
+    bn_grid(rows, cvecs, cpb, grid, rpb);
+    bn_apply_col_kernel<<<grid, 256, 0, s>>>(x, y, scale, shift, residual,
+                                             rows, cvecs, cpb, rpb, relu);
+  } else {
+    uint32_t total8 = rows * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_apply_kernel<<<blocks, 256, 0, s>>>(x, y, scale, shift, residual,
+                                           total8, dcv, relu);
+  }
+}
+
+template <typename elem_t>
+static void launch_bn_dx_coef(const elem_t* x, const elem_t* dy,
+                              const elem_t* y, const float* ka,
+                              const float* kb, const float* kc, elem_t* dx,
+                              elem_t* dres, uint32_t rows, uint32_t cvecs,
+                              int relu, hipStream_t s) {
+  if (bn_col_on()) {
+    uint32_t cpb = bn_cpb(cvecs);
+    dim3 grid;
+    uint32_t rpb;
+    bn_grid(rows, cvecs, cpb, grid, rpb);
+    bn_bwd_dx_col_kernel<<<grid, 256, 0, s>>>(x, dy, y, ka, kb, kc, dx, dres,
+                                              rows, cvecs, cpb, rpb, relu);
+  } else {
+    uint32_t total8 = rows * cvecs;
+    FastDiv dcv;
+    dcv.init(cvecs);
+    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
+    bn_bwd_dx_coef_kernel<<<blocks, 256, 0, s>>>(x, dy, y, ka, kb, kc, dx,
+                                                 dres, total8, dcv, relu);
+  }
 }
 
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamma,
@@ -651,15 +820,11 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x, const at::Tensor& gamm
         save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
         scale.data_ptr<float>(), shift.data_ptr<float>(), C, nslabs, ncv, rows,
         momentum, eps);
-    uint32_t total8 = rows * cvecs;
-    FastDiv dcv;
-    dcv.init(cvecs);
-    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-    bn_apply_kernel<<<blocks, 256, 0, s>>>(
+    launch_bn_apply<elem_t>(
         (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
         scale.data_ptr<float>(), shift.data_ptr<float>(),
         residual.has_value() ? (const elem_t*)residual->data_ptr() : nullptr,
-        total8, dcv, fuse_relu ? 1 : 0);
+        rows, cvecs, fuse_relu ? 1 : 0, s);
   });
   return {y, save_mean, save_invstd};
 }
@@ -681,15 +846,11 @@ at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
         (const elem_t*)gamma.data_ptr(), (const elem_t*)beta.data_ptr(),
         running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
         scale.data_ptr<float>(), shift.data_ptr<float>(), C, eps);
-    uint32_t total8 = rows * cvecs;
-    FastDiv dcv;
-    dcv.init(cvecs);
-    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-    bn_apply_kernel<<<blocks, 256, 0, s>>>(
+    launch_bn_apply<elem_t>(
         (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
         scale.data_ptr<float>(), shift.data_ptr<float>(),
         residual.has_value() ? (const elem_t*)residual->data_ptr() : nullptr,
-        total8, dcv, fuse_relu ? 1 : 0);
+        rows, cvecs, fuse_relu ? 1 : 0, s);
   });
   return y;
 }
@@ -739,12 +900,12 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
           (const elem_t*)gamma.data_ptr(), save_mean.data_ptr<float>(),
           save_invstd.data_ptr<float>(), ka.data_ptr<float>(),
           kb.data_ptr<float>(), kc.data_ptr<float>(), C, 1.f / rows);
-      bn_bwd_dx_coef_kernel<<<blocks, 256, 0, s>>>(
+      launch_bn_dx_coef<elem_t>(
           (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
           (const elem_t*)y.data_ptr(), ka.data_ptr<float>(),
           kb.data_ptr<float>(), kc.data_ptr<float>(), (elem_t*)dx.data_ptr(),
-          want_dres ? (elem_t*)dres.data_ptr() : nullptr, total8, dcv,
-          fuse_relu ? 1 : 0);
+          want_dres ? (elem_t*)dres.data_ptr() : nullptr, rows, cvecs,
+          fuse_relu ? 1 : 0, s);
     } else {
       bn_bwd_dx_kernel<<<blocks, 256, 0, s>>>(
           (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
@@ -876,15 +1037,11 @@ std::vector<at::Tensor> bn_fwd_presummed(
         save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
         scale.data_ptr<float>(), shift.data_ptr<float>(), C, /*nslabs=*/1,
         (uint32_t)count, momentum, eps);
-    uint32_t total8 = N * H * W * cvecs;
-    FastDiv dcv;
-    dcv.init(cvecs);
-    uint32_t blocks = std::min<uint32_t>((total8 + 255) / 256, 2048);
-    bn_apply_kernel<<<blocks, 256, 0, s>>>(
+    launch_bn_apply<elem_t>(
         (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(),
         scale.data_ptr<float>(), shift.data_ptr<float>(),
         residual.has_value() ? (const elem_t*)residual->data_ptr() : nullptr,
-        total8, dcv, fuse_relu ? 1 : 0);
+        N * H * W, cvecs, fuse_relu ? 1 : 0, s);
   });
   return {y, save_mean, save_invstd};
 }
@@ -944,12 +1101,12 @@ std::vector<at::Tensor> bn_bwd_dx_presummed(
           (const elem_t*)gamma.data_ptr(), save_mean.data_ptr<float>(),
           save_invstd.data_ptr<float>(), ka.data_ptr<float>(),
           kb.data_ptr<float>(), kc.data_ptr<float>(), C, 1.f / (float)count);
-      bn_bwd_dx_coef_kernel<<<blocks, 256, 0, s>>>(
+      launch_bn_dx_coef<elem_t>(
           (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
           (const elem_t*)y.data_ptr(), ka.data_ptr<float>(),
           kb.data_ptr<float>(), kc.data_ptr<float>(), (elem_t*)dx.data_ptr(),
-          want_dres ? (elem_t*)dres.data_ptr() : nullptr, total8, dcv,
-          fuse_relu ? 1 : 0);
+          want_dres ? (elem_t*)dres.data_ptr() : nullptr, N * H * W, cvecs,
+          fuse_relu ? 1 : 0, s);
     } else {
       bn_bwd_dx_kernel<<<blocks, 256, 0, s>>>(
           (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
