@@ -79,6 +79,7 @@ at::Tensor quantize_2bit(const at::Tensor&, at::Tensor, double);
 at::Tensor dequantize_2bit(const at::Tensor&, long, double);
 // gemm256.hip
 at::Tensor gemm256_nt(const at::Tensor&, const at::Tensor&);
+at::Tensor gemm256n_nt(const at::Tensor&, const at::Tensor&);
 // indexing.hip
 at::Tensor take_fwd(const at::Tensor&, const at::Tensor&);
 at::Tensor take_bwd(const at::Tensor&, const at::Tensor&, long);
@@ -126,6 +127,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &dtmx::layer_norm_fwd);
   m.def("layer_norm_bwd", &dtmx::layer_norm_bwd);
   m.def("gemm256_nt", &dtmx::gemm256_nt);
+  m.def("gemm256n_nt", &dtmx::gemm256n_nt);
   m.def("take_fwd", &dtmx::take_fwd);
   m.def("take_bwd", &dtmx::take_bwd);
   m.def("quantize_2bit", &dtmx::quantize_2bit);
