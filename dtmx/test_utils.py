@@ -58,3 +58,26 @@ def rand_ndarray(shape, dtype=torch.float32, scale=1.0, seed=None) -> torch.Tens
     if seed is not None:
         g.manual_seed(seed)
     return (torch.randn(*shape, generator=g) * scale).to(dtype)
+
+
+def same(a, b) -> bool:
+    """Exact equality (reference test_utils.py:454 same)."""
+    a = a.detach().cpu() if isinstance(a, torch.Tensor) else torch.as_tensor(a)
+    b = b.detach().cpu() if isinstance(b, torch.Tensor) else torch.as_tensor(b)
+    return bool(torch.equal(a, b))
+
+
+def assert_almost_equal_ignore_nan(a, b, rtol: float = 1e-5,
+                                   atol: float = 1e-8, names=("a", "b")):
+    """Compare ignoring positions where EITHER side is NaN (reference
+    test_utils.py:519)."""
+    a = a.detach().cpu().float() if isinstance(a, torch.Tensor) else torch.as_tensor(a, dtype=torch.float32)
+    b = b.detach().cpu().float() if isinstance(b, torch.Tensor) else torch.as_tensor(b, dtype=torch.float32)
+    mask = ~(torch.isnan(a) | torch.isnan(b))
+    assert_almost_equal(a[mask], b[mask], rtol=rtol, atol=atol, names=names)
+
+
+def default_context() -> torch.device:
+    """The default test device (reference test_utils.py:53 default_context):
+    cuda:0 when a GPU is visible, else cpu."""
+    return torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
