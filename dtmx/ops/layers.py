@@ -94,6 +94,32 @@ class BatchNorm2dNHWC(nn.Module):
         )
 
 
+class GroupedConv2dNHWC(nn.Module):
+    """Grouped / depthwise conv (mobilenet, resnext). The hand HIP conv
+    kernels are dense-gather only, so groups>1 routes through torch's conv
+    on channels_last memory (MIOpen on ROCm) — the documented long-tail
+    substrate path; groups==1 uses Conv2dNHWC instead."""
+
+    def __init__(self, in_channels, out_channels, kernel_size, stride=1,
+                 padding=0, groups=1, bias=False):
+        super().__init__()
+        assert in_channels % groups == 0 and out_channels % groups == 0
+        self.stride = stride
+        self.padding = padding
+        self.groups = groups
+        w = torch.empty(out_channels, in_channels // groups, kernel_size,
+                        kernel_size)
+        nn.init.kaiming_normal_(w, mode="fan_in", nonlinearity="relu")
+        self.weight = nn.Parameter(w)
+        self.bias = nn.Parameter(torch.zeros(out_channels)) if bias else None
+
+    def forward(self, x):
+        import torch.nn.functional as F
+        return F.conv2d(x, self.weight.to(x.dtype),
+                        self.bias.to(x.dtype) if self.bias is not None else None,
+                        self.stride, self.padding, 1, self.groups)
+
+
 class ReLU(nn.Module):
     def forward(self, x):
         return DF.relu(x)

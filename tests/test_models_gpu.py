@@ -19,6 +19,13 @@ CASES = [
     ("inception-bn", {}, (4, 3, 224, 224), 0.002),
     ("lenet", {}, (16, 1, 28, 28), 0.01),
     ("mlp", {}, (32, 784), 0.01),
+    # round-2 zoo additions (reference symbols/ tail)
+    ("googlenet", {}, (4, 3, 224, 224), 0.002),
+    ("inception-v4", {}, (2, 3, 299, 299), 0.002),
+    ("inception-resnet-v2", {}, (2, 3, 299, 299), 0.002),
+    ("mobilenet", {}, (8, 3, 224, 224), 0.002),
+    ("mobilenetv2", {}, (8, 3, 224, 224), 0.002),
+    ("resnext", {"num_layers": 50}, (4, 3, 224, 224), 0.01),
 ]
 
 
