@@ -26,7 +26,12 @@ from test_elastic import _free_port, _spawn, _wait_epoch  # noqa: E402
 
 
 @pytest.mark.timeout(300)
-def test_mid_epoch_death_fails_loudly_and_is_pruned(tmp_path):
+def test_mid_epoch_death_fails_loudly_and_is_pruned(tmp_path, monkeypatch):
+    # scheduler-side staleness threshold (module-level default 30 s) must
+    # sit INSIDE the worker heartbeat cadence set below, or the prune loop
+    # races the survivor's own exit going stale
+    from dtmx.parallel import rendezvous as rz
+    monkeypatch.setattr(rz, "_HB_TIMEOUT", 3.0)
     port = _free_port()
     sched = Scheduler("127.0.0.1", port, ["127.0.0.1#0", "127.0.0.1#1"],
                       hostfile=str(tmp_path / "hosts"))
