@@ -530,6 +530,43 @@ def test_bench_json_contract(tmp_path):
     assert d["config"]["parallelism"] == "dp1"
 
 
+def test_bench_dist_contract_world2(tmp_path):
+    """bench.py under the driver's multi-rank launch pattern (torchrun env:
+    RANK/WORLD_SIZE/LOCAL_RANK/MASTER_*), world 2 on gloo/CPU: both ranks
+    finish, rank 0 prints ONE whole-job JSON line with n_gpus=2 — de-risks
+    the round-end 8-GPU scaling run."""
+    import json as _json
+    import socket
+    import subprocess
+    import sys as _sys
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank),
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                   DTMX_BACKEND="gloo", DTMX_BENCH_AUX="0")
+        procs.append(subprocess.Popen(
+            [_sys.executable, os.path.join(ROOT, "bench.py"), "--gpus", "2",
+             "--steps", "1", "--warmup", "0", "--batch-size", "4"],
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, env=env))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=600)
+        assert p.returncode == 0, err.decode()[-2000:]
+        outs.append(out.decode())
+    json_lines = [ln for o in outs for ln in o.splitlines() if ln.startswith("{")]
+    assert len(json_lines) == 1, json_lines  # rank 0 only
+    d = _json.loads(json_lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 8  # whole-job aggregate
+
+
 def test_random_api():
     import dtmx.random as rnd
 
