@@ -21,11 +21,13 @@ if _os.environ.get("DTMX_BLOCKING", "0") == "1":
 
 from . import callback, context, initializer, io, lr_scheduler, metric, model, monitor
 from . import gluon, kvstore, models, ndarray, optimizer, parallel, profiler, recordio
-from . import image, random, rnn, visualization
+from . import autograd, image, nd, random, rnn, visualization
 from . import module as mod
 from .context import Context, cpu, gpu, num_gpus
 from .module import Module
 
-nd = ndarray
+# dtmx.nd is the mx.nd-style op namespace (nd.py); it re-exports the
+# byte-compatible save/load/array/zeros from dtmx.ndarray, so the round-1
+# `nd = ndarray` alias is subsumed
 
 __version__ = "0.1.0"
