@@ -13,6 +13,7 @@ from typing import Dict, Optional, Sequence
 
 import torch
 import torch.nn as nn
+import torch.nn as _tnn
 
 from . import kvstore as kvs
 from .optimizer import Optimizer, create as opt_create, get_updater
@@ -104,3 +105,79 @@ class SoftmaxCrossEntropyLoss(nn.Module):
         return torch.nn.functional.cross_entropy(
             pred.float(), label.reshape(-1).long(), reduction="none"
         )
+
+
+class L1Loss(nn.Module):
+    def forward(self, pred, label):
+        return (pred - label.reshape(pred.shape)).abs()
+
+
+class _GluonNN:
+    """`gluon.nn` namespace (reference gluon/nn/): gluon layer names mapped
+    onto the substrate's modules with gluon-style ctor args. Lazy in_units /
+    in_channels (gluon's deferred shape init) is NOT mirrored — pass the
+    input size explicitly, as dtmx layers do."""
+
+    Sequential = _tnn.Sequential
+    HybridSequential = _tnn.Sequential
+    Flatten = _tnn.Flatten
+    Dropout = _tnn.Dropout
+
+    @staticmethod
+    def Dense(units, in_units, activation=None, use_bias=True):
+        lin = _tnn.Linear(in_units, units, bias=use_bias)
+        if activation is None:
+            return lin
+        acts = {"relu": _tnn.ReLU(), "sigmoid": _tnn.Sigmoid(), "tanh": _tnn.Tanh()}
+        return _tnn.Sequential(lin, acts[activation])
+
+    @staticmethod
+    def Conv2D(channels, kernel_size, in_channels, strides=1, padding=0,
+               use_bias=True, activation=None):
+        conv = _tnn.Conv2d(in_channels, channels, kernel_size, strides, padding,
+                         bias=use_bias)
+        if activation is None:
+            return conv
+        return _tnn.Sequential(conv, {"relu": _tnn.ReLU()}[activation])
+
+    @staticmethod
+    def MaxPool2D(pool_size=2, strides=None, padding=0):
+        return _tnn.MaxPool2d(pool_size, strides or pool_size, padding)
+
+    @staticmethod
+    def AvgPool2D(pool_size=2, strides=None, padding=0):
+        return _tnn.AvgPool2d(pool_size, strides or pool_size, padding)
+
+    @staticmethod
+    def BatchNorm(in_channels, momentum=0.9, epsilon=1e-5):
+        return _tnn.BatchNorm2d(in_channels, eps=epsilon, momentum=1 - momentum)
+
+    @staticmethod
+    def Activation(activation):
+        return {"relu": _tnn.ReLU(), "sigmoid": _tnn.Sigmoid(),
+                "tanh": _tnn.Tanh(), "softrelu": _tnn.Softplus()}[activation]
+
+    @staticmethod
+    def GlobalAvgPool2D():
+        return _tnn.AdaptiveAvgPool2d(1)
+
+
+class _GluonLoss:
+    L2Loss = L2Loss
+    L1Loss = L1Loss
+    SoftmaxCrossEntropyLoss = SoftmaxCrossEntropyLoss
+    SoftmaxCELoss = SoftmaxCrossEntropyLoss
+
+
+class _GluonData:
+    """`gluon.data` essentials (reference gluon/data/): torch's dataset/
+    loader are the substrate equivalents."""
+    from torch.utils.data import DataLoader, Dataset, TensorDataset
+    ArrayDataset = TensorDataset
+
+
+nn_ = _GluonNN  # gluon.nn / gluon.loss / gluon.data namespaces
+loss = _GluonLoss
+data = _GluonData
+globals()["nn"] = _GluonNN  # keep `from dtmx.gluon import nn` working
+                            # (the torch.nn import above is module-internal)
