@@ -9,7 +9,7 @@ single/static-distributed workflow.
 """
 from __future__ import annotations
 
-from typing import Dict, Optional, Sequence
+from typing import Optional
 
 import torch
 import torch.nn as nn
