@@ -5,7 +5,7 @@ from __future__ import annotations
 
 from . import (alexnet, googlenet, inception_bn, inception_resnet_v2,
                inception_v3, inception_v4, lenet, mlp, mobilenet, resnet,
-               resnext, vgg)
+               resnet_v2, resnext, vgg)
 
 _REGISTRY = {
     "resnet": resnet.get_symbol,
@@ -26,6 +26,8 @@ _REGISTRY = {
     "mobilenet": mobilenet.get_symbol,
     "mobilenetv2": mobilenet.get_symbol_v2,
     "resnext": resnext.get_symbol,
+    # the reference's own resnet.py topology (pre-activation v2)
+    "resnet-v2": resnet_v2.get_symbol,
 }
 
 

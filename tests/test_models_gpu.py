@@ -26,6 +26,7 @@ CASES = [
     ("mobilenet", {}, (8, 3, 224, 224), 0.002),
     ("mobilenetv2", {}, (8, 3, 224, 224), 0.002),
     ("resnext", {"num_layers": 50}, (4, 3, 224, 224), 0.01),
+    ("resnet-v2", {"num_layers": 50}, (4, 3, 224, 224), 0.01),
 ]
 
 

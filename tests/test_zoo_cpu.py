@@ -13,6 +13,8 @@ CASES = [
     ("mobilenet", {}, (1, 3, 224, 224)),
     ("mobilenetv2", {}, (1, 3, 224, 224)),
     ("resnext", {"num_layers": 50}, (1, 3, 224, 224)),
+    ("resnet-v2", {"num_layers": 50}, (1, 3, 224, 224)),
+    ("resnet-v2", {"num_layers": 18}, (1, 3, 32, 32)),
 ]
 
 
