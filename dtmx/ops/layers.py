@@ -146,6 +146,25 @@ class GlobalAvgPool(nn.Module):
         return DF.global_avg_pool(x)
 
 
+class LRN(nn.Module):
+    """Cross-channel local response norm (reference mx.sym.LRN,
+    src/operator/nn/lrn.cc): y = x / (knorm + alpha/nsize * sum_n x^2)^beta.
+    torch's local_response_norm computes exactly this convention — the
+    substrate path for this long-tail op (alexnet/googlenet era only)."""
+
+    def __init__(self, nsize=5, alpha=1e-4, beta=0.75, knorm=2.0):
+        super().__init__()
+        self.nsize = nsize
+        self.alpha = alpha
+        self.beta = beta
+        self.knorm = knorm
+
+    def forward(self, x):
+        import torch.nn.functional as F
+        return F.local_response_norm(x, self.nsize, alpha=self.alpha,
+                                     beta=self.beta, k=self.knorm)
+
+
 class LinearBF16(nn.Module):
     def __init__(self, in_features, out_features, bias=True):
         super().__init__()
