@@ -138,7 +138,10 @@ def abs(t):  # noqa: A001
 
 
 def transpose(t, axes=None):
-    return t.t() if axes is None and t.dim() == 2 else t.permute(axes)
+    # no axes = reverse all dims (mx.nd.transpose semantics)
+    if axes is None:
+        axes = tuple(range(t.dim() - 1, -1, -1))
+    return t.permute(axes)
 
 
 def reshape(t, shape):
