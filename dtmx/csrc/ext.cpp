@@ -80,6 +80,8 @@ at::Tensor dequantize_2bit(const at::Tensor&, long, double);
 // gemm256.hip
 at::Tensor gemm256_nt(const at::Tensor&, const at::Tensor&);
 at::Tensor gemm256n_nt(const at::Tensor&, const at::Tensor&);
+at::Tensor lrn_fwd(const at::Tensor&, double, double, double);
+at::Tensor lrn_bwd(const at::Tensor&, const at::Tensor&, double, double, double);
 // indexing.hip
 at::Tensor take_fwd(const at::Tensor&, const at::Tensor&);
 at::Tensor take_bwd(const at::Tensor&, const at::Tensor&, long);
@@ -128,6 +130,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_bwd", &dtmx::layer_norm_bwd);
   m.def("gemm256_nt", &dtmx::gemm256_nt);
   m.def("gemm256n_nt", &dtmx::gemm256n_nt);
+  m.def("lrn_fwd", &dtmx::lrn_fwd);
+  m.def("lrn_bwd", &dtmx::lrn_bwd);
   m.def("take_fwd", &dtmx::take_fwd);
   m.def("take_bwd", &dtmx::take_bwd);
   m.def("quantize_2bit", &dtmx::quantize_2bit);

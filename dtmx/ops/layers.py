@@ -160,13 +160,11 @@ class LRN(nn.Module):
         self.knorm = knorm
 
     def forward(self, x):
-        # substrate path: a cumsum-over-channels reformulation measured
-        # SLOWER on GPU (18.8k vs 28.1k img/s alexnet bs2048 — the channel
-        # scan + index_select materializations beat the pad+avg_pool3d
-        # lowering it was meant to avoid), so the plain torch op stays
-        import torch.nn.functional as F
-        return F.local_response_norm(x, self.nsize, alpha=self.alpha,
-                                     beta=self.beta, k=self.knorm)
+        # HIP window-5 kernel on GPU 16-bit; torch op otherwise. (A
+        # cumsum-over-channels torch reformulation measured SLOWER than
+        # even pad+avg_pool3d — 18.8k vs 28.1k img/s alexnet bs2048.)
+        return DF.local_response_norm(x, self.nsize, alpha=self.alpha,
+                                      beta=self.beta, knorm=self.knorm)
 
 
 class LinearBF16(nn.Module):

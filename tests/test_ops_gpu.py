@@ -261,6 +261,28 @@ def test_relu_and_add_relu():
     assert_close(z, F.relu(a.float() + b.float()), name="add_relu")
 
 
+def test_lrn_fwd_bwd():
+    """HIP window-5 LRN vs torch fp32 reference (reference
+    src/operator/nn/lrn.cc; alexnet.py:34)."""
+    import torch
+
+    from dtmx.ops import functional as DF
+    torch.manual_seed(21)
+    for C in (8, 64, 192):
+        x32 = torch.randn(4, C, 9, 9)
+        x = (x32.to(torch.bfloat16).to(DEV)
+             .contiguous(memory_format=torch.channels_last).requires_grad_(True))
+        y = DF.local_response_norm(x, 5, alpha=1e-4, beta=0.75, knorm=2.0)
+        xr = x32.to(torch.bfloat16).float().requires_grad_(True)
+        ref = F.local_response_norm(xr, 5, alpha=1e-4, beta=0.75, k=2.0)
+        assert_close(y, ref, name=f"lrn_fwd_C{C}")
+        dy = torch.randn(4, C, 9, 9)
+        y.backward(dy.to(torch.bfloat16).to(DEV)
+                     .contiguous(memory_format=torch.channels_last))
+        ref.backward(dy.to(torch.bfloat16).float())
+        assert_close(x.grad, xr.grad, name=f"lrn_bwd_C{C}")
+
+
 # --------------------------------------------------------------- softmax-ce
 
 def test_softmax_ce():
