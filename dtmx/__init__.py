@@ -21,7 +21,7 @@ if _os.environ.get("DTMX_BLOCKING", "0") == "1":
 
 from . import callback, context, initializer, io, lr_scheduler, metric, model, monitor
 from . import gluon, kvstore, models, ndarray, optimizer, parallel, profiler, recordio
-from . import autograd, image, nd, random, rnn, visualization
+from . import autograd, image, nd, random, rnn, symbol, visualization
 from . import module as mod
 from .context import Context, cpu, gpu, num_gpus
 from .module import Module
