@@ -567,6 +567,22 @@ def test_bench_dist_contract_world2(tmp_path):
     assert d["config"]["global_batch"] == 8  # whole-job aggregate
 
 
+def test_symbol_stub_and_lrn_fallback():
+    """dtmx.symbol raises with directions (no symbolic layer by design);
+    LRN's CPU path matches torch's reference op."""
+    import pytest as _pytest
+    import torch.nn.functional as F
+
+    import dtmx
+    from dtmx.ops.layers import LRN
+
+    with _pytest.raises(AttributeError, match="no symbolic graph"):
+        dtmx.symbol.Variable("data")
+    x = torch.randn(2, 16, 5, 5)
+    torch.testing.assert_close(LRN()(x),
+                               F.local_response_norm(x, 5, 1e-4, 0.75, 2.0))
+
+
 def test_random_api():
     import dtmx.random as rnd
 
