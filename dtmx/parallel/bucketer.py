@@ -236,12 +236,6 @@ class GradBucketer:
             out /= world
         flat.view(-1).copy_(out)
 
-    def _on_grad_ready_compressed(self, param):
-        bi = self._param_bucket[id(param)]
-        self._pending[bi] -= 1
-        if self._pending[bi] == 0 and self._enabled:
-            self._compressed_reduce(bi)
-
     def rebuild_after_membership_change(self):
         """Communicator changed (elastic re-form): nothing to re-shard in the
         replicated-DP layout; hooks and buffers stay valid, only the process
