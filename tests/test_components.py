@@ -567,6 +567,29 @@ def test_bench_dist_contract_world2(tmp_path):
     assert d["config"]["global_batch"] == 8  # whole-job aggregate
 
 
+def test_gluon_namespaces():
+    """gluon.nn / gluon.loss / gluon.data map gluon ctor idioms onto the
+    substrate (reference gluon/nn, gluon/loss, gluon/data)."""
+    from dtmx import gluon
+
+    net = gluon.nn.Sequential(
+        gluon.nn.Dense(8, in_units=4, activation="relu"),
+        gluon.nn.Dense(2, in_units=8))
+    y = net(torch.randn(3, 4))
+    assert y.shape == (3, 2)
+    conv = gluon.nn.Conv2D(6, 3, in_channels=2, padding=1)
+    assert conv(torch.randn(1, 2, 5, 5)).shape == (1, 6, 5, 5)
+    assert gluon.nn.MaxPool2D(2)(torch.randn(1, 2, 4, 4)).shape == (1, 2, 2, 2)
+    losses = gluon.loss.SoftmaxCrossEntropyLoss()(y, torch.tensor([0, 1, 0]))
+    assert losses.shape == (3,)
+    assert gluon.loss.L1Loss()(y, torch.zeros(3, 2)).shape == (3, 2)
+    ds = gluon.data.ArrayDataset(torch.randn(10, 4), torch.arange(10))
+    dl = gluon.data.DataLoader(ds, batch_size=5)
+    assert len(list(dl)) == 2
+    # Block/HybridBlock are nn.Module aliases
+    assert gluon.Block is torch.nn.Module
+
+
 def test_symbol_stub_and_lrn_fallback():
     """dtmx.symbol raises with directions (no symbolic layer by design);
     LRN's CPU path matches torch's reference op."""
