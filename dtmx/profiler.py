@@ -73,3 +73,17 @@ def dump(aggregate: bool = False):
     if aggregate and _prof is not None:
         logging.info("\n%s", dumps())
     set_state("stop")
+
+
+def memory_summary(device=None) -> str:
+    """Device-memory profile (reference profiler memory profiling /
+    gpu_memory_profiler): peak allocated/reserved plus the caching
+    allocator's pool breakdown, from the substrate's accounting."""
+    if not torch.cuda.is_available():
+        return "(no GPU: memory profiling is device-side)"
+    dev = device if device is not None else torch.cuda.current_device()
+    alloc = torch.cuda.max_memory_allocated(dev) / 2 ** 30
+    reserv = torch.cuda.max_memory_reserved(dev) / 2 ** 30
+    head = (f"peak allocated {alloc:.2f} GiB, peak reserved {reserv:.2f} GiB "
+            f"(of 288 GB HBM3E)\n")
+    return head + torch.cuda.memory_summary(dev, abbreviated=True)
