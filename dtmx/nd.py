@@ -175,6 +175,40 @@ def waitall():
         torch.cuda.synchronize()
 
 
+def expand_dims(t, axis: int):
+    return t.unsqueeze(axis)
+
+
+def flip(t, axis):
+    return torch.flip(t, [axis] if isinstance(axis, int) else list(axis))
+
+
+def tile(t, reps):
+    return t.repeat(reps if isinstance(reps, (tuple, list)) else (reps,))
+
+
+def repeat(t, repeats: int, axis=None):
+    if axis is None:
+        return t.reshape(-1).repeat_interleave(repeats)
+    return t.repeat_interleave(repeats, dim=axis)
+
+
+def maximum(a, b):
+    return torch.maximum(a, torch.as_tensor(b, dtype=a.dtype, device=a.device))
+
+
+def minimum(a, b):
+    return torch.minimum(a, torch.as_tensor(b, dtype=a.dtype, device=a.device))
+
+
+def flatten(t):
+    return t.reshape(t.shape[0], -1)
+
+
+def squeeze(t, axis=None):
+    return t.squeeze() if axis is None else t.squeeze(axis)
+
+
 def take(t, indices, axis: int = 0):
     from .ops import functional as DF
     if axis == 0 and t.dim() == 2:

@@ -43,6 +43,19 @@ def test_nd_math_matches_closed_forms():
     nd.waitall()  # no-op on CPU, must not raise
 
 
+def test_nd_shape_ops():
+    t = torch.arange(6.0).reshape(2, 3)
+    assert nd.expand_dims(t, 0).shape == (1, 2, 3)
+    assert nd.flip(t, 1)[0].tolist() == [2.0, 1.0, 0.0]
+    assert nd.tile(torch.ones(2), 3).shape == (6,)
+    assert nd.repeat(torch.tensor([1.0, 2.0]), 2).tolist() == [1.0, 1.0, 2.0, 2.0]
+    assert nd.maximum(t, 2.0).max() == 5.0 and nd.maximum(t, 2.0).min() == 2.0
+    assert nd.minimum(t, 2.0).max() == 2.0
+    assert nd.flatten(torch.ones(2, 3, 4)).shape == (2, 12)
+    assert nd.squeeze(torch.ones(1, 3, 1)).shape == (3,)
+    assert nd.squeeze(torch.ones(1, 3, 1), 0).shape == (3, 1)
+
+
 def test_nd_save_load_roundtrip(tmp_path):
     f = str(tmp_path / "x.params")
     nd.save(f, {"w": torch.randn(3, 3)})
