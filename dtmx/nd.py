@@ -184,7 +184,10 @@ def flip(t, axis):
 
 
 def tile(t, reps):
-    return t.repeat(reps if isinstance(reps, (tuple, list)) else (reps,))
+    reps = tuple(reps) if isinstance(reps, (tuple, list)) else (reps,)
+    if len(reps) < t.dim():  # mx.nd.tile pads reps with 1s on the left
+        reps = (1,) * (t.dim() - len(reps)) + reps
+    return t.repeat(reps)
 
 
 def repeat(t, repeats: int, axis=None):
