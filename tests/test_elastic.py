@@ -147,6 +147,30 @@ def test_removal_beats_addition(tmp_path):
     assert "c#0" in sched.members  # initial worker retained
 
 
+def test_remove_then_readd_same_host(tmp_path):
+    """A removed (non-initial) worker id can re-join later: the scheduler
+    publishes a fresh generation and re-seeds its heartbeat (reference
+    workflow: shrink the hostfile, later grow it again with the same
+    host)."""
+    port = _free_port()
+    sched = Scheduler("127.0.0.1", port, ["a", "b"],
+                      hostfile=str(tmp_path / "hosts"))
+    gens = sched.publish(["a", "b", "c"])      # add c
+    assert sched.members == ["a", "b", "c"] and len(gens) == 1
+    sched.publish(["a", "b"])                  # remove c
+    assert sched.members == ["a", "b"]
+    v_before = sched.version
+    gens = sched.publish(["a", "b", "c"])      # re-add the same id
+    assert sched.members == ["a", "b", "c"]
+    assert sched.version == v_before + 1
+    # re-admission re-seeds the heartbeat so it is not instantly "dead"
+    import time as _time
+    ts = float(sched.store.get("hb/c"))
+    assert _time.time() - ts < 5
+    log = (tmp_path / "hosts_log").read_text()
+    assert log.count("ADDED c") == 2 and log.count("REMOVED c") == 1
+
+
 def test_heartbeats_and_dead_node_detection(monkeypatch):
     """Store-based heartbeats + num_dead_node (reference van.cc:686-698
     heartbeat ledger / Postoffice dead-node accounting)."""
