@@ -27,6 +27,24 @@ def test_cpu_quantize_matches_reference():
     assert np.allclose(out2.numpy(), q2)
 
 
+def test_residuals_keyed_per_kvstore_key():
+    """Two SAME-SHAPED keys pushed alternately must keep independent
+    error-feedback residuals (advisor finding: metadata-keyed residuals
+    aliased same-shaped params; reference keeps residual_[key] per key,
+    kvstore_dist.h:778). Repro: 0.3-gradients quantize to 0 with residual
+    0.3; an aliased residual would make the SECOND key emit 0.5."""
+    comp = TwoBitCompression(0.5)
+    ga = torch.full((8,), 0.3)
+    gb = torch.full((8,), 0.3)
+    out_a = comp.compress_decompress(ga.clone(), key=1)
+    out_b = comp.compress_decompress(gb.clone(), key=2)
+    assert out_a.eq(0).all()
+    assert out_b.eq(0).all(), "key 2 consumed key 1's residual"
+    # round 2: each key's own residual (0.3) + 0.3 crosses the threshold
+    assert comp.compress_decompress(ga.clone(), key=1).eq(0.5).all()
+    assert comp.compress_decompress(gb.clone(), key=2).eq(0.5).all()
+
+
 def test_pack_unpack_roundtrip():
     comp = TwoBitCompression(0.25)
     g = torch.randn(100)
