@@ -147,6 +147,60 @@ def _dist_worker(rank, world, port, q):
         q.put(("err", rank, traceback.format_exc()))
 
 
+def _dist_compress_worker(rank, world, port, q):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+        DTMX_BACKEND="gloo",
+    )
+    try:
+        import dtmx
+
+        kv = dtmx.kvstore.create("dist_sync")
+        kv.set_gradient_compression({"type": "2bit", "threshold": 0.5})
+        kv.init("a", torch.zeros(8))
+        kv.init("b", torch.zeros(8))
+        # 0.3-gradients: quantize to 0 (residual 0.3) on push 1, to 0.5 on
+        # push 2 — per-key residuals must not alias (kvstore_dist.h:778)
+        for _ in range(2):
+            kv.push("a", torch.full((8,), 0.3))
+            kv.push("b", torch.full((8,), 0.3))
+        a = kv.pull("a")[0].clone()
+        b = kv.pull("b")[0].clone()
+        kv.close()
+        q.put(("ok", rank, {"a": a.tolist(), "b": b.tolist()}))
+    except Exception:
+        import traceback
+        q.put(("err", rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(180)
+def test_dist_compressed_push_per_key_residuals():
+    """2-bit compression through DistKVStore.push on gloo world 2: stored =
+    sum over ranks of quantized grads; second push crosses the threshold
+    via each key's OWN residual (reference dist_sync_kvstore.py compression
+    checks + the per-key residual fix)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = torch.randint(20000, 40000, (1,)).item()
+    procs = [ctx.Process(target=_dist_compress_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in procs:
+        status, rank, payload = q.get(timeout=120)
+        assert status == "ok", payload
+        res[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+    assert res[0] == res[1]
+    # push 1: both ranks quantize 0.3 -> 0; push 2: residual 0.3+0.3 -> 0.5
+    # per rank, summed over 2 ranks = 1.0 per element (no updater: SUM)
+    assert res[0]["a"] == [1.0] * 8, res[0]["a"]
+    assert res[0]["b"] == [1.0] * 8, res[0]["b"]
+
+
 @pytest.mark.timeout(180)
 def test_dist_row_sparse_push_merges_rows():
     ctx = mp.get_context("spawn")
