@@ -26,7 +26,9 @@ import torch  # noqa: E402
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--gpus", type=int, default=1,
+                    help="informational (driver flag); the actual rank count "
+                         "comes from WORLD_SIZE set by torch.distributed.run")
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--batch-size", type=int, default=4096, help="per-GPU batch (218 GiB of the 288 GB HBM3E at 4096)")
